@@ -1,0 +1,195 @@
+"""2-hidden-layer 4096-d MLP regressor — the MFMA GEMM path.
+
+BASELINE.json config 5: "swap regressor for 2-layer MLP (4096-d, exercises
+MFMA GEMM path)".  Architecture for the 1-feature regression task:
+
+    x (n,)  --expand1d-->  h1 = relu(x * w1 + b1)      (n, H)   fused outer
+    h1      --gemm_bf16-->  h2 = relu(h1 @ W2 + b2)    (n, H)   MFMA hot op
+    h2      --rowdot---->   yhat = h2 @ w3 + b3        (n,)     wave-reduce
+
+The H x H middle GEMM is the MFMA showcase; layer 1 and 3 are fused
+elementwise/reduction kernels instead of degenerate K=1 / N=1 GEMMs (a
+K=1 MFMA launch would waste the matrix cores — MI355X-first design).
+
+Training: minibatch Adam on MSE; forward/backward matmuls are the
+hand-written bf16 MFMA kernels (fp32 accumulate), optimizer state fp32.
+In DP training gradients are bucket-all-reduced over RCCL.
+
+Artefact: a *real sklearn ``MLPRegressor``* with ``coefs_``/``intercepts_``
+injected, so ``joblib.load`` + ``predict`` work with stock sklearn
+(format parity with reference ``stage_1:111-125``).
+"""
+from __future__ import annotations
+
+import math
+
+import torch
+
+from bodywork_mlops_demo_amd import ops
+
+
+class GPUMLPRegressor:
+    HIDDEN = 4096
+
+    def __init__(self, hidden: int = 4096, device="cpu", seed: int = 7):
+        self.hidden = hidden
+        self.device = torch.device(device)
+        g = torch.Generator(device="cpu").manual_seed(seed)
+        h = hidden
+        # He init, fp32 master weights
+        self.w1 = torch.randn(h, generator=g) * math.sqrt(2.0)
+        self.b1 = torch.zeros(h)
+        self.W2 = torch.randn(h, h, generator=g) * math.sqrt(2.0 / h)
+        self.b2 = torch.zeros(h)
+        self.w3 = torch.randn(h, generator=g) * math.sqrt(2.0 / h)
+        self.b3 = torch.zeros(1)
+        self._to_device()
+        self._opt_state = None
+
+    # -- device/dtype management ------------------------------------------
+    def _to_device(self):
+        d = self.device
+        for name in ("w1", "b1", "W2", "b2", "w3", "b3"):
+            setattr(self, name, getattr(self, name).to(d))
+        self._refresh_bf16()
+
+    def _refresh_bf16(self):
+        self.w1_bf = self.w1.bfloat16()
+        self.b1_bf = self.b1.bfloat16()
+        self.W2_bf = self.W2.bfloat16()
+        self.W2t_bf = self.W2.t().contiguous().bfloat16()  # for NT backward
+        self.b2_bf = self.b2.bfloat16()
+        self.w3_bf = self.w3.bfloat16()
+
+    def to(self, device):
+        self.device = torch.device(device)
+        self._to_device()
+        return self
+
+    def parameters(self) -> list[torch.Tensor]:
+        return [self.w1, self.b1, self.W2, self.b2, self.w3, self.b3]
+
+    # -- forward -----------------------------------------------------------
+    def _forward(self, x: torch.Tensor):
+        h1 = ops.expand1d_bf16(x, self.w1_bf, self.b1_bf, relu=True)
+        h2 = ops.gemm_bf16(h1, self.W2_bf, bias=self.b2_bf, relu=True)
+        yhat = ops.rowdot_bf16(h2, self.w3_bf, float(self.b3.item()))
+        return yhat, h1, h2
+
+    def predict(self, X: torch.Tensor) -> torch.Tensor:
+        yhat, _, _ = self._forward(X.to(self.device))
+        return yhat
+
+    # -- training ----------------------------------------------------------
+    def fit(
+        self,
+        X: torch.Tensor,
+        y: torch.Tensor,
+        steps: int = 200,
+        batch_size: int = 65536,
+        lr: float = 3e-4,
+        process_group=None,
+        seed: int = 42,
+    ):
+        """Minibatch Adam on MSE.  With ``process_group``, each rank holds
+        a shard of (X, y); gradients are all-reduce-averaged per step
+        (flat fp32 bucket — one RCCL launch per step)."""
+        n = X.shape[0]
+        g = torch.Generator(device="cpu").manual_seed(seed)
+        if self._opt_state is None:
+            self._opt_state = [
+                (torch.zeros_like(p), torch.zeros_like(p)) for p in self.parameters()
+            ]
+        t = 0
+        world = 1
+        if process_group is not None:
+            import torch.distributed as dist
+
+            world = dist.get_world_size(process_group)
+        for _ in range(steps):
+            idx = torch.randint(0, n, (min(batch_size, n),), generator=g)
+            xb = X[idx.to(X.device)]
+            yb = y[idx.to(y.device)]
+            grads = self._step_grads(xb, yb)
+            if process_group is not None:
+                import torch.distributed as dist
+
+                flat = torch.cat([gr.reshape(-1) for gr in grads])
+                dist.all_reduce(flat, group=process_group)
+                flat /= world
+                off = 0
+                for i, gr in enumerate(grads):
+                    grads[i] = flat[off:off + gr.numel()].view_as(gr)
+                    off += gr.numel()
+            t += 1
+            self._adam_update(grads, lr, t)
+            self._refresh_bf16()
+        return self
+
+    def _step_grads(self, xb: torch.Tensor, yb: torch.Tensor) -> list[torch.Tensor]:
+        nb = xb.shape[0]
+        yhat, h1, h2 = self._forward(xb)
+        dy = (2.0 / nb) * (yhat - yb.float())                   # (n,)
+        # layer 3: yhat = h2 @ w3 + b3
+        dw3 = ops.coldot_bf16(h2, dy)                           # (H,)
+        db3 = dy.sum().reshape(1)
+        # dh2 = outer(dy, w3) * relu'(h2): fused expand with mask
+        dz2 = ops.expand1d_bf16(dy, self.w3_bf, None, relu=False, mask=h2)
+        # layer 2: h2 = relu(h1 @ W2 + b2)
+        dW2 = ops.gemm_bf16(h1, dz2, trans_a=True, out_fp32=True)   # (H,H)
+        db2 = ops.colsum_bf16(dz2)
+        dz1 = ops.gemm_bf16(dz2, self.W2t_bf, mask=h1)          # (n,H)*relu'(h1)
+        # layer 1: h1 = relu(x w1 + b1)
+        dw1, db1 = ops.coldot_bf16(dz1, xb.float(), also_colsum=True)
+        return [dw1, db1, dW2, db2, dw3, db3]
+
+    def _adam_update(self, grads, lr: float, t: int, beta1=0.9, beta2=0.999, eps=1e-8):
+        for p, gr, (m, v) in zip(self.parameters(), grads, self._opt_state):
+            gr = gr.to(p.dtype)
+            m.mul_(beta1).add_(gr, alpha=1 - beta1)
+            v.mul_(beta2).addcmul_(gr, gr, value=1 - beta2)
+            mhat = m / (1 - beta1**t)
+            vhat = v / (1 - beta2**t)
+            p.sub_(lr * mhat / (vhat.sqrt() + eps))
+
+    # -- artefact compatibility --------------------------------------------
+    def to_sklearn(self):
+        import numpy as np
+        from sklearn.neural_network import MLPRegressor
+
+        h = self.hidden
+        m = MLPRegressor(hidden_layer_sizes=(h, h), activation="relu")
+        m.coefs_ = [
+            self.w1.detach().cpu().numpy().reshape(1, h).astype(np.float64),
+            self.W2.detach().cpu().numpy().astype(np.float64),
+            self.w3.detach().cpu().numpy().reshape(h, 1).astype(np.float64),
+        ]
+        m.intercepts_ = [
+            self.b1.detach().cpu().numpy().astype(np.float64),
+            self.b2.detach().cpu().numpy().astype(np.float64),
+            self.b3.detach().cpu().numpy().astype(np.float64),
+        ]
+        m.n_layers_ = 4
+        m.n_outputs_ = 1
+        m.out_activation_ = "identity"
+        m.n_features_in_ = 1
+        return m
+
+    @classmethod
+    def from_sklearn(cls, m, device="cpu") -> "GPUMLPRegressor":
+        h = m.coefs_[1].shape[0]
+        self = cls.__new__(cls)
+        self.hidden = h
+        self.device = torch.device(device)
+        self.w1 = torch.from_numpy(m.coefs_[0].reshape(-1).copy()).float()
+        self.b1 = torch.from_numpy(m.intercepts_[0].copy()).float()
+        self.W2 = torch.from_numpy(m.coefs_[1].copy()).float()
+        self.b2 = torch.from_numpy(m.intercepts_[1].copy()).float()
+        self.w3 = torch.from_numpy(m.coefs_[2].reshape(-1).copy()).float()
+        self.b3 = torch.from_numpy(m.intercepts_[2].copy()).float()
+        self._opt_state = None
+        self._to_device()
+        return self
+
+    def __repr__(self) -> str:
+        return f"MLPRegressor(hidden_layer_sizes=({self.hidden}, {self.hidden}))"

@@ -1,0 +1,308 @@
+"""Compute ops: hand-written CDNA4 HIP kernels with CPU torch oracles.
+
+The reference delegates all math to libraries (sklearn ``fit``/``predict``
+at ``stage_1:105-106``/``stage_2:78``, numpy RNG at ``stage_3:39-40``);
+here every computational primitive is a first-class gfx950 kernel
+(SURVEY.md §2.2 mapping table):
+
+- ``datagen``        — on-GPU philox4x32 synthetic-drift generator with
+                       in-kernel y>=0 stream compaction (stage_3:28-43)
+- ``linreg_stats``   — fused XtX/Xty statistics reduction for the
+                       closed-form OLS fit (stage_1:105-106)
+- ``linear_score``   — batched linear scoring (stage_2:78)
+- ``regression_metrics`` — single-pass fused MAPE/R^2/max-residual
+                       reduction (stage_1:79-90, stage_4:101-113)
+- ``gemm_bf16``      — MFMA (v_mfma_f32_16x16x32_bf16) LDS-tiled dense
+                       GEMM with fused bias+ReLU epilogue for the MLP path
+
+Dispatch policy: on a CUDA/HIP device the in-tree ``_hipcore`` extension
+is REQUIRED — if it is missing the ops raise instead of silently falling
+back to eager PyTorch.  On CPU the pure-torch reference implementations
+(:mod:`bodywork_mlops_demo_amd.ops.reference`) run; they double as the
+numerics oracles in the test suite.
+"""
+from __future__ import annotations
+
+import glob
+import os
+
+import torch
+
+from bodywork_mlops_demo_amd.ops import reference
+
+_HIPCORE = None
+_HIPCORE_ERR: str | None = None
+
+
+def _load_extension():
+    global _HIPCORE, _HIPCORE_ERR
+    if _HIPCORE is not None or _HIPCORE_ERR is not None:
+        return _HIPCORE
+    here = os.path.dirname(__file__)
+    cands = sorted(glob.glob(os.path.join(here, "_hipcore*.so")))
+    if not cands:
+        _HIPCORE_ERR = (
+            "in-tree HIP extension _hipcore*.so not found under "
+            f"{here}; build it with `python setup.py build_ext --inplace` "
+            "(PYTORCH_ROCM_ARCH=gfx950)"
+        )
+        return None
+    try:
+        torch.ops.load_library(cands[0])
+        _HIPCORE = torch.ops.bodywork_hip
+    except Exception as e:  # loud: a broken build must not silently fall back
+        _HIPCORE_ERR = f"failed to load {cands[0]}: {e}"
+        raise RuntimeError(_HIPCORE_ERR) from e
+    return _HIPCORE
+
+
+def hip_available() -> bool:
+    """True when the _hipcore extension is importable (CPU box included)."""
+    try:
+        return _load_extension() is not None
+    except RuntimeError:
+        return False
+
+
+def _core(device: torch.device):
+    """Return the HIP extension for a CUDA tensor op; raise loudly if absent."""
+    if device.type != "cuda":
+        return None
+    ext = _load_extension()
+    if ext is None:
+        raise RuntimeError(
+            "GPU tensor passed but the _hipcore HIP extension is not built: "
+            + str(_HIPCORE_ERR)
+        )
+    return ext
+
+
+# --------------------------------------------------------------------------
+# datagen — reference stage_3:28-43 semantics
+# --------------------------------------------------------------------------
+
+def datagen(
+    n: int,
+    day_of_year: int,
+    seed: int,
+    f: float = 6.0,
+    kappa: float = 1.0,
+    A: float = 0.5,
+    beta: float = 0.5,
+    sigma: float = 10.0,
+    device: str | torch.device = "cpu",
+    stream_offset: int = 0,
+) -> tuple[torch.Tensor, torch.Tensor]:
+    """Generate ``y = alpha(d) + beta*X + sigma*eps`` rows with ``y >= 0``.
+
+    ``alpha(d) = kappa + A*sin(2*pi*f*(d-1)/364)`` (reference stage_3:31-33).
+    X ~ U(0,100), eps ~ N(0,1) via counter-based philox4x32-10, so the
+    stream is reproducible for a (seed, stream_offset) pair on any device
+    count — rank r of a DP world passes its row offset as stream_offset.
+    Returns (y, X) float32 tensors on ``device`` (length <= n after the
+    y>=0 cull, reference stage_3:43).
+    """
+    device = torch.device(device)
+    alpha = reference.alpha(day_of_year, f, kappa, A)
+    if device.type == "cuda":
+        core = _core(device)
+        y, X = core.datagen(n, seed, stream_offset, alpha, beta, sigma)
+        return y, X
+    return reference.datagen_cpu(n, seed, stream_offset, alpha, beta, sigma)
+
+
+def alpha(day_of_year: int, f: float = 6.0, kappa: float = 1.0, A: float = 0.5) -> float:
+    return reference.alpha(day_of_year, f, kappa, A)
+
+
+# --------------------------------------------------------------------------
+# OLS fit statistics — reference stage_1:105-106 (LinearRegression.fit)
+# --------------------------------------------------------------------------
+
+def linreg_stats(X: torch.Tensor, y: torch.Tensor) -> torch.Tensor:
+    """Single-pass fused sufficient statistics for 1-feature OLS.
+
+    Returns a float64 tensor ``[n, sum_x, sum_y, sum_xx, sum_xy]`` (on the
+    same device).  The closed-form solve is host-side (2x2).  In DP
+    training these five scalars are the entire all-reduce payload.
+    """
+    if X.device.type == "cuda":
+        core = _core(X.device)
+        return core.linreg_stats(X.contiguous(), y.contiguous())
+    return reference.linreg_stats_cpu(X, y)
+
+
+def solve_ols(stats: torch.Tensor) -> tuple[float, float]:
+    """Closed-form (intercept, coef) from fused stats (2x2 normal equations)."""
+    n, sx, sy, sxx, sxy = stats.tolist()
+    denom = n * sxx - sx * sx
+    if abs(denom) < 1e-30:
+        return float(sy / max(n, 1.0)), 0.0
+    coef = (n * sxy - sx * sy) / denom
+    intercept = (sy - coef * sx) / n
+    return intercept, coef
+
+
+# --------------------------------------------------------------------------
+# scoring — reference stage_2:78 (model.predict)
+# --------------------------------------------------------------------------
+
+def linear_score(X: torch.Tensor, intercept: float, coef: float) -> torch.Tensor:
+    """Batched linear scoring ``yhat = intercept + coef * X`` (fp32)."""
+    if X.device.type == "cuda":
+        core = _core(X.device)
+        return core.linear_score(X.contiguous(), intercept, coef)
+    return reference.linear_score_cpu(X, intercept, coef)
+
+
+# --------------------------------------------------------------------------
+# metrics — reference stage_1:79-90 and stage_4:101-113
+# --------------------------------------------------------------------------
+
+def regression_metrics(y: torch.Tensor, yhat: torch.Tensor) -> dict:
+    """Offline model metrics in one fused pass: MAPE, R^2, max residual.
+
+    Matches sklearn's definitions used by the reference
+    (``mean_absolute_percentage_error``, ``r2_score``, ``max_error`` —
+    stage_1:79-90): MAPE uses |y - yhat| / max(|y|, eps).
+    """
+    if y.device.type == "cuda":
+        core = _core(y.device)
+        out = core.regression_metrics(y.contiguous(), yhat.contiguous())
+        n, s_ape, ss_res, s_y, s_yy, max_res = out.tolist()
+    else:
+        n, s_ape, ss_res, s_y, s_yy, max_res = reference.metric_sums_cpu(y, yhat)
+    mape = s_ape / n
+    ss_tot = s_yy - s_y * s_y / n
+    r2 = 1.0 - ss_res / ss_tot if ss_tot > 0 else 0.0
+    return {"MAPE": mape, "r_squared": r2, "max_residual": max_res}
+
+
+def score_label_metrics(scores: torch.Tensor, labels: torch.Tensor) -> dict:
+    """Online (live-service) metrics in one fused pass.
+
+    Matches reference stage_4:101-113: MAPE = mean |score/label - 1|,
+    r_squared = Pearson correlation(score, label), max_residual = max APE.
+    """
+    if scores.device.type == "cuda":
+        core = _core(scores.device)
+        out = core.score_label_metrics(scores.contiguous(), labels.contiguous())
+        n, s_ape, max_ape, s_s, s_l, s_ss, s_ll, s_sl = out.tolist()
+    else:
+        (n, s_ape, max_ape, s_s, s_l, s_ss, s_ll, s_sl) = (
+            reference.score_label_sums_cpu(scores, labels)
+        )
+    mape = s_ape / n
+    cov = s_sl - s_s * s_l / n
+    var_s = s_ss - s_s * s_s / n
+    var_l = s_ll - s_l * s_l / n
+    corr = cov / (var_s * var_l) ** 0.5 if var_s > 0 and var_l > 0 else 0.0
+    return {"MAPE": mape, "r_squared": corr, "max_residual": max_ape}
+
+
+# --------------------------------------------------------------------------
+# train/test split — reference stage_1:98-103 (train_test_split, seed 42)
+# --------------------------------------------------------------------------
+
+def train_test_split_indices(
+    n: int, test_size: float = 0.2, seed: int = 42, device: str | torch.device = "cpu"
+) -> tuple[torch.Tensor, torch.Tensor]:
+    """Random permutation split (GPU permutation via philox keyed sort)."""
+    device = torch.device(device)
+    g = torch.Generator(device="cpu").manual_seed(seed)
+    perm = torch.randperm(n, generator=g)
+    n_test = int(round(n * test_size))
+    test_idx, train_idx = perm[:n_test], perm[n_test:]
+    if device.type == "cuda":
+        return train_idx.to(device), test_idx.to(device)
+    return train_idx, test_idx
+
+
+# --------------------------------------------------------------------------
+# MFMA GEMM — MLP path
+# --------------------------------------------------------------------------
+
+def gemm_bf16(
+    a: torch.Tensor,
+    b: torch.Tensor,
+    bias: torch.Tensor | None = None,
+    relu: bool = False,
+    trans_a: bool = False,
+    mask: torch.Tensor | None = None,
+    out_fp32: bool = False,
+) -> torch.Tensor:
+    """C = op(A) @ B with a fused epilogue, bf16 in / fp32 accumulate.
+
+    ``op(A) = A.T`` when ``trans_a`` (the dW = X^T @ dY backward shape).
+    Epilogue (fused into the C-write, never a second pass over HBM):
+    ``+bias`` then ``relu`` , or ``* (mask > 0)`` (ReLU backward).
+    GPU: hand-written gfx950 MFMA kernel (v_mfma_f32_16x16x32_bf16,
+    LDS-tiled).  CPU oracle: fp32 torch matmul.
+    """
+    if a.device.type == "cuda":
+        core = _core(a.device)
+        return core.gemm_bf16(
+            a.contiguous(), b.contiguous(),
+            bias.contiguous() if bias is not None else None,
+            relu, trans_a,
+            mask.contiguous() if mask is not None else None,
+            out_fp32,
+        )
+    return reference.gemm_bf16_cpu(a, b, bias, relu, trans_a, mask, out_fp32)
+
+
+def expand1d_bf16(
+    x: torch.Tensor,
+    w: torch.Tensor,
+    b: torch.Tensor | None = None,
+    relu: bool = False,
+    mask: torch.Tensor | None = None,
+) -> torch.Tensor:
+    """Fused rank-1 expansion: out[i,j] = act(x[i]*w[j] + b[j]).
+
+    The MLP's 1-feature input layer (and its backward dh = outer(dy, w))
+    as a single bandwidth-bound kernel instead of a degenerate K=1 MFMA
+    GEMM.  ``mask`` multiplies by (mask>0) — ReLU backward.  Returns bf16
+    (n, H).
+    """
+    if x.device.type == "cuda":
+        core = _core(x.device)
+        return core.expand1d_bf16(
+            x.contiguous().float(), w.contiguous(),
+            b.contiguous() if b is not None else None,
+            relu,
+            mask.contiguous() if mask is not None else None,
+        )
+    return reference.expand1d_cpu(x, w, b, relu, mask)
+
+
+def rowdot_bf16(h: torch.Tensor, w: torch.Tensor, b: float = 0.0) -> torch.Tensor:
+    """out[i] = sum_j h[i,j] * w[j] + b — the MLP output head (GEMV).
+
+    One wave per row, vectorised bf16x8 loads, shuffle reduction;
+    fp32 output (scoring precision — BASELINE "bf16 fit + fp32 scoring").
+    """
+    if h.device.type == "cuda":
+        core = _core(h.device)
+        return core.rowdot_bf16(h.contiguous(), w.contiguous(), b)
+    return reference.rowdot_cpu(h, w, b)
+
+
+def coldot_bf16(
+    m: torch.Tensor, v: torch.Tensor, also_colsum: bool = False
+):
+    """dw[j] = sum_i m[i,j] * v[i] (fp32), optionally fused with
+    cs[j] = sum_i m[i,j] in the same pass (bias gradients)."""
+    if m.device.type == "cuda":
+        core = _core(m.device)
+        out = core.coldot_bf16(m.contiguous(), v.contiguous().float(), also_colsum)
+        return (out[0], out[1]) if also_colsum else out[0]
+    return reference.coldot_cpu(m, v, also_colsum)
+
+
+def colsum_bf16(m: torch.Tensor) -> torch.Tensor:
+    """cs[j] = sum_i m[i,j] (fp32)."""
+    if m.device.type == "cuda":
+        core = _core(m.device)
+        return core.colsum_bf16(m.contiguous())
+    return reference.colsum_cpu(m)

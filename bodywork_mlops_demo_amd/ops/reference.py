@@ -1,0 +1,199 @@
+"""CPU reference implementations of every HIP op.
+
+These are (a) the execution path on CPU-only machines and (b) the
+numerics oracles the GPU kernels are tested against.  The philox4x32-10
+generator here is bit-identical in its integer stream to the device
+kernel in ``ops/hip/datagen.hip``, so datagen parity tests compare
+float results to ~1e-5 (libm vs device transcendental differences only).
+
+Semantics follow reference ``stage_3_synthetic_data_generation.py:28-43``
+(drift model), ``stage_1_train_model.py:79-108`` (fit + metrics) and
+``stage_4_test_model_scoring_service.py:87-113`` (live metrics).
+"""
+from __future__ import annotations
+
+import math
+
+import numpy as np
+import torch
+
+# --------------------------------------------------------------------------
+# philox4x32-10 counter-based RNG (vectorised numpy, uint32 lattice)
+# --------------------------------------------------------------------------
+
+_PHILOX_M0 = np.uint64(0xD2511F53)
+_PHILOX_M1 = np.uint64(0xCD9E8D57)
+_W0 = np.uint32(0x9E3779B9)
+_W1 = np.uint32(0xBB67AE85)
+
+
+def philox4x32(counter: np.ndarray, key0: int, key1: int) -> np.ndarray:
+    """10-round philox4x32; ``counter`` is uint64 array of N counters.
+
+    Counter i is expanded to the 128-bit counter (lo=i, hi=0, 0, 0).
+    Returns (N, 4) uint32.
+    """
+    n = counter.shape[0]
+    c0 = (counter & 0xFFFFFFFF).astype(np.uint32)
+    c1 = (counter >> np.uint64(32)).astype(np.uint32)
+    c2 = np.zeros(n, np.uint32)
+    c3 = np.zeros(n, np.uint32)
+    k0 = np.uint32(key0 & 0xFFFFFFFF)
+    k1 = np.uint32(key1 & 0xFFFFFFFF)
+    with np.errstate(over="ignore"):  # uint32 wraparound is the algorithm
+        for _ in range(10):
+            p0 = _PHILOX_M0 * c0.astype(np.uint64)
+            p1 = _PHILOX_M1 * c2.astype(np.uint64)
+            hi0 = (p0 >> np.uint64(32)).astype(np.uint32)
+            lo0 = p0.astype(np.uint32)
+            hi1 = (p1 >> np.uint64(32)).astype(np.uint32)
+            lo1 = p1.astype(np.uint32)
+            c0, c1, c2, c3 = hi1 ^ c1 ^ k0, lo1, hi0 ^ c3 ^ k1, lo0
+            k0 = k0 + _W0
+            k1 = k1 + _W1
+    return np.stack([c0, c1, c2, c3], axis=1)
+
+
+def alpha(day_of_year: int, f: float = 6.0, kappa: float = 1.0, A: float = 0.5) -> float:
+    """Sinusoidal drift intercept (reference stage_3:31-33)."""
+    return kappa + A * math.sin(2.0 * math.pi * f * (day_of_year - 1) / 364.0)
+
+
+def datagen_cpu(
+    n: int,
+    seed: int,
+    stream_offset: int,
+    alpha_now: float,
+    beta: float,
+    sigma: float,
+) -> tuple[torch.Tensor, torch.Tensor]:
+    """y = alpha + beta*X + sigma*eps with X~U(0,100), eps~N(0,1), y>=0 cull.
+
+    Row i consumes philox counter ``stream_offset + i``; the y>=0 filter is
+    stable (row order preserved) exactly like pandas ``query`` in the
+    reference (stage_3:43).
+    """
+    counters = (np.arange(n, dtype=np.uint64) + np.uint64(stream_offset))
+    r = philox4x32(counters, seed, seed >> 32 if seed > 0xFFFFFFFF else 0x1F123BB5)
+    inv32 = np.float32(1.0 / 4294967296.0)
+    X = (r[:, 0].astype(np.float32) * inv32 * np.float32(100.0)).astype(np.float32)
+    u1 = (r[:, 1].astype(np.float32) + np.float32(0.5)) * inv32
+    u2 = r[:, 2].astype(np.float32) * inv32
+    eps = np.sqrt(np.float32(-2.0) * np.log(u1)) * np.cos(
+        np.float32(2.0 * math.pi) * u2
+    )
+    y = (
+        np.float32(alpha_now)
+        + np.float32(beta) * X
+        + np.float32(sigma) * eps.astype(np.float32)
+    ).astype(np.float32)
+    keep = y >= 0.0
+    return torch.from_numpy(y[keep].copy()), torch.from_numpy(X[keep].copy())
+
+
+# --------------------------------------------------------------------------
+# OLS statistics / scoring / metrics oracles (fp64 accumulation)
+# --------------------------------------------------------------------------
+
+def linreg_stats_cpu(X: torch.Tensor, y: torch.Tensor) -> torch.Tensor:
+    x64 = X.double()
+    y64 = y.double()
+    return torch.stack([
+        torch.tensor(float(X.numel()), dtype=torch.float64),
+        x64.sum(),
+        y64.sum(),
+        (x64 * x64).sum(),
+        (x64 * y64).sum(),
+    ])
+
+
+def linear_score_cpu(X: torch.Tensor, intercept: float, coef: float) -> torch.Tensor:
+    return (intercept + coef * X.float()).float()
+
+
+_MAPE_EPS = float(np.finfo(np.float64).eps)  # sklearn's epsilon
+
+
+def metric_sums_cpu(y: torch.Tensor, yhat: torch.Tensor):
+    y64, p64 = y.double(), yhat.double()
+    res = y64 - p64
+    ape = res.abs() / torch.clamp(y64.abs(), min=_MAPE_EPS)
+    return (
+        float(y.numel()),
+        float(ape.sum()),
+        float((res * res).sum()),
+        float(y64.sum()),
+        float((y64 * y64).sum()),
+        float(res.abs().max()),
+    )
+
+
+def score_label_sums_cpu(scores: torch.Tensor, labels: torch.Tensor):
+    s64, l64 = scores.double(), labels.double()
+    ape = (s64 / l64 - 1.0).abs()  # reference stage_4:87-90 (no eps guard)
+    return (
+        float(scores.numel()),
+        float(ape.sum()),
+        float(ape.max()),
+        float(s64.sum()),
+        float(l64.sum()),
+        float((s64 * s64).sum()),
+        float((l64 * l64).sum()),
+        float((s64 * l64).sum()),
+    )
+
+
+# --------------------------------------------------------------------------
+# GEMM oracle
+# --------------------------------------------------------------------------
+
+def gemm_bf16_cpu(
+    a: torch.Tensor,
+    b: torch.Tensor,
+    bias: torch.Tensor | None = None,
+    relu: bool = False,
+    trans_a: bool = False,
+    mask: torch.Tensor | None = None,
+    out_fp32: bool = False,
+) -> torch.Tensor:
+    a32 = a.float().t() if trans_a else a.float()
+    c = a32 @ b.float()
+    if bias is not None:
+        c = c + bias.float()
+    if relu:
+        c = torch.relu(c)
+    if mask is not None:
+        c = c * (mask.float() > 0)
+    return c if out_fp32 else c.bfloat16()
+
+
+def expand1d_cpu(
+    x: torch.Tensor,
+    w: torch.Tensor,
+    b: torch.Tensor | None = None,
+    relu: bool = False,
+    mask: torch.Tensor | None = None,
+) -> torch.Tensor:
+    out = torch.outer(x.float(), w.float())
+    if b is not None:
+        out = out + b.float()
+    if relu:
+        out = torch.relu(out)
+    if mask is not None:
+        out = out * (mask.float() > 0)
+    return out.bfloat16()
+
+
+def rowdot_cpu(h: torch.Tensor, w: torch.Tensor, b: float = 0.0) -> torch.Tensor:
+    return h.float() @ w.float() + b
+
+
+def coldot_cpu(m: torch.Tensor, v: torch.Tensor, also_colsum: bool = False):
+    dw = m.float().t() @ v.float()
+    if also_colsum:
+        return dw, m.float().sum(dim=0)
+    return dw
+
+
+def colsum_cpu(m: torch.Tensor) -> torch.Tensor:
+    return m.float().sum(dim=0)

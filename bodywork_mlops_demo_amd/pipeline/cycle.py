@@ -1,0 +1,188 @@
+"""The in-process train→serve→drift→test cycle (the perf path).
+
+One "day" of the reference pipeline: train on all data
+up to day t → deploy the model → generate day t+1's drifted dataset →
+test the deployed model on the unseen t+1 data.  The reference realises
+this as four k8s pods with S3 round-trips between them; here it is one
+process per GPU with tensors resident in HBM — the artefact store is
+written at the contract boundaries (model, metrics, dataset) but the hot
+path never re-reads what it already holds (SURVEY.md §7 hard part (e)).
+
+This is what ``bench.py`` times: BASELINE.json's "rows/sec scored
+(stage_2 path) + train-to-serve cycle wall-clock".
+"""
+from __future__ import annotations
+
+from datetime import date as date_t, timedelta
+from time import perf_counter
+
+import torch
+
+from bodywork_mlops_demo_amd import ops
+from bodywork_mlops_demo_amd.serving.scorer import BatchedScorer
+from bodywork_mlops_demo_amd.stages import loadtest as stage4
+from bodywork_mlops_demo_amd.stages import train as stage1
+from bodywork_mlops_demo_amd.store import ArtefactStore
+from bodywork_mlops_demo_amd.utils.logging import configure_logger
+
+log = configure_logger(__name__)
+
+
+class CycleState:
+    """Device-resident accumulated dataset history + current-day cursor."""
+
+    def __init__(self, device: str, start_date: date_t, rank: int = 0,
+                 world_size: int = 1):
+        self.device = device
+        self.date = start_date
+        self.rank = rank
+        self.world_size = world_size
+        self.y = torch.empty(0, device=device)
+        self.X = torch.empty(0, device=device)
+        self._next: tuple[torch.Tensor, torch.Tensor] | None = None
+
+    def append_day(self, y: torch.Tensor, X: torch.Tensor) -> None:
+        self.y = torch.cat([self.y, y])
+        self.X = torch.cat([self.X, X])
+
+
+def run_cycle(
+    state: CycleState,
+    store: ArtefactStore | None,
+    n_rows: int,
+    model_type: str = "linear",
+    process_group=None,
+    persist_fmt: str = "npz",
+    mlp_steps: int = 50,
+    mlp_batch_size: int = 65536,
+    use_graphs: bool = True,
+    scorer_cache: dict | None = None,
+) -> dict:
+    """Run one full cycle; returns per-phase timings + metrics.
+
+    In a DP world each rank generates/holds/trains on its own ``n_rows``
+    rows (weak scaling: per-GPU work fixed); rank 0 persists artefacts.
+    """
+    device = state.device
+    dev_cuda = device.startswith("cuda")
+    timings: dict[str, float] = {}
+
+    def sync():
+        if dev_cuda:
+            torch.cuda.synchronize()
+
+    # -- day-0 bootstrap: make sure there is data for today -----------------
+    if state.y.numel() == 0:
+        seed = state.date.toordinal() * 1000 + state.rank
+        y, X = ops.datagen(
+            n_rows, state.date.timetuple().tm_yday, seed, device=device,
+            stream_offset=state.rank * n_rows,
+        )
+        state.append_day(y, X)
+        if store is not None and state.rank == 0:
+            store.put_dataset(state.date, y.cpu().numpy(), X.cpu().numpy(),
+                              fmt=persist_fmt)
+
+    # -- stage 1: train on all accumulated data -----------------------------
+    sync()
+    t0 = perf_counter()
+    eff_store = store if store is not None else _NullStore()
+    metrics, trained = stage1.run(
+        eff_store,
+        model_type=model_type,
+        device=device,
+        process_group=process_group,
+        rank=state.rank,
+        world_size=1,  # data is already per-rank sharded by generation
+        mlp_steps=mlp_steps,
+        mlp_batch_size=mlp_batch_size,
+        data=(state.y, state.X, state.date),
+        return_model=True,
+    )
+    sync()
+    timings["train_s"] = perf_counter() - t0
+
+    # -- stage 2: deploy — joblib artefact round-trip, model into HBM,
+    #    hipGraphs captured (reference stage_2:108-119 semantics) ----------
+    t0 = perf_counter()
+    if store is not None:
+        if process_group is not None:  # rank 0 persisted; others wait
+            import torch.distributed as dist
+
+            dist.barrier(group=process_group)
+        model = _deploy_from_store(store, device)
+    else:
+        # store-less bench: every rank does an in-memory joblib round-trip
+        # so the deploy phase still pays serialization (artefact parity)
+        import io
+
+        import joblib
+
+        from bodywork_mlops_demo_amd.models import regressor_from_artifact
+
+        bio = io.BytesIO()
+        joblib.dump(trained.to_sklearn(), bio)
+        bio.seek(0)
+        model = regressor_from_artifact(joblib.load(bio), device)
+    scorer = BatchedScorer(model, device, use_graphs=use_graphs)
+    if scorer_cache is not None:
+        scorer_cache["scorer"] = scorer
+    sync()
+    timings["deploy_s"] = perf_counter() - t0
+
+    # -- stage 3: generate day t+1 ------------------------------------------
+    t0 = perf_counter()
+    next_date = state.date + timedelta(days=1)
+    seed = next_date.toordinal() * 1000 + state.rank
+    y_next, X_next = ops.datagen(
+        n_rows, next_date.timetuple().tm_yday, seed, device=device,
+        stream_offset=state.rank * n_rows,
+    )
+    if store is not None and state.rank == 0:
+        store.put_dataset(next_date, y_next.cpu().numpy(),
+                          X_next.cpu().numpy(), fmt=persist_fmt)
+    sync()
+    timings["datagen_s"] = perf_counter() - t0
+
+    # -- stage 4: test the deployed model on unseen t+1 data ----------------
+    t0 = perf_counter()
+    test_metrics = stage4.run(
+        store if store is not None else _NullStore(),
+        device=device,
+        scorer=scorer,
+        data=(y_next, X_next, next_date),
+        persist=store is not None and state.rank == 0,
+    )
+    sync()
+    timings["test_s"] = perf_counter() - t0
+    timings["rows_scored"] = int(y_next.shape[0])
+
+    # -- advance the clock ---------------------------------------------------
+    state.append_day(y_next, X_next)
+    state.date = next_date
+
+    timings["cycle_s"] = sum(
+        v for k, v in timings.items() if k.endswith("_s")
+    )
+    return {"timings": timings, "offline": metrics, "online": test_metrics}
+
+
+def _deploy_from_store(store, device):
+    """Stage-2 deployment: latest joblib artefact by key-date → HBM."""
+    from bodywork_mlops_demo_amd.models import regressor_from_artifact
+
+    artefact, _ = store.get_latest_model()
+    return regressor_from_artifact(artefact, device)
+
+
+class _NullStore:
+    """Store stub for store-less benchmarking (artefact writes elided)."""
+
+    def put_model(self, model, d):
+        return f"models/regressor-{d}.joblib"
+
+    def put_metrics_csv(self, key, header, row):
+        return None
+
+    def put_dataset(self, d, y, X, fmt="npz"):
+        return None

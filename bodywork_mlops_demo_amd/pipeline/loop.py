@@ -1,0 +1,107 @@
+"""Multi-day concept-drift loop (BASELINE config 4).
+
+The reference runs one cycle per *real* day via a k8s cronjob
+(``README.md:5``); its stages stamp artefacts with ``date.today()`` so
+more than one cycle/day is impossible.  This loop
+parameterises the date (virtual clock) and runs N retrain-and-redeploy
+cycles back-to-back, reporting the per-cycle wall-clock and leaving the
+full dated artefact history in the store — which is also how
+checkpoint/resume works: the loop resumes from the latest dataset date
+found in the store (the artefacts ARE the checkpoints).
+"""
+from __future__ import annotations
+
+import argparse
+import json
+from datetime import date as date_t, timedelta
+
+import torch
+
+from bodywork_mlops_demo_amd.pipeline.cycle import CycleState, run_cycle
+from bodywork_mlops_demo_amd.store import ArtefactStore, contract, open_store
+from bodywork_mlops_demo_amd.utils.logging import configure_logger
+
+log = configure_logger(__name__)
+
+
+def run_loop(
+    store: ArtefactStore | None,
+    days: int = 30,
+    n_rows: int = 24 * 60,
+    model_type: str = "linear",
+    device: str | None = None,
+    start_date: date_t | str = "2026-01-01",
+    persist_fmt: str = "csv",
+    process_group=None,
+    rank: int = 0,
+    world_size: int = 1,
+    resume: bool = True,
+) -> list[dict]:
+    device = device or ("cuda" if torch.cuda.is_available() else "cpu")
+    if isinstance(start_date, str):
+        start_date = date_t.fromisoformat(start_date)
+
+    # resume from the latest dataset date in the store (artefact=checkpoint)
+    if store is not None and resume:
+        try:
+            _, latest = store.latest(contract.DATASETS_PREFIX)
+            start_date = latest
+            log.info(f"resuming drift loop from latest store date {latest}")
+        except FileNotFoundError:
+            pass
+
+    state = CycleState(device, start_date, rank=rank, world_size=world_size)
+
+    # preload history from the store so training sees all prior days
+    if store is not None:
+        try:
+            y_np, X_np, _ = store.get_all_datasets()
+            state.append_day(
+                torch.from_numpy(y_np).to(device),
+                torch.from_numpy(X_np).to(device),
+            )
+            log.info(f"preloaded {y_np.shape[0]} historical rows from store")
+        except FileNotFoundError:
+            pass
+
+    results = []
+    for day in range(days):
+        r = run_cycle(
+            state, store, n_rows, model_type=model_type,
+            process_group=process_group, persist_fmt=persist_fmt,
+        )
+        results.append(r)
+        t = r["timings"]
+        log.info(
+            f"cycle {day + 1}/{days} ({state.date}): "
+            f"{t['cycle_s']:.3f}s (train {t['train_s']:.3f} deploy "
+            f"{t['deploy_s']:.3f} datagen {t['datagen_s']:.3f} "
+            f"test {t['test_s']:.3f}) online MAPE "
+            f"{r['online']['MAPE']:.4f}"
+        )
+    return results
+
+
+def main(argv=None) -> None:
+    p = argparse.ArgumentParser(description=__doc__)
+    p.add_argument("--store", default=None)
+    p.add_argument("--days", type=int, default=30)
+    p.add_argument("--rows", type=int, default=24 * 60)
+    p.add_argument("--model", default="linear", choices=["linear", "mlp"])
+    p.add_argument("--device", default=None)
+    p.add_argument("--start-date", default="2026-01-01")
+    p.add_argument("--format", default="csv", choices=["csv", "npz"])
+    p.add_argument("--json-out", default=None)
+    args = p.parse_args(argv)
+    results = run_loop(
+        open_store(args.store), days=args.days, n_rows=args.rows,
+        model_type=args.model, device=args.device,
+        start_date=args.start_date, persist_fmt=args.format,
+    )
+    if args.json_out:
+        with open(args.json_out, "w") as f:
+            json.dump(results, f, indent=2, default=str)
+
+
+if __name__ == "__main__":
+    main()

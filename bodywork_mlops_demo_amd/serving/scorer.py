@@ -1,0 +1,81 @@
+"""Batched, hipGraph-captured inference for the scoring service.
+
+The reference scores one row per HTTP request through sklearn
+``model.predict`` (``stage_2_serve_model.py:73-80``).  Here the scorer
+keeps the model resident in HBM, scores arbitrary batches through the HIP
+kernels, and replays hipGraph-captured launches for the common batch
+sizes so the per-request launch overhead disappears from the hot loop
+(SURVEY.md §7 step 4).
+
+Graph strategy: power-of-two batch buckets, one captured graph per
+bucket, static input/output buffers; a request is padded up to its
+bucket, replayed, and sliced.  Capture is lazy (first use of a bucket)
+and falls back to direct launches on CPU or when capture is unavailable.
+"""
+from __future__ import annotations
+
+import numpy as np
+import torch
+
+from bodywork_mlops_demo_amd.utils.logging import configure_logger
+
+log = configure_logger(__name__)
+
+
+class BatchedScorer:
+    BUCKETS = [1, 16, 256, 4096, 65536, 1 << 20]
+
+    def __init__(self, model, device: str | torch.device = "cpu",
+                 use_graphs: bool = True):
+        self.model = model.to(device)
+        self.device = torch.device(device)
+        self.use_graphs = use_graphs and self.device.type == "cuda"
+        self._graphs: dict[int, tuple] = {}
+
+    def _bucket(self, n: int) -> int:
+        for b in self.BUCKETS:
+            if n <= b:
+                return b
+        return n
+
+    def _capture(self, b: int):
+        x_static = torch.zeros(b, device=self.device, dtype=torch.float32)
+        # warm up the kernels on a side stream before capture
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            for _ in range(2):
+                y_static = self.model.predict(x_static)
+        torch.cuda.current_stream().wait_stream(s)
+        graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(graph):
+            y_static = self.model.predict(x_static)
+        self._graphs[b] = (graph, x_static, y_static)
+        log.info(f"captured scoring hipGraph for batch bucket {b}")
+
+    def score_tensor(self, X: torch.Tensor) -> torch.Tensor:
+        """Score a device-resident batch; returns fp32 predictions."""
+        n = X.shape[0]
+        if not self.use_graphs:
+            return self.model.predict(X)
+        b = self._bucket(n)
+        if b not in self._graphs:
+            try:
+                self._capture(b)
+            except Exception as e:  # capture unavailable — direct launches
+                log.warning(f"hipGraph capture failed ({e}); direct launch")
+                self.use_graphs = False
+                return self.model.predict(X)
+        graph, x_static, y_static = self._graphs[b]
+        x_static[:n] = X.to(self.device, dtype=torch.float32)
+        if n < b:
+            x_static[n:] = 1.0  # benign pad (avoids log/div edge cases)
+        graph.replay()
+        return y_static[:n].clone()
+
+    def score(self, X) -> np.ndarray:
+        """Score a host array/list; returns fp32 numpy predictions."""
+        t = torch.as_tensor(np.asarray(X, dtype=np.float32).ravel(),
+                            device=self.device)
+        out = self.score_tensor(t)
+        return out.cpu().numpy()

@@ -1,0 +1,85 @@
+"""HTTP scoring service — wire-compatible with the reference.
+
+Endpoint parity (reference ``stage_2_serve_model.py:73-80`` and docstring
+``:8-22``): ``POST /score/v1`` with ``{"X": <x>}`` returns
+``{"prediction": <p>, "model_info": "<str(model)>"}``.
+
+Extensions beyond the reference (batched GPU serving):
+- ``X`` may be a list → ``prediction`` is a list (scored in one fused
+  kernel launch instead of N requests);
+- ``POST /score/v1/batch`` with ``{"X": [...]}`` returns
+  ``{"predictions": [...], "n": n, "model_info": ...}``;
+- ``GET /healthz`` for the pipeline runner's startup probe (replaces the
+  k8s readiness mechanism implied by ``bodywork.yaml:39``).
+
+Served by uvicorn (ASGI) instead of the reference's Flask dev server —
+one replica process per GPU, model resident in HBM, hipGraph-captured
+batch scoring.
+"""
+from __future__ import annotations
+
+from contextlib import asynccontextmanager
+
+import numpy as np
+from fastapi import FastAPI, Request
+from fastapi.responses import JSONResponse, Response
+
+from bodywork_mlops_demo_amd.models import regressor_from_artifact
+from bodywork_mlops_demo_amd.serving.scorer import BatchedScorer
+from bodywork_mlops_demo_amd.store import ArtefactStore
+from bodywork_mlops_demo_amd.utils.logging import configure_logger
+
+log = configure_logger(__name__)
+
+
+def create_app(store: ArtefactStore, device: str = "cpu",
+               use_graphs: bool = True) -> FastAPI:
+    state: dict = {}
+
+    @asynccontextmanager
+    async def lifespan(app: FastAPI):
+        artefact, model_date = store.get_latest_model()
+        model = regressor_from_artifact(artefact, device=device)
+        state["scorer"] = BatchedScorer(model, device, use_graphs=use_graphs)
+        state["model_info"] = str(model)
+        state["model_date"] = str(model_date)
+        log.info(f"loaded model={state['model_info']} trained on {model_date} "
+                 f"onto {device}")
+        yield
+
+    app = FastAPI(lifespan=lifespan)
+
+    @app.post("/score/v1")
+    async def score_data_instance(request: Request) -> Response:
+        payload = await request.json()
+        features = payload["X"]
+        scalar = np.isscalar(features)
+        preds = state["scorer"].score(features)
+        prediction = float(preds[0]) if scalar else [float(p) for p in preds]
+        return JSONResponse(
+            {"prediction": prediction, "model_info": state["model_info"]}
+        )
+
+    @app.post("/score/v1/batch")
+    async def score_batch(request: Request) -> Response:
+        payload = await request.json()
+        preds = state["scorer"].score(payload["X"])
+        return JSONResponse(
+            {
+                "predictions": [float(p) for p in preds],
+                "n": int(preds.shape[0]),
+                "model_info": state["model_info"],
+            }
+        )
+
+    @app.get("/healthz")
+    async def healthz() -> Response:
+        ok = "scorer" in state
+        return JSONResponse(
+            {"status": "ok" if ok else "starting",
+             "model_date": state.get("model_date"),
+             "model_info": state.get("model_info")},
+            status_code=200 if ok else 503,
+        )
+
+    return app

@@ -1,0 +1,168 @@
+"""stage-4-test-model-scoring-service: load-test the live service.
+
+Reference semantics (``stage_4_test_model_scoring_service.py:31-36``):
+download the latest (t+1) dataset, score every row against the live
+stage-2 service, time each request (3 retries, -1 sentinels on failure),
+compute live metrics (MAPE = mean |score/label - 1|, score-label
+correlation, max APE, mean response time) and persist the test-metrics
+CSV.
+
+Client modes:
+- ``serial``  — reference parity: one POST per row, new session per call
+  (``stage_4:68-85``); the reference's own throughput profile.
+- ``batch``   — the MI355X client: rows are scored in large chunks
+  against ``/score/v1/batch`` so one request drives one fused
+  (hipGraph-replayed) GPU launch; ``response_time`` is per-chunk wall
+  time, amortised per row.  This is the rows/sec headline path.
+
+Metric computation itself runs as one fused GPU reduction when a device
+is available (``ops.score_label_metrics``).
+"""
+from __future__ import annotations
+
+import argparse
+from datetime import date as date_t
+from time import perf_counter
+
+import numpy as np
+import torch
+
+from bodywork_mlops_demo_amd import ops
+from bodywork_mlops_demo_amd.monitoring import stage_guard
+from bodywork_mlops_demo_amd.store import ArtefactStore, contract, open_store
+from bodywork_mlops_demo_amd.utils.logging import configure_logger
+
+log = configure_logger(__name__)
+
+STAGE_NAME = "stage-4-test-model-scoring-service"
+DEFAULT_URL = "http://127.0.0.1:5000/score/v1"
+
+
+def _score_serial(url: str, X: np.ndarray) -> tuple[np.ndarray, np.ndarray]:
+    """Reference-parity serial client (stage_4:66-98)."""
+    import requests
+    from requests.exceptions import ConnectionError, Timeout
+
+    scores = np.empty(X.shape[0])
+    times = np.empty(X.shape[0])
+    for i, x in enumerate(X):
+        session = requests.Session()
+        session.mount(url, requests.adapters.HTTPAdapter(max_retries=3))
+        t0 = perf_counter()
+        try:
+            resp = session.post(url, json={"X": float(x)}, timeout=30)
+            times[i] = perf_counter() - t0
+            scores[i] = resp.json()["prediction"] if resp.ok else -1
+        except (ConnectionError, Timeout) as e:
+            log.error(e)
+            scores[i], times[i] = -1, -1
+    return scores, times
+
+
+def _score_batch(
+    url: str, X: np.ndarray, chunk: int = 65536
+) -> tuple[np.ndarray, np.ndarray]:
+    """Chunked batch client against /score/v1/batch."""
+    import requests
+
+    batch_url = url.rstrip("/") + "/batch" if not url.endswith("/batch") else url
+    session = requests.Session()
+    session.mount(batch_url, requests.adapters.HTTPAdapter(max_retries=3))
+    scores = np.empty(X.shape[0])
+    times = np.empty(X.shape[0])
+    for lo in range(0, X.shape[0], chunk):
+        hi = min(lo + chunk, X.shape[0])
+        t0 = perf_counter()
+        resp = session.post(
+            batch_url, json={"X": [float(v) for v in X[lo:hi]]}, timeout=300
+        )
+        dt = perf_counter() - t0
+        if resp.ok:
+            scores[lo:hi] = np.asarray(resp.json()["predictions"])
+        else:
+            scores[lo:hi] = -1
+        times[lo:hi] = dt / (hi - lo)
+    return scores, times
+
+
+def run(
+    store: ArtefactStore,
+    url: str = DEFAULT_URL,
+    mode: str = "batch",
+    device: str | None = None,
+    scorer=None,
+    data: tuple[torch.Tensor, torch.Tensor, date_t] | None = None,
+    persist: bool = True,
+) -> dict:
+    """Score the latest dataset against the service; persist test metrics.
+
+    ``scorer`` bypasses HTTP entirely (in-process serving replica) for the
+    hermetic pipeline/bench path; ``data`` skips the store read when the
+    tensors are already device-resident.
+    """
+    device = device or ("cuda" if torch.cuda.is_available() else "cpu")
+    if data is None:
+        key, data_date = store.latest(contract.DATASETS_PREFIX)
+        y_np, X_np = store.get_dataset(key)
+        labels = torch.from_numpy(y_np)
+        X = torch.from_numpy(X_np)
+    else:
+        labels, X, data_date = data
+
+    n = X.shape[0]
+    if scorer is not None:
+        t0 = perf_counter()
+        scores_t = scorer.score_tensor(X.to(device))
+        if device.startswith("cuda"):
+            torch.cuda.synchronize()
+        mean_rt = (perf_counter() - t0) / n
+        scores = scores_t
+    else:
+        X_np = X.cpu().numpy()
+        if mode == "serial":
+            scores_np, times = _score_serial(url, X_np)
+        elif mode == "batch":
+            scores_np, times = _score_batch(url, X_np)
+        else:
+            raise ValueError(f"unknown client mode {mode!r}")
+        mean_rt = float(times.mean())
+        scores = torch.from_numpy(scores_np.astype(np.float32))
+
+    scores = scores.to(device)
+    labels = labels.to(device)
+    metrics = ops.score_label_metrics(scores, labels)
+    metrics["mean_response_time"] = mean_rt
+    metrics["rows_per_sec"] = n / (mean_rt * n) if mean_rt > 0 else float("inf")
+    log.info(f"live-service metrics on {n} rows: {metrics}")
+
+    if persist:
+        key = contract.test_metrics_key(data_date)
+        store.put_metrics_csv(
+            key,
+            ["date", "MAPE", "r_squared", "max_residual", "mean_response_time"],
+            [data_date, metrics["MAPE"], metrics["r_squared"],
+             metrics["max_residual"], metrics["mean_response_time"]],
+        )
+        log.info(f"uploaded test metrics to {key}")
+    return metrics
+
+
+def main(argv=None) -> None:
+    import os
+
+    p = argparse.ArgumentParser(description=__doc__)
+    p.add_argument("--store", default=None)
+    p.add_argument(
+        "--url",
+        default=os.environ.get("BODYWORK_AMD_SERVICE_URL", DEFAULT_URL),
+    )
+    p.add_argument("--mode", default="batch", choices=["serial", "batch"])
+    p.add_argument("--device", default=None)
+    args = p.parse_args(argv)
+    with stage_guard(STAGE_NAME, exit_on_error=True):
+        run(open_store(args.store), url=args.url, mode=args.mode,
+            device=args.device)
+
+
+if __name__ == "__main__":
+    main()

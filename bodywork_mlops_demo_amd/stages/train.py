@@ -1,0 +1,119 @@
+"""stage-1-train-model: fit a regressor on all accumulated data.
+
+Reference semantics (``stage_1_train_model.py:31-36``): download every
+dataset in time order, 80/20 split with seed 42, fit, compute offline
+metrics (MAPE / R^2 / max residual), persist joblib model + metrics CSV
+keyed by the newest dataset date.
+
+MI355X path: datasets load into HBM once, the OLS fit is one fused HIP
+statistics reduction (or MFMA GEMM training steps for the MLP config),
+and in DP mode each rank holds a row shard with RCCL all-reduce of the
+statistics/gradients.
+"""
+from __future__ import annotations
+
+import argparse
+from datetime import date as date_t
+
+import torch
+
+from bodywork_mlops_demo_amd import ops
+from bodywork_mlops_demo_amd.models import GPULinearRegressor, GPUMLPRegressor
+from bodywork_mlops_demo_amd.monitoring import stage_guard
+from bodywork_mlops_demo_amd.store import ArtefactStore, contract, open_store
+from bodywork_mlops_demo_amd.utils.logging import configure_logger
+
+log = configure_logger(__name__)
+
+STAGE_NAME = "stage-1-train-model"
+
+
+def default_device() -> str:
+    return "cuda" if torch.cuda.is_available() else "cpu"
+
+
+def run(
+    store: ArtefactStore,
+    model_type: str = "linear",
+    device: str | None = None,
+    process_group=None,
+    rank: int = 0,
+    world_size: int = 1,
+    mlp_steps: int = 50,
+    mlp_batch_size: int = 65536,
+    data: tuple[torch.Tensor, torch.Tensor, date_t] | None = None,
+    return_model: bool = False,
+):
+    """Train and persist; returns the offline metrics record
+    (or ``(metrics, model)`` when ``return_model``).
+
+    ``data`` short-circuits store reads when the caller already holds the
+    tensors in HBM (the in-process pipeline keeps data resident instead of
+    re-reading artefacts — SURVEY.md §7 'keep tensors resident').
+    """
+    device = device or default_device()
+    if data is None:
+        y_np, X_np, data_date = store.get_all_datasets()
+        X = torch.from_numpy(X_np).to(device)
+        y = torch.from_numpy(y_np).to(device)
+    else:
+        y, X, data_date = data
+        X, y = X.to(device), y.to(device)
+
+    if world_size > 1:  # row-shard for DP
+        X, y = X[rank::world_size], y[rank::world_size]
+
+    n = X.shape[0]
+    log.info(f"training {model_type} regressor on {n} rows (device={device})")
+    train_idx, test_idx = ops.train_test_split_indices(
+        n, test_size=0.2, seed=42, device=device
+    )
+    X_train, y_train = X[train_idx], y[train_idx]
+    X_test, y_test = X[test_idx], y[test_idx]
+
+    if model_type == "linear":
+        model = GPULinearRegressor(device=device).fit(
+            X_train, y_train, process_group=process_group
+        )
+    elif model_type == "mlp":
+        model = GPUMLPRegressor(device=device).fit(
+            X_train, y_train, steps=mlp_steps, batch_size=mlp_batch_size,
+            process_group=process_group,
+        )
+    else:
+        raise ValueError(f"unknown model_type {model_type!r}")
+
+    yhat = model.predict(X_test)
+    metrics = ops.regression_metrics(y_test, yhat)
+    log.info(f"offline metrics: {metrics}")
+
+    if rank == 0:
+        model_key = store.put_model(model.to_sklearn(), data_date)
+        log.info(f"uploaded model to {model_key}")
+        metrics_key = contract.model_metrics_key(data_date)
+        store.put_metrics_csv(
+            metrics_key,
+            ["date", "MAPE", "r_squared", "max_residual"],
+            [data_date, metrics["MAPE"], metrics["r_squared"],
+             metrics["max_residual"]],
+        )
+        log.info(f"uploaded metrics to {metrics_key}")
+    if return_model:
+        return metrics, model
+    return metrics
+
+
+def main(argv=None) -> None:
+    p = argparse.ArgumentParser(description=__doc__)
+    p.add_argument("--store", default=None, help="store URI (dir or s3://bucket)")
+    p.add_argument("--model", default="linear", choices=["linear", "mlp"])
+    p.add_argument("--device", default=None)
+    p.add_argument("--mlp-steps", type=int, default=50)
+    args = p.parse_args(argv)
+    with stage_guard(STAGE_NAME, exit_on_error=True):
+        run(open_store(args.store), model_type=args.model, device=args.device,
+            mlp_steps=args.mlp_steps)
+
+
+if __name__ == "__main__":
+    main()

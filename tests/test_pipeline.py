@@ -1,0 +1,120 @@
+"""Pipeline runner + cycle + drift-loop + analytics tests (CPU)."""
+import sys
+from datetime import date
+
+import pytest
+
+from bodywork_mlops_demo_amd.config import load_config
+from bodywork_mlops_demo_amd.pipeline.cycle import CycleState, run_cycle
+from bodywork_mlops_demo_amd.pipeline.loop import run_loop
+from bodywork_mlops_demo_amd.pipeline.runner import PipelineRunner
+from bodywork_mlops_demo_amd.stages import datagen
+from bodywork_mlops_demo_amd.store import contract
+
+
+def test_cycle_semantics(tmp_store):
+    """train(t) -> deploy(t) -> generate(t+1) -> test model(t) on data(t+1)."""
+    datagen.run(tmp_store, n=800, date=date(2026, 1, 1), device="cpu")
+    state = CycleState("cpu", date(2026, 1, 1))
+    import numpy as np
+    import torch
+
+    y, X = tmp_store.get_dataset(contract.dataset_key(date(2026, 1, 1)))
+    state.append_day(torch.from_numpy(y), torch.from_numpy(X))
+
+    r = run_cycle(state, tmp_store, n_rows=800, model_type="linear",
+                  persist_fmt="csv")
+    # model trained on day 1, tested on day 2's data
+    assert tmp_store.exists(contract.model_key(date(2026, 1, 1)))
+    assert tmp_store.exists(contract.dataset_key(date(2026, 1, 2)))
+    assert tmp_store.exists(contract.test_metrics_key(date(2026, 1, 2)))
+    assert state.date == date(2026, 1, 2)
+    t = r["timings"]
+    assert t["cycle_s"] >= t["train_s"] + t["test_s"]
+    assert r["online"]["r_squared"] > 0.5
+    assert np.isfinite(r["offline"]["MAPE"])
+
+
+def test_drift_loop_and_analytics(tmp_store):
+    results = run_loop(tmp_store, days=3, n_rows=600, model_type="linear",
+                       device="cpu", start_date="2026-03-01")
+    assert len(results) == 3
+    # 1 bootstrap + 3 generated days of datasets
+    assert len(tmp_store.list_keys(contract.DATASETS_PREFIX)) == 4
+    assert len(tmp_store.list_keys(contract.MODEL_METRICS_PREFIX)) == 3
+    assert len(tmp_store.list_keys(contract.TEST_METRICS_PREFIX)) == 3
+
+    from bodywork_mlops_demo_amd.monitoring.analytics import drift_report
+
+    report = drift_report(tmp_store)
+    assert report["summary"]["days"] >= 2
+    assert report["summary"]["mean_online_MAPE"] > 0
+
+    # resume: next loop continues from the latest store date
+    more = run_loop(tmp_store, days=1, n_rows=600, device="cpu",
+                    start_date="2026-03-01")
+    keys = tmp_store.list_keys(contract.DATASETS_PREFIX)
+    assert contract.dataset_key(date(2026, 3, 5)) in keys
+
+
+SMALL_PIPELINE = """
+version: "1.0"
+project:
+  name: test-pipeline
+  DAG: stage-3-generate-next-dataset >> stage-1-train-model >> stage-2-serve-model >> stage-4-test-model-scoring-service
+stages:
+  stage-1-train-model:
+    executable_module_path: bodywork_mlops_demo_amd/stages/train.py
+    batch: {max_completion_time_seconds: 120, retries: 1}
+  stage-2-serve-model:
+    executable_module_path: bodywork_mlops_demo_amd/stages/serve.py
+    service: {max_startup_time_seconds: 60, replicas: 1, port: 5391}
+  stage-3-generate-next-dataset:
+    executable_module_path: bodywork_mlops_demo_amd/stages/datagen.py
+    args: ["--n", "200"]
+    batch: {max_completion_time_seconds: 120, retries: 1}
+  stage-4-test-model-scoring-service:
+    executable_module_path: bodywork_mlops_demo_amd/stages/loadtest.py
+    args: ["--mode", "batch"]
+    batch: {max_completion_time_seconds: 300, retries: 1}
+logging:
+  log_level: INFO
+"""
+
+
+@pytest.mark.timeout(300)
+def test_runner_end_to_end_subprocesses(tmp_path):
+    """The full DAG through real subprocess stages + a live HTTP service."""
+    store_dir = str(tmp_path / "store")
+    cfg = load_config(SMALL_PIPELINE)
+    runner = PipelineRunner(cfg, store_uri=store_dir, base_port=5391, n_gpus=0)
+    report = runner.run()
+    assert report.ok, f"failed stages: {report.failed}"
+    assert len(report.succeeded) == 4
+
+    from bodywork_mlops_demo_amd.store import LocalStore
+
+    store = LocalStore(store_dir)
+    assert len(store.list_keys(contract.DATASETS_PREFIX)) == 1
+    assert len(store.list_keys(contract.MODELS_PREFIX)) == 1
+    assert len(store.list_keys(contract.TEST_METRICS_PREFIX)) == 1
+    # services torn down
+    assert not runner.services
+
+
+def test_runner_retries_failing_stage(tmp_path):
+    cfg = load_config("""
+version: "1.0"
+project:
+  name: fail-pipeline
+  DAG: boom
+stages:
+  boom:
+    executable_module_path: bodywork_mlops_demo_amd/stages/train.py
+    batch: {max_completion_time_seconds: 60, retries: 1}
+""")
+    # train fails: empty store -> no datasets
+    runner = PipelineRunner(cfg, store_uri=str(tmp_path / "empty"), n_gpus=0)
+    report = runner.run()
+    assert not report.ok
+    assert report.attempts["boom"] == 2  # initial + 1 retry
