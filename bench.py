@@ -1,0 +1,183 @@
+"""Flagship benchmark — the driver contract.
+
+Measures BASELINE.json's headline metric on MI355X: scoring rows/sec
+through the stage-2 path with the full train->serve->drift->test cycle as
+the step (`ms_per_step` = cycle wall-clock).  One step = one pipeline
+"day" on `--rows` synthetic rows per GPU (weak scaling):
+
+  stage 1  train on the day's rows (fused-stats OLS or MFMA-GEMM MLP,
+           DP all-reduce over RCCL when N>1)
+  stage 2  deploy: joblib artefact round-trip -> model into HBM,
+           hipGraph-captured batch scoring
+  stage 3  generate day t+1 on-GPU (philox drift kernel + compaction)
+  stage 4  score day t+1 through the deployed scorer + fused metrics,
+           persist test-metrics artefact
+
+Launch (driver): python -m torch.distributed.run --nnodes=1
+  --nproc-per-node N --master-addr 127.0.0.1 bench.py --gpus N ...
+Rank 0 prints ONE JSON line; `value` is whole-job rows/sec scored.
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import shutil
+import tempfile
+from datetime import date
+from time import perf_counter
+
+import torch
+
+
+def main() -> None:
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=4)
+    p.add_argument("--warmup", type=int, default=2)
+    p.add_argument("--rows", type=int, default=10_000_000,
+                   help="synthetic rows per GPU per day (BASELINE config 2)")
+    p.add_argument("--model", default="linear", choices=["linear", "mlp"],
+                   help="mlp = BASELINE config 5 (4096-d MFMA GEMM path)")
+    p.add_argument("--mlp-steps", type=int, default=50)
+    p.add_argument("--mlp-batch", type=int, default=65536)
+    p.add_argument("--no-graphs", action="store_true")
+    p.add_argument("--store", default=None,
+                   help="artefact store dir (default: fresh tmpdir)")
+    args = p.parse_args()
+
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    if world > 1 and args.gpus != world:
+        args.gpus = world
+
+    use_cuda = torch.cuda.is_available()
+    device = f"cuda:{local_rank}" if use_cuda else "cpu"
+
+    if use_cuda:
+        # the HIP extension is the compute path — fail loudly if missing
+        from bodywork_mlops_demo_amd import ops
+
+        if not ops.hip_available():
+            raise RuntimeError(
+                "bench requires the in-tree _hipcore HIP extension on GPU; "
+                "build with: PYTORCH_ROCM_ARCH=gfx950 python setup.py "
+                "build_ext --inplace"
+            )
+        torch.cuda.set_device(local_rank)
+
+    pg = None
+    if world > 1:
+        import torch.distributed as dist
+
+        from bodywork_mlops_demo_amd.parallel import init_distributed
+
+        init_distributed()
+        pg = dist.group.WORLD
+
+    from bodywork_mlops_demo_amd.pipeline.cycle import CycleState, run_cycle
+    from bodywork_mlops_demo_amd.store import LocalStore
+
+    # shared store dir across ranks
+    if args.store:
+        store_dir = args.store
+    else:
+        store_dir = os.path.join(
+            tempfile.gettempdir(),
+            f"bodywork_bench_{os.environ.get('MASTER_PORT', 'solo')}",
+        )
+    if rank == 0:
+        shutil.rmtree(store_dir, ignore_errors=True)
+        os.makedirs(store_dir, exist_ok=True)
+    if pg is not None:
+        torch.distributed.barrier(pg)
+    store = LocalStore(store_dir)
+
+    state = CycleState(device, date(2026, 1, 1), rank=rank, world_size=world,
+                       history_days=1)
+
+    def one_cycle():
+        return run_cycle(
+            state, store, args.rows, model_type=args.model,
+            process_group=pg, persist_fmt="npz",
+            mlp_steps=args.mlp_steps, mlp_batch_size=args.mlp_batch,
+            use_graphs=not args.no_graphs and use_cuda,
+        )
+
+    def barrier_sync():
+        if pg is not None:
+            torch.distributed.barrier(pg)
+        if use_cuda:
+            torch.cuda.synchronize()
+
+    for _ in range(args.warmup):
+        one_cycle()
+
+    barrier_sync()
+    t0 = perf_counter()
+    rows_scored = 0
+    last = None
+    for _ in range(args.steps):
+        last = one_cycle()
+        rows_scored += last["timings"]["rows_scored"]
+    barrier_sync()
+    elapsed = perf_counter() - t0
+
+    # max elapsed over ranks, total rows over ranks
+    if pg is not None:
+        buf = torch.tensor([elapsed], dtype=torch.float64,
+                           device=device if use_cuda else "cpu")
+        torch.distributed.all_reduce(buf, op=torch.distributed.ReduceOp.MAX,
+                                     group=pg)
+        elapsed = float(buf.item())
+        rbuf = torch.tensor([float(rows_scored)], dtype=torch.float64,
+                            device=device if use_cuda else "cpu")
+        torch.distributed.all_reduce(rbuf, group=pg)
+        rows_scored = int(rbuf.item())
+
+    if rank == 0:
+        ms_per_step = elapsed / args.steps * 1000.0
+        value = rows_scored / elapsed
+        print(json.dumps({
+            "metric": "rows/sec scored (stage_2 path, full "
+                      "train-serve-drift-test cycle)",
+            "value": value,
+            "unit": "rows/s",
+            "n_gpus": args.gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": ms_per_step,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16" if args.model == "mlp" else "fp32",
+            "data": "synthetic (on-GPU philox drift generator, "
+                    f"{args.rows} rows/GPU/day, random-init weights)",
+            "config": {
+                "model": ("mlp-4096x2" if args.model == "mlp"
+                          else "linear-ols"),
+                "rows_per_gpu_per_day": args.rows,
+                "history_days": 1,
+                "parallelism": f"dp{args.gpus}",
+                "serving": "1 replica/GPU, hipGraph-captured batches",
+                "train_phase": ("adam bf16 MFMA GEMM, "
+                                f"{args.mlp_steps}x{args.mlp_batch}"
+                                if args.model == "mlp"
+                                else "fused-stats closed-form OLS"),
+                "online_mape": (last["online"]["MAPE"] if last else None),
+                "phase_ms": {
+                    k: round(v * 1000.0, 3)
+                    for k, v in (last["timings"].items() if last else [])
+                    if k.endswith("_s")
+                },
+            },
+        }))
+
+    if pg is not None:
+        torch.distributed.barrier(pg)
+        torch.distributed.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

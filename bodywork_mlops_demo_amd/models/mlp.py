@@ -54,10 +54,13 @@ class GPUMLPRegressor:
         self._refresh_bf16()
 
     def _refresh_bf16(self):
+        # W2 master is [in,out] (sklearn coefs_[1] layout); the MFMA linear
+        # kernel wants K-contiguous weights both ways, so keep two bf16
+        # copies: [out,in] for forward, [in,out] for backward-data.
         self.w1_bf = self.w1.bfloat16()
         self.b1_bf = self.b1.bfloat16()
-        self.W2_bf = self.W2.bfloat16()
-        self.W2t_bf = self.W2.t().contiguous().bfloat16()  # for NT backward
+        self.W2w_bf = self.W2.t().contiguous().bfloat16()   # [out,in]
+        self.W2wt_bf = self.W2.contiguous().bfloat16()      # [in,out]
         self.b2_bf = self.b2.bfloat16()
         self.w3_bf = self.w3.bfloat16()
 
@@ -72,7 +75,7 @@ class GPUMLPRegressor:
     # -- forward -----------------------------------------------------------
     def _forward(self, x: torch.Tensor):
         h1 = ops.expand1d_bf16(x, self.w1_bf, self.b1_bf, relu=True)
-        h2 = ops.gemm_bf16(h1, self.W2_bf, bias=self.b2_bf, relu=True)
+        h2 = ops.linear_bf16(h1, self.W2w_bf, bias=self.b2_bf, relu=True)
         yhat = ops.rowdot_bf16(h2, self.w3_bf, float(self.b3.item()))
         return yhat, h1, h2
 
@@ -135,10 +138,10 @@ class GPUMLPRegressor:
         db3 = dy.sum().reshape(1)
         # dh2 = outer(dy, w3) * relu'(h2): fused expand with mask
         dz2 = ops.expand1d_bf16(dy, self.w3_bf, None, relu=False, mask=h2)
-        # layer 2: h2 = relu(h1 @ W2 + b2)
-        dW2 = ops.gemm_bf16(h1, dz2, trans_a=True, out_fp32=True)   # (H,H)
+        # layer 2: h2 = relu(h1 @ W2 + b2), W2 [in,out]
+        dW2 = ops.gemm_tn_bf16(h1, dz2, out_fp32=True)          # (in,out)
         db2 = ops.colsum_bf16(dz2)
-        dz1 = ops.gemm_bf16(dz2, self.W2t_bf, mask=h1)          # (n,H)*relu'(h1)
+        dz1 = ops.linear_bf16(dz2, self.W2wt_bf, mask=h1)       # dz2 @ W2^T
         # layer 1: h1 = relu(x w1 + b1)
         dw1, db1 = ops.coldot_bf16(dz1, xb.float(), also_colsum=True)
         return [dw1, db1, dW2, db2, dw3, db3]

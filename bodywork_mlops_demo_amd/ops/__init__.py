@@ -222,33 +222,45 @@ def train_test_split_indices(
 # MFMA GEMM — MLP path
 # --------------------------------------------------------------------------
 
-def gemm_bf16(
-    a: torch.Tensor,
-    b: torch.Tensor,
+def linear_bf16(
+    x: torch.Tensor,
+    w: torch.Tensor,
     bias: torch.Tensor | None = None,
     relu: bool = False,
-    trans_a: bool = False,
     mask: torch.Tensor | None = None,
     out_fp32: bool = False,
 ) -> torch.Tensor:
-    """C = op(A) @ B with a fused epilogue, bf16 in / fp32 accumulate.
+    """C[M,N] = x[M,K] @ w[N,K]^T with a fused epilogue — bf16 MFMA GEMM.
 
-    ``op(A) = A.T`` when ``trans_a`` (the dW = X^T @ dY backward shape).
-    Epilogue (fused into the C-write, never a second pass over HBM):
-    ``+bias`` then ``relu`` , or ``* (mask > 0)`` (ReLU backward).
-    GPU: hand-written gfx950 MFMA kernel (v_mfma_f32_16x16x32_bf16,
-    LDS-tiled).  CPU oracle: fp32 torch matmul.
+    torch.nn.functional.linear convention: both operands K-contiguous in
+    memory, so LDS staging is coalesced and fragment reads are single
+    ds_read_b128s (ops/hip/gemm.hip).  Epilogue (fused into the C-write):
+    ``+bias`` then ``relu``, or ``* (mask > 0)`` (ReLU backward).
+    CPU oracle: fp32 torch matmul.
     """
-    if a.device.type == "cuda":
-        core = _core(a.device)
-        return core.gemm_bf16(
-            a.contiguous(), b.contiguous(),
+    if x.device.type == "cuda":
+        core = _core(x.device)
+        return core.linear_bf16(
+            x.contiguous(), w.contiguous(),
             bias.contiguous() if bias is not None else None,
-            relu, trans_a,
+            relu,
             mask.contiguous() if mask is not None else None,
             out_fp32,
         )
-    return reference.gemm_bf16_cpu(a, b, bias, relu, trans_a, mask, out_fp32)
+    return reference.linear_bf16_cpu(x, w, bias, relu, mask, out_fp32)
+
+
+def gemm_tn_bf16(
+    a: torch.Tensor, b: torch.Tensor, out_fp32: bool = False
+) -> torch.Tensor:
+    """C[M,N] = a[R,M]^T @ b[R,N] — the dW = dY^T @ X backward shape.
+
+    Same MFMA kernel with transpose-staged LDS tiles.
+    """
+    if a.device.type == "cuda":
+        core = _core(a.device)
+        return core.gemm_tn_bf16(a.contiguous(), b.contiguous(), out_fp32)
+    return reference.gemm_tn_bf16_cpu(a, b, out_fp32)
 
 
 def expand1d_bf16(

@@ -1,0 +1,75 @@
+// torch library registration for the gfx950 HIP kernels.
+//
+// Registered under torch.ops.bodywork_hip.* so the ops are first-class
+// torch ops (dispatcher + hipGraph capture compatible).
+#include <torch/extension.h>
+
+// datagen.hip
+std::tuple<at::Tensor, at::Tensor> datagen_hip(int64_t n, int64_t seed,
+                                               int64_t stream_offset,
+                                               double alpha, double beta,
+                                               double sigma);
+// linreg.hip
+at::Tensor linreg_stats_hip(const at::Tensor& x, const at::Tensor& y);
+at::Tensor linear_score_hip(const at::Tensor& x, double intercept,
+                            double coef);
+at::Tensor regression_metrics_hip(const at::Tensor& y, const at::Tensor& yhat);
+at::Tensor score_label_metrics_hip(const at::Tensor& s, const at::Tensor& l);
+// mlp_small.hip
+at::Tensor expand1d_bf16_hip(const at::Tensor& x, const at::Tensor& w,
+                             const c10::optional<at::Tensor>& b, bool relu,
+                             const c10::optional<at::Tensor>& mask);
+at::Tensor rowdot_bf16_hip(const at::Tensor& h, const at::Tensor& w,
+                           double bias);
+at::Tensor coldot_bf16_hip(const at::Tensor& m, const at::Tensor& v,
+                           bool also_colsum);
+at::Tensor colsum_bf16_hip(const at::Tensor& m);
+// gemm.hip
+at::Tensor linear_bf16_hip(const at::Tensor& x, const at::Tensor& w,
+                           const c10::optional<at::Tensor>& bias, bool relu,
+                           const c10::optional<at::Tensor>& mask,
+                           bool out_fp32);
+at::Tensor gemm_tn_bf16_hip(const at::Tensor& a, const at::Tensor& b,
+                            bool out_fp32);
+
+TORCH_LIBRARY(bodywork_hip, m) {
+  m.def("datagen(int n, int seed, int stream_offset, float alpha, float beta, "
+        "float sigma) -> (Tensor, Tensor)");
+  m.def("linreg_stats(Tensor x, Tensor y) -> Tensor");
+  m.def("linear_score(Tensor x, float intercept, float coef) -> Tensor");
+  m.def("regression_metrics(Tensor y, Tensor yhat) -> Tensor");
+  m.def("score_label_metrics(Tensor s, Tensor l) -> Tensor");
+  m.def("expand1d_bf16(Tensor x, Tensor w, Tensor? b, bool relu, "
+        "Tensor? mask) -> Tensor");
+  m.def("rowdot_bf16(Tensor h, Tensor w, float bias) -> Tensor");
+  m.def("coldot_bf16(Tensor m, Tensor v, bool also_colsum) -> Tensor");
+  m.def("colsum_bf16(Tensor m) -> Tensor");
+  m.def("linear_bf16(Tensor x, Tensor w, Tensor? bias, bool relu, "
+        "Tensor? mask, bool out_fp32) -> Tensor");
+  m.def("gemm_tn_bf16(Tensor a, Tensor b, bool out_fp32) -> Tensor");
+}
+
+TORCH_LIBRARY_IMPL(bodywork_hip, CUDA, m) {
+  m.impl("datagen", datagen_hip);
+  m.impl("linreg_stats", linreg_stats_hip);
+  m.impl("linear_score", linear_score_hip);
+  m.impl("regression_metrics", regression_metrics_hip);
+  m.impl("score_label_metrics", score_label_metrics_hip);
+  m.impl("expand1d_bf16", expand1d_bf16_hip);
+  m.impl("rowdot_bf16", rowdot_bf16_hip);
+  m.impl("coldot_bf16", coldot_bf16_hip);
+  m.impl("colsum_bf16", colsum_bf16_hip);
+  m.impl("linear_bf16", linear_bf16_hip);
+  m.impl("gemm_tn_bf16", gemm_tn_bf16_hip);
+}
+
+// datagen takes no tensor argument, so the dispatcher cannot route it by
+// device -- register a CompositeExplicitAutograd fallback that forwards
+// to the CUDA implementation (it allocates on the current CUDA device).
+TORCH_LIBRARY_IMPL(bodywork_hip, CompositeExplicitAutograd, m) {
+  m.impl("datagen", datagen_hip);
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.doc() = "bodywork_mlops_demo_amd gfx950 HIP kernels";
+}

@@ -1,0 +1,274 @@
+// Hand-written gfx950 MFMA bf16 dense GEMM — the MLP hot path.
+//
+// Replaces the sklearn/library matmuls of the reference's fit/predict
+// (stage_1:105-106, stage_2:78) for the MLP config (BASELINE config 5)
+// with v_mfma_f32_16x16x32_bf16 tiles, LDS-staged, fp32 accumulate.
+//
+// Two kernels, chosen so every global load is K-contiguous and every LDS
+// fragment read is a single ds_read_b128 (CDNA4 fragment layout: a lane
+// holds 8 contiguous K elements):
+//   linear_nt:  C[M,N] = X[M,K] @ W[N,K]^T (+bias) (+ReLU | *mask) —
+//               torch.nn.functional.linear convention; both operands
+//               K-major so LDS staging is coalesced and linear.
+//   gemm_tn:    C[M,N] = A[R,M]^T @ B[R,N] — the dW = dY^T @ X backward
+//               shape; operands are R-major so tiles are transpose-staged
+//               into the same K-contiguous LDS images.
+//
+// Geometry: 128x128 C-tile per 256-thread block (4 waves as 2x2, each
+// wave a 64x64 sub-tile = 4x4 fragments of 16x16), BK=64, LDS rows padded
+// to 72 shorts (row stride 144 B = 36 banks: a 16-lane fragment-read group
+// touches 16 distinct banks - conflict-free without an XOR swizzle).
+// fp32 C/D per guide §3: col = lane&15, row = (lane>>4)*4 + reg.
+#include <ATen/cuda/CUDAContext.h>
+#include <hip/hip_runtime.h>
+#include <torch/extension.h>
+
+#include "bf16_utils.h"
+
+#define BM 128
+#define BN 128
+#define BK 64
+#define LDS_STRIDE 72  // BK + 8 shorts pad
+#define GEMM_THREADS 256
+
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8_v;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+
+union lds_vec {
+  bf16x8_v v;
+  short s[8];
+};
+
+// stage a [ROWS x BK] K-major tile (row-major source, K contiguous):
+// thread t loads 16 B chunks; zero-fill outside (r1, k1) bounds.
+__device__ __forceinline__ void stage_k_major(
+    short* __restrict__ lds, const bf16_t* __restrict__ src, long long ld,
+    long long row0, long long row_end, long long k0, long long k_end) {
+#pragma unroll
+  for (int it = 0; it < (BM * (BK / 8)) / GEMM_THREADS; ++it) {
+    int chunk = threadIdx.x + it * GEMM_THREADS;
+    int r = chunk >> 3;          // row within tile
+    int c8 = chunk & 7;          // 8-col chunk within BK
+    long long gr = row0 + r;
+    long long gk = k0 + c8 * 8;
+    lds_vec val;
+    if (gr < row_end && gk + 7 < k_end) {
+      val = *(const lds_vec*)(src + gr * ld + gk);
+    } else {
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        long long kk = gk + e;
+        val.s[e] = (gr < row_end && kk < k_end)
+                       ? ((const short*)(src + gr * ld))[kk]
+                       : (short)0;
+      }
+    }
+    *(lds_vec*)(lds + r * LDS_STRIDE + c8 * 8) = val;
+  }
+}
+
+// stage a [ROWS x BK] K-major LDS tile from an R-major source
+// (src[R, C] with R the GEMM K dim): transpose during the LDS write.
+__device__ __forceinline__ void stage_transposed(
+    short* __restrict__ lds, const bf16_t* __restrict__ src, long long ld,
+    long long col0, long long col_end /* tile rows = source cols */,
+    long long k0, long long k_end /* tile cols = source rows */) {
+#pragma unroll
+  for (int it = 0; it < (BM * (BK / 8)) / GEMM_THREADS; ++it) {
+    int chunk = threadIdx.x + it * GEMM_THREADS;
+    int kr = chunk >> 4;         // source row (= tile k) [0,64)
+    int c8 = chunk & 15;         // 8-col chunk along source cols [0,16)
+    long long gk = k0 + kr;
+    long long gc = col0 + c8 * 8;
+    lds_vec val;
+    if (gk < k_end && gc + 7 < col_end) {
+      val = *(const lds_vec*)(src + gk * ld + gc);
+    } else {
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        long long cc = gc + e;
+        val.s[e] = (gk < k_end && cc < col_end)
+                       ? ((const short*)(src + gk * ld))[cc]
+                       : (short)0;
+      }
+    }
+#pragma unroll
+    for (int e = 0; e < 8; ++e)
+      lds[(c8 * 8 + e) * LDS_STRIDE + kr] = val.s[e];
+  }
+}
+
+// epilogue modes
+#define EPI_NONE 0
+#define EPI_BIAS_RELU 1  // +bias then relu (bias may be null -> relu only)
+#define EPI_MASK 2       // multiply by (mask > 0)
+
+template <bool TN, int EPI, bool HAS_BIAS, bool OUT_FP32>
+__launch_bounds__(GEMM_THREADS)
+__global__ void gemm_bf16_kernel(
+    const bf16_t* __restrict__ A,  // NT: [M,K]; TN: [R=K, M]
+    const bf16_t* __restrict__ B,  // NT: [N,K]; TN: [R=K, N]
+    const float* __restrict__ bias, const bf16_t* __restrict__ mask,
+    void* __restrict__ C, long long M, long long N, long long K) {
+  __shared__ short lds_all[2 * BM * LDS_STRIDE];
+  short* As = lds_all;
+  short* Bs = lds_all + BM * LDS_STRIDE;
+
+  const long long m0 = (long long)blockIdx.y * BM;
+  const long long n0 = (long long)blockIdx.x * BN;
+  const int wave = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int wm = wave >> 1;  // wave row 0..1
+  const int wn = wave & 1;   // wave col 0..1
+  const int fl = lane & 15;       // fragment row/col within 16
+  const int kg = lane >> 4;       // k-group 0..3 (8 elems each)
+
+  f32x4 acc[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  for (long long k0 = 0; k0 < K; k0 += BK) {
+    if (TN) {
+      stage_transposed(As, A, M, m0, M, k0, K);
+      stage_transposed(Bs, B, N, n0, N, k0, K);
+    } else {
+      stage_k_major(As, A, K, m0, M, k0, K);
+      stage_k_major(Bs, B, K, n0, N, k0, K);
+    }
+    __syncthreads();
+#pragma unroll
+    for (int ks = 0; ks < BK / 32; ++ks) {
+      bf16x8_v a_frag[4], b_frag[4];
+#pragma unroll
+      for (int f = 0; f < 4; ++f) {
+        a_frag[f] = ((const lds_vec*)(As + (wm * 64 + f * 16 + fl) * LDS_STRIDE +
+                                      ks * 32 + kg * 8))->v;
+        b_frag[f] = ((const lds_vec*)(Bs + (wn * 64 + f * 16 + fl) * LDS_STRIDE +
+                                      ks * 32 + kg * 8))->v;
+      }
+#pragma unroll
+      for (int i = 0; i < 4; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a_frag[i], b_frag[j], acc[i][j], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  // epilogue: C[row][col], row = tile_m + wm*64 + i*16 + kg*4 + r,
+  //           col = tile_n + wn*64 + j*16 + fl
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      long long col = n0 + wn * 64 + j * 16 + fl;
+      if (col >= N) continue;
+      float bval = (EPI == EPI_BIAS_RELU && HAS_BIAS) ? bias[col] : 0.0f;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        long long row = m0 + wm * 64 + i * 16 + kg * 4 + r;
+        if (row >= M) continue;
+        float v = acc[i][j][r];
+        if (EPI == EPI_BIAS_RELU) {
+          v += bval;
+          v = fmaxf(v, 0.0f);
+        } else if (EPI == EPI_MASK) {
+          float mv = bf16_to_f32(mask[row * N + col]);
+          v = mv > 0.0f ? v : 0.0f;
+        }
+        if (OUT_FP32)
+          ((float*)C)[row * N + col] = v;
+        else
+          ((bf16_t*)C)[row * N + col] = f32_to_bf16(v);
+      }
+    }
+  }
+}
+
+// ---- launchers ------------------------------------------------------------
+
+static void launch_gemm(bool tn, int epi, bool has_bias, bool out_fp32,
+                        const at::Tensor& a, const at::Tensor& b,
+                        const float* bias, const bf16_t* mask, at::Tensor& c,
+                        long long M, long long N, long long K) {
+  dim3 grid((N + BN - 1) / BN, (M + BM - 1) / BM);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  const bf16_t* ap = (const bf16_t*)a.data_ptr();
+  const bf16_t* bp = (const bf16_t*)b.data_ptr();
+  void* cp = c.data_ptr();
+
+#define G_LAUNCH(TN_, EPI_, HB_, OF_)                                       \
+  hipLaunchKernelGGL((gemm_bf16_kernel<TN_, EPI_, HB_, OF_>), grid,         \
+                     dim3(GEMM_THREADS), 0, stream, ap, bp, bias, mask, cp, \
+                     M, N, K)
+#define G_EPI(TN_)                                                          \
+  do {                                                                      \
+    if (epi == EPI_BIAS_RELU) {                                             \
+      if (has_bias) { if (out_fp32) G_LAUNCH(TN_, EPI_BIAS_RELU, true, true);   \
+                      else          G_LAUNCH(TN_, EPI_BIAS_RELU, true, false); }\
+      else          { if (out_fp32) G_LAUNCH(TN_, EPI_BIAS_RELU, false, true);  \
+                      else          G_LAUNCH(TN_, EPI_BIAS_RELU, false, false);}\
+    } else if (epi == EPI_MASK) {                                           \
+      if (out_fp32) G_LAUNCH(TN_, EPI_MASK, false, true);                   \
+      else          G_LAUNCH(TN_, EPI_MASK, false, false);                  \
+    } else {                                                                \
+      if (out_fp32) G_LAUNCH(TN_, EPI_NONE, false, true);                   \
+      else          G_LAUNCH(TN_, EPI_NONE, false, false);                  \
+    }                                                                       \
+  } while (0)
+  if (tn) G_EPI(true); else G_EPI(false);
+#undef G_EPI
+#undef G_LAUNCH
+}
+
+// C = x @ w^T (+bias)(+relu | *mask); x [M,K], w [N,K]
+at::Tensor linear_bf16_hip(const at::Tensor& x, const at::Tensor& w,
+                           const c10::optional<at::Tensor>& bias, bool relu,
+                           const c10::optional<at::Tensor>& mask,
+                           bool out_fp32) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 2 && w.dim() == 2);
+  TORCH_CHECK(x.scalar_type() == at::kBFloat16 &&
+              w.scalar_type() == at::kBFloat16);
+  long long M = x.size(0), K = x.size(1), N = w.size(0);
+  TORCH_CHECK(w.size(1) == K, "linear_bf16: w must be [N,K]");
+  TORCH_CHECK(!(relu && mask.has_value()), "relu and mask are exclusive");
+  auto c = at::empty({M, N},
+                     x.options().dtype(out_fp32 ? at::kFloat : at::kBFloat16));
+  at::Tensor bias_f;
+  const float* bp = nullptr;
+  if (bias.has_value()) {
+    bias_f = bias->to(at::kFloat);
+    bp = bias_f.data_ptr<float>();
+  }
+  int epi = mask.has_value() ? EPI_MASK
+            : (relu || bias.has_value()) ? EPI_BIAS_RELU
+                                         : EPI_NONE;
+  // bias without relu: fold via EPI_BIAS_RELU only when relu requested;
+  // else add bias in the same kernel without clamping -> use NONE + host
+  // add would cost a pass, so clamp-free bias is handled here:
+  TORCH_CHECK(!(bias.has_value() && !relu),
+              "linear_bf16: bias currently requires relu epilogue");
+  const bf16_t* mp =
+      mask.has_value() ? (const bf16_t*)mask->data_ptr() : nullptr;
+  launch_gemm(false, epi, bias.has_value(), out_fp32, x, w, bp, mp, c, M, N,
+              K);
+  return c;
+}
+
+// C = a^T @ b; a [R,M], b [R,N] (the dW backward shape)
+at::Tensor gemm_tn_bf16_hip(const at::Tensor& a, const at::Tensor& b,
+                            bool out_fp32) {
+  TORCH_CHECK(a.is_cuda() && a.dim() == 2 && b.dim() == 2);
+  TORCH_CHECK(a.scalar_type() == at::kBFloat16 &&
+              b.scalar_type() == at::kBFloat16);
+  long long R = a.size(0), M = a.size(1), N = b.size(1);
+  TORCH_CHECK(b.size(0) == R, "gemm_tn: row counts must match");
+  auto c = at::empty({M, N},
+                     a.options().dtype(out_fp32 ? at::kFloat : at::kBFloat16));
+  launch_gemm(true, EPI_NONE, false, out_fp32, a, b, nullptr, nullptr, c, M,
+              N, R);
+  return c;
+}

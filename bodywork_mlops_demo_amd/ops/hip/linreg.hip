@@ -1,0 +1,212 @@
+// Fused OLS statistics, batched scoring and metric reductions — gfx950.
+//
+// Replaces sklearn LinearRegression.fit / predict and the sklearn metric
+// calls (reference stage_1:79-108, stage_2:78, stage_4:101-113) with:
+//   - linreg_stats:   single pass over (X, y) producing fp64
+//                     [n, sum_x, sum_y, sum_xx, sum_xy] — the whole
+//                     closed-form fit input AND the whole DP all-reduce
+//                     payload.
+//   - linear_score:   yhat = intercept + coef*X, float4-vectorised,
+//                     grid-stride (hipGraph-capturable: no allocs/syncs).
+//   - regression_metrics / score_label_metrics: one fused pass producing
+//     every sum the MAPE/R^2/max-residual/Pearson formulas need.
+// All reductions: per-wave shuffle -> LDS -> one fp64 atomicAdd per block
+// (Guideline 12); accumulation in fp64 so 1B-row sums stay exact to ~2^-40.
+#include <ATen/cuda/CUDAContext.h>
+#include <hip/hip_runtime.h>
+#include <torch/extension.h>
+
+#include "reduce.h"
+
+#define RED_BLOCK 256
+#define RED_WAVES (RED_BLOCK / 64)
+
+static inline int reduce_grid(long long n, int per_thread = 4) {
+  long long want = (n + (long long)RED_BLOCK * per_thread - 1) /
+                   ((long long)RED_BLOCK * per_thread);
+  return (int)std::min<long long>(want, 2048);  // Guideline 11 cap
+}
+
+// ---- linreg_stats ---------------------------------------------------------
+
+__global__ void linreg_stats_kernel(const float* __restrict__ x,
+                                    const float* __restrict__ y,
+                                    double* __restrict__ out, long long n) {
+  double sx = 0, sy = 0, sxx = 0, sxy = 0;
+  const long long stride = (long long)gridDim.x * RED_BLOCK;
+  long long i = (long long)blockIdx.x * RED_BLOCK + threadIdx.x;
+  // float4 main loop (coalesced 16B/lane)
+  const long long n4 = n & ~3ll;
+  for (long long j = i * 4; j < n4; j += stride * 4) {
+    if (j + 3 < n4) {
+      float4 xv = *(const float4*)(x + j);
+      float4 yv = *(const float4*)(y + j);
+      sx += (double)xv.x + xv.y + xv.z + xv.w;
+      sy += (double)yv.x + yv.y + yv.z + yv.w;
+      sxx += (double)xv.x * xv.x + (double)xv.y * xv.y +
+             (double)xv.z * xv.z + (double)xv.w * xv.w;
+      sxy += (double)xv.x * yv.x + (double)xv.y * yv.y +
+             (double)xv.z * yv.z + (double)xv.w * yv.w;
+    }
+  }
+  // tail
+  for (long long j = n4 + i; j < n; j += stride) {
+    double xv = x[j], yv = y[j];
+    sx += xv; sy += yv; sxx += xv * xv; sxy += xv * yv;
+  }
+  __shared__ double lds[RED_WAVES];
+  double t;
+  t = block_sum_f64<RED_WAVES>(sx, lds);
+  if (threadIdx.x == 0) atomicAdd(&out[1], t);
+  t = block_sum_f64<RED_WAVES>(sy, lds);
+  if (threadIdx.x == 0) atomicAdd(&out[2], t);
+  t = block_sum_f64<RED_WAVES>(sxx, lds);
+  if (threadIdx.x == 0) atomicAdd(&out[3], t);
+  t = block_sum_f64<RED_WAVES>(sxy, lds);
+  if (threadIdx.x == 0) atomicAdd(&out[4], t);
+}
+
+at::Tensor linreg_stats_hip(const at::Tensor& x, const at::Tensor& y) {
+  TORCH_CHECK(x.is_cuda() && y.is_cuda() && x.numel() == y.numel());
+  TORCH_CHECK(x.scalar_type() == at::kFloat && y.scalar_type() == at::kFloat);
+  long long n = x.numel();
+  auto out = at::zeros({5}, x.options().dtype(at::kDouble));
+  auto stream = at::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL(linreg_stats_kernel, dim3(reduce_grid(n)),
+                     dim3(RED_BLOCK), 0, stream, x.data_ptr<float>(),
+                     y.data_ptr<float>(), out.data_ptr<double>(), n);
+  out[0].fill_((double)n);
+  return out;
+}
+
+// ---- linear_score ---------------------------------------------------------
+
+__global__ void linear_score_kernel(const float* __restrict__ x,
+                                    float* __restrict__ out, float a, float b,
+                                    long long n) {
+  const long long stride = (long long)gridDim.x * blockDim.x;
+  long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long long n4 = n >> 2;
+  const float4* x4 = (const float4*)x;
+  float4* o4 = (float4*)out;
+  for (long long j = i; j < n4; j += stride) {
+    float4 v = x4[j];
+    o4[j] = {fmaf(b, v.x, a), fmaf(b, v.y, a), fmaf(b, v.z, a),
+             fmaf(b, v.w, a)};
+  }
+  for (long long j = n4 * 4 + i; j < n; j += stride)
+    out[j] = fmaf(b, x[j], a);
+}
+
+at::Tensor linear_score_hip(const at::Tensor& x, double intercept,
+                            double coef) {
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kFloat);
+  long long n = x.numel();
+  auto out = at::empty_like(x);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL(linear_score_kernel, dim3(reduce_grid(n)),
+                     dim3(RED_BLOCK), 0, stream, x.data_ptr<float>(),
+                     out.data_ptr<float>(), (float)intercept, (float)coef, n);
+  return out;
+}
+
+// ---- regression_metrics (offline: stage_1:79-90 semantics) ---------------
+// out = [n, sum_ape, ss_res, sum_y, sum_yy, max_resid]
+
+__global__ void regression_metrics_kernel(const float* __restrict__ y,
+                                          const float* __restrict__ yhat,
+                                          double* __restrict__ out,
+                                          long long n) {
+  const double MAPE_EPS = 2.220446049250313e-16;  // sklearn epsilon
+  double s_ape = 0, ss_res = 0, s_y = 0, s_yy = 0, max_res = 0;
+  const long long stride = (long long)gridDim.x * RED_BLOCK;
+  for (long long j = (long long)blockIdx.x * RED_BLOCK + threadIdx.x; j < n;
+       j += stride) {
+    double yv = y[j], pv = yhat[j];
+    double r = yv - pv;
+    double ar = fabs(r);
+    s_ape += ar / fmax(fabs(yv), MAPE_EPS);
+    ss_res += r * r;
+    s_y += yv;
+    s_yy += yv * yv;
+    max_res = fmax(max_res, ar);
+  }
+  __shared__ double lds[RED_WAVES];
+  double t;
+  t = block_sum_f64<RED_WAVES>(s_ape, lds);
+  if (threadIdx.x == 0) atomicAdd(&out[1], t);
+  t = block_sum_f64<RED_WAVES>(ss_res, lds);
+  if (threadIdx.x == 0) atomicAdd(&out[2], t);
+  t = block_sum_f64<RED_WAVES>(s_y, lds);
+  if (threadIdx.x == 0) atomicAdd(&out[3], t);
+  t = block_sum_f64<RED_WAVES>(s_yy, lds);
+  if (threadIdx.x == 0) atomicAdd(&out[4], t);
+  max_res = wave_max_f64(max_res);
+  if ((threadIdx.x & 63) == 0) atomic_max_nonneg_f64(&out[5], max_res);
+}
+
+at::Tensor regression_metrics_hip(const at::Tensor& y,
+                                  const at::Tensor& yhat) {
+  TORCH_CHECK(y.is_cuda() && yhat.is_cuda() && y.numel() == yhat.numel());
+  long long n = y.numel();
+  auto yf = y.scalar_type() == at::kFloat ? y : y.to(at::kFloat);
+  auto pf = yhat.scalar_type() == at::kFloat ? yhat : yhat.to(at::kFloat);
+  auto out = at::zeros({6}, y.options().dtype(at::kDouble));
+  auto stream = at::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL(regression_metrics_kernel, dim3(reduce_grid(n)),
+                     dim3(RED_BLOCK), 0, stream, yf.data_ptr<float>(),
+                     pf.data_ptr<float>(), out.data_ptr<double>(), n);
+  out[0].fill_((double)n);
+  return out;
+}
+
+// ---- score_label_metrics (online: stage_4:87-113 semantics) ---------------
+// out = [n, sum_ape, max_ape, s_s, s_l, s_ss, s_ll, s_sl]
+
+__global__ void score_label_metrics_kernel(const float* __restrict__ s,
+                                           const float* __restrict__ l,
+                                           double* __restrict__ out,
+                                           long long n) {
+  double s_ape = 0, max_ape = 0, s_s = 0, s_l = 0, s_ss = 0, s_ll = 0,
+         s_sl = 0;
+  const long long stride = (long long)gridDim.x * RED_BLOCK;
+  for (long long j = (long long)blockIdx.x * RED_BLOCK + threadIdx.x; j < n;
+       j += stride) {
+    double sv = s[j], lv = l[j];
+    double ape = fabs(sv / lv - 1.0);  // reference has no eps guard
+    s_ape += ape;
+    max_ape = fmax(max_ape, ape);
+    s_s += sv; s_l += lv;
+    s_ss += sv * sv; s_ll += lv * lv; s_sl += sv * lv;
+  }
+  __shared__ double lds[RED_WAVES];
+  double t;
+  t = block_sum_f64<RED_WAVES>(s_ape, lds);
+  if (threadIdx.x == 0) atomicAdd(&out[1], t);
+  t = block_sum_f64<RED_WAVES>(s_s, lds);
+  if (threadIdx.x == 0) atomicAdd(&out[3], t);
+  t = block_sum_f64<RED_WAVES>(s_l, lds);
+  if (threadIdx.x == 0) atomicAdd(&out[4], t);
+  t = block_sum_f64<RED_WAVES>(s_ss, lds);
+  if (threadIdx.x == 0) atomicAdd(&out[5], t);
+  t = block_sum_f64<RED_WAVES>(s_ll, lds);
+  if (threadIdx.x == 0) atomicAdd(&out[6], t);
+  t = block_sum_f64<RED_WAVES>(s_sl, lds);
+  if (threadIdx.x == 0) atomicAdd(&out[7], t);
+  max_ape = wave_max_f64(max_ape);
+  if ((threadIdx.x & 63) == 0) atomic_max_nonneg_f64(&out[2], max_ape);
+}
+
+at::Tensor score_label_metrics_hip(const at::Tensor& s, const at::Tensor& l) {
+  TORCH_CHECK(s.is_cuda() && l.is_cuda() && s.numel() == l.numel());
+  long long n = s.numel();
+  auto sf = s.scalar_type() == at::kFloat ? s : s.to(at::kFloat);
+  auto lf = l.scalar_type() == at::kFloat ? l : l.to(at::kFloat);
+  auto out = at::zeros({8}, s.options().dtype(at::kDouble));
+  auto stream = at::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL(score_label_metrics_kernel, dim3(reduce_grid(n)),
+                     dim3(RED_BLOCK), 0, stream, sf.data_ptr<float>(),
+                     lf.data_ptr<float>(), out.data_ptr<double>(), n);
+  out[0].fill_((double)n);
+  return out;
+}

@@ -147,23 +147,28 @@ def score_label_sums_cpu(scores: torch.Tensor, labels: torch.Tensor):
 # GEMM oracle
 # --------------------------------------------------------------------------
 
-def gemm_bf16_cpu(
-    a: torch.Tensor,
-    b: torch.Tensor,
+def linear_bf16_cpu(
+    x: torch.Tensor,
+    w: torch.Tensor,
     bias: torch.Tensor | None = None,
     relu: bool = False,
-    trans_a: bool = False,
     mask: torch.Tensor | None = None,
     out_fp32: bool = False,
 ) -> torch.Tensor:
-    a32 = a.float().t() if trans_a else a.float()
-    c = a32 @ b.float()
+    c = x.float() @ w.float().t()
     if bias is not None:
         c = c + bias.float()
     if relu:
         c = torch.relu(c)
     if mask is not None:
         c = c * (mask.float() > 0)
+    return c if out_fp32 else c.bfloat16()
+
+
+def gemm_tn_bf16_cpu(
+    a: torch.Tensor, b: torch.Tensor, out_fp32: bool = False
+) -> torch.Tensor:
+    c = a.float().t() @ b.float()
     return c if out_fp32 else c.bfloat16()
 
 

@@ -29,21 +29,33 @@ log = configure_logger(__name__)
 
 
 class CycleState:
-    """Device-resident accumulated dataset history + current-day cursor."""
+    """Device-resident accumulated dataset history + current-day cursor.
+
+    ``history_days`` bounds the training window (None = all history, the
+    reference's read-all semantics, stage_1:59-71).  The benchmark pins it
+    to 1 so per-step work is constant (steady-state cycle timing).
+    """
 
     def __init__(self, device: str, start_date: date_t, rank: int = 0,
-                 world_size: int = 1):
+                 world_size: int = 1, history_days: int | None = None):
         self.device = device
         self.date = start_date
         self.rank = rank
         self.world_size = world_size
+        self.history_days = history_days
         self.y = torch.empty(0, device=device)
         self.X = torch.empty(0, device=device)
-        self._next: tuple[torch.Tensor, torch.Tensor] | None = None
+        self._day_sizes: list[int] = []
 
     def append_day(self, y: torch.Tensor, X: torch.Tensor) -> None:
         self.y = torch.cat([self.y, y])
         self.X = torch.cat([self.X, X])
+        self._day_sizes.append(int(y.shape[0]))
+        if self.history_days is not None:
+            while len(self._day_sizes) > self.history_days:
+                drop = self._day_sizes.pop(0)
+                self.y = self.y[drop:].contiguous()
+                self.X = self.X[drop:].contiguous()
 
 
 def run_cycle(

@@ -113,19 +113,18 @@ class TestFitAndMetrics:
 
 
 class TestMLPOps:
-    def test_gemm_cpu_variants(self):
-        a = torch.randn(64, 32).bfloat16()
-        b = torch.randn(32, 48).bfloat16()
+    def test_linear_and_tn_cpu(self):
+        x = torch.randn(64, 32).bfloat16()
+        w = torch.randn(48, 32).bfloat16()
         bias = torch.randn(48).bfloat16()
-        c = ops.gemm_bf16(a, b, bias=bias, relu=True)
-        want = torch.relu(a.float() @ b.float() + bias.float()).bfloat16()
+        c = ops.linear_bf16(x, w, bias=bias, relu=True)
+        want = torch.relu(x.float() @ w.float().t() + bias.float()).bfloat16()
         assert torch.equal(c, want)
 
-        at = torch.randn(32, 64).bfloat16()
-        c2 = ops.gemm_bf16(at, b.t().contiguous()[:32, :], trans_a=True,
-                           out_fp32=True)
-        want2 = at.float().t() @ b.t().contiguous()[:32, :].float()
-        assert torch.allclose(c2, want2)
+        a = torch.randn(32, 64).bfloat16()
+        b = torch.randn(32, 48).bfloat16()
+        c2 = ops.gemm_tn_bf16(a, b, out_fp32=True)
+        assert torch.allclose(c2, a.float().t() @ b.float())
 
     def test_expand_rowdot_coldot(self):
         x = torch.randn(16)

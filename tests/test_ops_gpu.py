@@ -1,0 +1,194 @@
+"""GPU kernel numerics: every HIP kernel vs its CPU/fp32 torch oracle.
+
+All tests marked gpu; the CPU implementations in ops/reference.py are the
+oracles (plain PyTorch fp32/fp64 math).
+"""
+import numpy as np
+import pytest
+import torch
+
+from bodywork_mlops_demo_amd import ops
+from bodywork_mlops_demo_amd.ops import reference
+
+pytestmark = pytest.mark.gpu
+
+DEV = "cuda:0"
+
+
+@pytest.fixture(autouse=True)
+def _require_hip():
+    if torch.cuda.is_available():
+        assert ops.hip_available(), (
+            "GPU present but _hipcore extension missing — the HIP path "
+            "must be the one that runs (no silent eager fallback)"
+        )
+
+
+class TestDatagenGPU:
+    def test_matches_cpu_oracle_bitstream(self):
+        # kappa=60 -> no y<0 culls, so streams align row-for-row
+        yg, Xg = ops.datagen(100_000, 30, seed=1234, kappa=60.0, device=DEV)
+        yc, Xc = ops.datagen(100_000, 30, seed=1234, kappa=60.0, device="cpu")
+        assert yg.shape[0] == 100_000 == yc.shape[0]
+        # X path is pure fp32 multiplies of the same philox ints: bitwise
+        assert torch.equal(Xg.cpu(), Xc)
+        # y path uses device log/cos: allow ulp-scale differences
+        torch.testing.assert_close(yg.cpu(), yc, rtol=1e-5, atol=1e-3)
+
+    def test_cull_and_determinism(self):
+        y1, X1 = ops.datagen(200_000, 100, seed=7, device=DEV)
+        y2, X2 = ops.datagen(200_000, 100, seed=7, device=DEV)
+        assert torch.equal(y1, y2) and torch.equal(X1, X2)
+        assert (y1 >= 0).all()
+        assert y1.shape[0] < 200_000  # the cull removed something
+        # kept fraction matches the CPU oracle within noise
+        yc, _ = ops.datagen(200_000, 100, seed=7, device="cpu")
+        assert abs(y1.shape[0] - yc.shape[0]) <= 50
+
+    def test_stable_order(self):
+        """Compaction preserves row order (pandas query parity)."""
+        yg, Xg = ops.datagen(50_000, 200, seed=3, device=DEV)
+        yc, Xc = ops.datagen(50_000, 200, seed=3, device="cpu")
+        n = min(yg.shape[0], yc.shape[0])
+        # order-sensitive comparison of the X stream (X is bitwise-stable)
+        mismatch = (Xg[:n].cpu() != Xc[:n]).sum().item()
+        assert mismatch <= 2  # only boundary-flip induced shifts allowed
+
+    def test_large_n(self):
+        y, X = ops.datagen(50_000_000, 10, seed=5, device=DEV)
+        assert y.shape[0] > 25_000_000
+        assert X.min() >= 0 and X.max() <= 100
+
+
+class TestLinregGPU:
+    def test_stats_match_fp64_oracle(self):
+        X = torch.rand(1_000_000, device=DEV) * 100
+        y = 1 + 0.5 * X + torch.randn_like(X) * 10
+        got = ops.linreg_stats(X, y).cpu()
+        want = reference.linreg_stats_cpu(X.cpu(), y.cpu())
+        torch.testing.assert_close(got, want, rtol=1e-12, atol=1e-6)
+
+    def test_score_exact(self):
+        X = torch.rand(100_001, device=DEV) * 100  # odd n: tail path
+        got = ops.linear_score(X, 1.25, 0.5)
+        want = 1.25 + 0.5 * X
+        # same fmaf on both paths? oracle is torch mul+add; allow 1 ulp
+        torch.testing.assert_close(got, want, rtol=1e-6, atol=1e-5)
+
+    def test_regression_metrics(self):
+        y = torch.rand(500_000, device=DEV) * 100 + 1
+        yhat = y + torch.randn_like(y) * 5
+        got = ops.regression_metrics(y, yhat)
+        want_sums = reference.metric_sums_cpu(y.cpu(), yhat.cpu())
+        want_mape = want_sums[1] / want_sums[0]
+        assert got["MAPE"] == pytest.approx(want_mape, rel=1e-9)
+        assert got["max_residual"] == pytest.approx(want_sums[5], rel=1e-9)
+
+    def test_score_label_metrics(self):
+        labels = torch.rand(300_000, device=DEV) * 100 + 1
+        scores = labels + torch.randn_like(labels) * 5
+        got = ops.score_label_metrics(scores, labels)
+        sc, lc = scores.cpu().double(), labels.cpu().double()
+        corr = float(np.corrcoef(sc.numpy(), lc.numpy())[0, 1])
+        assert got["r_squared"] == pytest.approx(corr, rel=1e-9)
+        assert got["MAPE"] == pytest.approx(
+            float((sc / lc - 1).abs().mean()), rel=1e-9)
+
+
+class TestMLPOpsGPU:
+    def test_expand1d(self):
+        x = torch.randn(1000, device=DEV)
+        w = torch.randn(4096, device=DEV).bfloat16()
+        b = torch.randn(4096, device=DEV).bfloat16()
+        got = ops.expand1d_bf16(x, w, b, relu=True)
+        want = reference.expand1d_cpu(x.cpu(), w.cpu(), b.cpu(), relu=True)
+        torch.testing.assert_close(got.cpu().float(), want.float(),
+                                   rtol=1e-2, atol=1e-2)
+
+    def test_expand1d_mask(self):
+        x = torch.randn(256, device=DEV)
+        w = torch.randn(512, device=DEV).bfloat16()
+        mask = torch.randn(256, 512, device=DEV).bfloat16()
+        got = ops.expand1d_bf16(x, w, None, relu=False, mask=mask)
+        want = reference.expand1d_cpu(x.cpu(), w.cpu(), None, False,
+                                      mask.cpu())
+        torch.testing.assert_close(got.cpu().float(), want.float(),
+                                   rtol=1e-2, atol=1e-2)
+
+    def test_rowdot(self):
+        h = torch.randn(777, 4096, device=DEV).bfloat16()
+        w = torch.randn(4096, device=DEV).bfloat16()
+        got = ops.rowdot_bf16(h, w, 0.5)
+        want = reference.rowdot_cpu(h.cpu(), w.cpu(), 0.5)
+        torch.testing.assert_close(got.cpu(), want, rtol=2e-2, atol=2e-1)
+
+    def test_coldot_colsum(self):
+        m = torch.randn(3000, 2048, device=DEV).bfloat16()
+        v = torch.randn(3000, device=DEV)
+        dw, cs = ops.coldot_bf16(m, v, also_colsum=True)
+        dw_w = reference.coldot_cpu(m.cpu(), v.cpu())
+        cs_w = reference.colsum_cpu(m.cpu())
+        torch.testing.assert_close(dw.cpu(), dw_w, rtol=2e-2, atol=2e-1)
+        torch.testing.assert_close(cs.cpu(), cs_w, rtol=2e-2, atol=2e-1)
+        cs2 = ops.colsum_bf16(m)
+        torch.testing.assert_close(cs2.cpu(), cs_w, rtol=2e-2, atol=2e-1)
+
+
+class TestGemmGPU:
+    """Asymmetric operands everywhere (transpose-detecting — guide G9)."""
+
+    @staticmethod
+    def _relerr(got, want):
+        return (got.float() - want.float()).abs().max().item() / (
+            want.float().abs().max().item() + 1e-9
+        )
+
+    @pytest.mark.parametrize("m,n,k", [
+        (128, 128, 64),       # single tile
+        (256, 256, 256),      # multi-tile
+        (200, 130, 96),       # edge tiles all dims
+        (1024, 4096, 4096),   # MLP shape
+        (64, 64, 64),         # sub-tile block
+    ])
+    def test_linear_nt(self, m, n, k):
+        x = (torch.randn(m, k, device=DEV) * 0.5).bfloat16()
+        w = (torch.randn(n, k, device=DEV) * 0.5).bfloat16()
+        got = ops.linear_bf16(x, w)
+        want = (x.float() @ w.float().t())
+        assert self._relerr(got.cpu(), want.cpu()) < 2e-2
+
+    def test_linear_bias_relu(self):
+        x = (torch.randn(300, 256, device=DEV) * 0.5).bfloat16()
+        w = (torch.randn(200, 256, device=DEV) * 0.5).bfloat16()
+        b = torch.randn(200, device=DEV).bfloat16()
+        got = ops.linear_bf16(x, w, bias=b, relu=True)
+        want = torch.relu(x.float() @ w.float().t() + b.float())
+        assert self._relerr(got.cpu(), want.cpu()) < 2e-2
+        assert (got.float() >= 0).all()
+
+    def test_linear_mask(self):
+        x = (torch.randn(256, 512, device=DEV) * 0.5).bfloat16()
+        w = (torch.randn(256, 512, device=DEV) * 0.5).bfloat16()
+        mask = torch.randn(256, 256, device=DEV).bfloat16()
+        got = ops.linear_bf16(x, w, mask=mask)
+        want = (x.float() @ w.float().t()) * (mask.float() > 0)
+        assert self._relerr(got.cpu(), want.cpu()) < 2e-2
+        assert (got.float()[mask.float() <= 0] == 0).all()
+
+    @pytest.mark.parametrize("r,m,n", [
+        (256, 128, 128),
+        (1000, 200, 130),
+        (8192, 4096, 4096),   # dW2 shape
+    ])
+    def test_gemm_tn(self, r, m, n):
+        a = (torch.randn(r, m, device=DEV) * 0.5).bfloat16()
+        b = (torch.randn(r, n, device=DEV) * 0.5).bfloat16()
+        got = ops.gemm_tn_bf16(a, b, out_fp32=True)
+        want = a.float().t() @ b.float()
+        assert self._relerr(got.cpu(), want.cpu()) < 2e-2
+
+    def test_out_fp32(self):
+        x = (torch.randn(128, 64, device=DEV)).bfloat16()
+        w = (torch.randn(128, 64, device=DEV)).bfloat16()
+        got = ops.linear_bf16(x, w, out_fp32=True)
+        assert got.dtype == torch.float32
