@@ -54,10 +54,22 @@ class BatchedScorer:
         log.info(f"captured scoring hipGraph for batch bucket {b}")
 
     def score_tensor(self, X: torch.Tensor) -> torch.Tensor:
-        """Score a device-resident batch; returns fp32 predictions."""
+        """Score a device-resident batch; returns fp32 predictions.
+
+        Batches beyond the largest graph bucket are scored in
+        largest-bucket chunks (bounds activation memory for the MLP:
+        a 2^20-row chunk holds ~16 GB of bf16 activations at H=4096).
+        """
         n = X.shape[0]
         if not self.use_graphs:
             return self.model.predict(X)
+        max_b = self.BUCKETS[-1]
+        if n > max_b:
+            out = torch.empty(n, device=self.device, dtype=torch.float32)
+            for lo in range(0, n, max_b):
+                hi = min(lo + max_b, n)
+                out[lo:hi] = self.score_tensor(X[lo:hi])
+            return out
         b = self._bucket(n)
         if b not in self._graphs:
             try:
