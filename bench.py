@@ -100,7 +100,7 @@ def main() -> None:
     def one_cycle():
         return run_cycle(
             state, store, args.rows, model_type=args.model,
-            process_group=pg, persist_fmt="npz",
+            process_group=pg, persist_fmt="npy",
             mlp_steps=args.mlp_steps, mlp_batch_size=args.mlp_batch,
             use_graphs=not args.no_graphs and use_cuda,
         )

@@ -204,6 +204,28 @@ def score_label_metrics(scores: torch.Tensor, labels: torch.Tensor) -> dict:
 # train/test split — reference stage_1:98-103 (train_test_split, seed 42)
 # --------------------------------------------------------------------------
 
+def random_split(
+    X: torch.Tensor,
+    y: torch.Tensor,
+    test_frac: float = 0.2,
+    seed: int = 42,
+) -> tuple[torch.Tensor, torch.Tensor, torch.Tensor, torch.Tensor]:
+    """Seeded random split into (X_train, y_train, X_test, y_test).
+
+    Row i is a test row iff ``philox(seed, i).x < test_frac * 2^32``
+    (E[test share] = test_frac exactly; count is binomial), both
+    partitions stable-ordered — a single fused GPU pass instead of the
+    reference's host-side permutation (``train_test_split`` at
+    stage_1:98-103, which cost 2/3 of the GPU train phase).  Same philox
+    stream on CPU, so splits are device-independent.
+    """
+    if X.device.type == "cuda":
+        core = _core(X.device)
+        return core.random_split(X.contiguous().float(),
+                                 y.contiguous().float(), test_frac, seed)
+    return reference.random_split_cpu(X, y, test_frac, seed)
+
+
 def train_test_split_indices(
     n: int, test_size: float = 0.2, seed: int = 42, device: str | torch.device = "cpu"
 ) -> tuple[torch.Tensor, torch.Tensor]:
@@ -288,16 +310,25 @@ def expand1d_bf16(
     return reference.expand1d_cpu(x, w, b, relu, mask)
 
 
-def rowdot_bf16(h: torch.Tensor, w: torch.Tensor, b: float = 0.0) -> torch.Tensor:
+def rowdot_bf16(
+    h: torch.Tensor, w: torch.Tensor, b: torch.Tensor | float = 0.0
+) -> torch.Tensor:
     """out[i] = sum_j h[i,j] * w[j] + b — the MLP output head (GEMV).
 
     One wave per row, vectorised bf16x8 loads, shuffle reduction;
     fp32 output (scoring precision — BASELINE "bf16 fit + fp32 scoring").
+    ``b`` may be a 1-element device tensor (read on-device, so the whole
+    forward is hipGraph-capturable) or a python float.
     """
     if h.device.type == "cuda":
         core = _core(h.device)
-        return core.rowdot_bf16(h.contiguous(), w.contiguous(), b)
-    return reference.rowdot_cpu(h, w, b)
+        if not torch.is_tensor(b):
+            b = torch.full((1,), float(b), device=h.device,
+                           dtype=torch.float32)
+        return core.rowdot_bf16(h.contiguous(), w.contiguous(),
+                                b.reshape(1).float())
+    return reference.rowdot_cpu(h, w, float(b) if not torch.is_tensor(b)
+                                else float(b.reshape(-1)[0]))
 
 
 def coldot_bf16(
@@ -318,3 +349,35 @@ def colsum_bf16(m: torch.Tensor) -> torch.Tensor:
         core = _core(m.device)
         return core.colsum_bf16(m.contiguous())
     return reference.colsum_cpu(m)
+
+
+def adam_step(
+    p: torch.Tensor,
+    g: torch.Tensor,
+    m: torch.Tensor,
+    v: torch.Tensor,
+    p_bf16: torch.Tensor | None,
+    lr: float,
+    t: int,
+    beta1: float = 0.9,
+    beta2: float = 0.999,
+    eps: float = 1e-8,
+) -> None:
+    """Fused in-place Adam update, optionally writing the bf16 shadow
+    weight in the same pass (one HBM traversal per parameter per step)."""
+    if p.device.type == "cuda":
+        core = _core(p.device)
+        core.adam_step(p.view(-1), g.reshape(-1).float(), m.view(-1),
+                       v.view(-1),
+                       p_bf16.view(-1) if p_bf16 is not None else None,
+                       lr, beta1, beta2, eps, t)
+        return
+    reference.adam_step_cpu(p, g, m, v, p_bf16, lr, t, beta1, beta2, eps)
+
+
+def transpose_to_bf16(src: torch.Tensor) -> torch.Tensor:
+    """dst[C,R] bf16 = src[R,C] fp32 transposed (LDS-tiled on GPU)."""
+    if src.device.type == "cuda":
+        core = _core(src.device)
+        return core.transpose_to_bf16(src.contiguous())
+    return src.t().contiguous().bfloat16()

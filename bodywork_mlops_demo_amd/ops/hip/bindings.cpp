@@ -9,6 +9,8 @@ std::tuple<at::Tensor, at::Tensor> datagen_hip(int64_t n, int64_t seed,
                                                int64_t stream_offset,
                                                double alpha, double beta,
                                                double sigma);
+std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor> random_split_hip(
+    const at::Tensor& X, const at::Tensor& y, double test_frac, int64_t seed);
 // linreg.hip
 at::Tensor linreg_stats_hip(const at::Tensor& x, const at::Tensor& y);
 at::Tensor linear_score_hip(const at::Tensor& x, double intercept,
@@ -20,7 +22,7 @@ at::Tensor expand1d_bf16_hip(const at::Tensor& x, const at::Tensor& w,
                              const c10::optional<at::Tensor>& b, bool relu,
                              const c10::optional<at::Tensor>& mask);
 at::Tensor rowdot_bf16_hip(const at::Tensor& h, const at::Tensor& w,
-                           double bias);
+                           const at::Tensor& bias);
 at::Tensor coldot_bf16_hip(const at::Tensor& m, const at::Tensor& v,
                            bool also_colsum);
 at::Tensor colsum_bf16_hip(const at::Tensor& m);
@@ -31,26 +33,39 @@ at::Tensor linear_bf16_hip(const at::Tensor& x, const at::Tensor& w,
                            bool out_fp32);
 at::Tensor gemm_tn_bf16_hip(const at::Tensor& a, const at::Tensor& b,
                             bool out_fp32);
+// optim.hip
+void adam_step_hip(at::Tensor p, const at::Tensor& g, at::Tensor m,
+                   at::Tensor v, const c10::optional<at::Tensor>& p_bf16,
+                   double lr, double beta1, double beta2, double eps,
+                   int64_t t);
+at::Tensor transpose_to_bf16_hip(const at::Tensor& src);
 
 TORCH_LIBRARY(bodywork_hip, m) {
   m.def("datagen(int n, int seed, int stream_offset, float alpha, float beta, "
         "float sigma) -> (Tensor, Tensor)");
+  m.def("random_split(Tensor X, Tensor y, float test_frac, int seed) -> "
+        "(Tensor, Tensor, Tensor, Tensor)");
   m.def("linreg_stats(Tensor x, Tensor y) -> Tensor");
   m.def("linear_score(Tensor x, float intercept, float coef) -> Tensor");
   m.def("regression_metrics(Tensor y, Tensor yhat) -> Tensor");
   m.def("score_label_metrics(Tensor s, Tensor l) -> Tensor");
   m.def("expand1d_bf16(Tensor x, Tensor w, Tensor? b, bool relu, "
         "Tensor? mask) -> Tensor");
-  m.def("rowdot_bf16(Tensor h, Tensor w, float bias) -> Tensor");
+  m.def("rowdot_bf16(Tensor h, Tensor w, Tensor bias) -> Tensor");
   m.def("coldot_bf16(Tensor m, Tensor v, bool also_colsum) -> Tensor");
   m.def("colsum_bf16(Tensor m) -> Tensor");
   m.def("linear_bf16(Tensor x, Tensor w, Tensor? bias, bool relu, "
         "Tensor? mask, bool out_fp32) -> Tensor");
   m.def("gemm_tn_bf16(Tensor a, Tensor b, bool out_fp32) -> Tensor");
+  m.def("adam_step(Tensor(a!) p, Tensor g, Tensor(b!) m, Tensor(c!) v, "
+        "Tensor(d!)? p_bf16, float lr, float beta1, float beta2, float eps, "
+        "int t) -> ()");
+  m.def("transpose_to_bf16(Tensor src) -> Tensor");
 }
 
 TORCH_LIBRARY_IMPL(bodywork_hip, CUDA, m) {
   m.impl("datagen", datagen_hip);
+  m.impl("random_split", random_split_hip);
   m.impl("linreg_stats", linreg_stats_hip);
   m.impl("linear_score", linear_score_hip);
   m.impl("regression_metrics", regression_metrics_hip);
@@ -61,6 +76,8 @@ TORCH_LIBRARY_IMPL(bodywork_hip, CUDA, m) {
   m.impl("colsum_bf16", colsum_bf16_hip);
   m.impl("linear_bf16", linear_bf16_hip);
   m.impl("gemm_tn_bf16", gemm_tn_bf16_hip);
+  m.impl("adam_step", adam_step_hip);
+  m.impl("transpose_to_bf16", transpose_to_bf16_hip);
 }
 
 // datagen takes no tensor argument, so the dispatcher cannot route it by

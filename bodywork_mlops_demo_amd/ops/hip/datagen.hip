@@ -203,3 +203,125 @@ std::tuple<at::Tensor, at::Tensor> datagen_hip(
   int64_t kept = total.to(at::kLong).item<int64_t>();
   return {y_out.narrow(0, 0, kept), X_out.narrow(0, 0, kept)};
 }
+
+// ---- random 80/20 split (replaces sklearn train_test_split, stage_1:98) ---
+// Element i is a TEST row iff philox(seed, i).x < tau; both partitions keep
+// row order (stable).  Single pass + the same two-level scan as datagen:
+// no host-side permutation, no gather indices — the reference's CPU
+// shuffle was 2/3 of the GPU cycle's train phase before this kernel.
+
+__global__ void split_flag_kernel(unsigned int* __restrict__ block_counts,
+                                  long long n, unsigned int key0,
+                                  unsigned int key1, unsigned int tau) {
+  const long long i = (long long)blockIdx.x * DG_BLOCK + threadIdx.x;
+  int is_test = 0;
+  if (i < n) {
+    Philox4 r = philox4x32((unsigned long long)i, key0, key1);
+    is_test = r.x < tau;
+  }
+  __shared__ unsigned int wave_cnt[DG_WAVES];
+  unsigned long long ballot = __ballot(is_test);
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  if (lane == 0) wave_cnt[wave] = (unsigned int)__popcll(ballot);
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    unsigned int c = 0;
+#pragma unroll
+    for (int w = 0; w < DG_WAVES; ++w) c += wave_cnt[w];
+    block_counts[blockIdx.x] = c;
+  }
+}
+
+__global__ void split_scatter_kernel(
+    const float* __restrict__ X, const float* __restrict__ y,
+    const unsigned int* __restrict__ test_offsets,
+    float* __restrict__ X_tr, float* __restrict__ y_tr,
+    float* __restrict__ X_te, float* __restrict__ y_te, long long n,
+    unsigned int key0, unsigned int key1, unsigned int tau) {
+  const long long i = (long long)blockIdx.x * DG_BLOCK + threadIdx.x;
+  int is_test = 0;
+  if (i < n) {
+    Philox4 r = philox4x32((unsigned long long)i, key0, key1);
+    is_test = r.x < tau;
+  }
+  __shared__ unsigned int wave_off[DG_WAVES];
+  unsigned long long ballot = __ballot(is_test);
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  if (lane == 0) wave_off[wave] = (unsigned int)__popcll(ballot);
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    unsigned int acc = 0;
+#pragma unroll
+    for (int w = 0; w < DG_WAVES; ++w) {
+      unsigned int c = wave_off[w];
+      wave_off[w] = acc;
+      acc += c;
+    }
+  }
+  __syncthreads();
+  if (i < n) {
+    unsigned int before =
+        (unsigned int)__popcll(ballot & ((1ull << lane) - 1ull));
+    unsigned int tests_before =
+        test_offsets[blockIdx.x] + wave_off[wave] + before;
+    if (is_test) {
+      X_te[tests_before] = X[i];
+      y_te[tests_before] = y[i];
+    } else {
+      long long train_pos = i - (long long)tests_before;
+      X_tr[train_pos] = X[i];
+      y_tr[train_pos] = y[i];
+    }
+  }
+}
+
+std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor> random_split_hip(
+    const at::Tensor& X, const at::Tensor& y, double test_frac,
+    int64_t seed) {
+  TORCH_CHECK(X.is_cuda() && y.is_cuda() && X.numel() == y.numel());
+  TORCH_CHECK(X.scalar_type() == at::kFloat && y.scalar_type() == at::kFloat);
+  long long n = X.numel();
+  const long long n_blocks = (n + DG_BLOCK - 1) / DG_BLOCK;
+  const long long n_chunks = (n_blocks + SCAN_CHUNK - 1) / SCAN_CHUNK;
+  TORCH_CHECK(n_chunks <= 4096, "random_split: n too large (max ~1.07e9)");
+  auto u32 = at::TensorOptions().dtype(at::kUInt32).device(X.device());
+  auto counts = at::empty({n_blocks}, u32);
+  auto offsets = at::empty({n_blocks}, u32);
+  auto chunk_sums = at::empty({std::max<long long>(n_chunks, 1)}, u32);
+  auto total = at::zeros({1}, u32);
+  unsigned int key0 = (unsigned int)(seed & 0xFFFFFFFFll);
+  unsigned int key1 = (seed > 0xFFFFFFFFll) ? (unsigned int)(seed >> 32)
+                                            : 0x85EBCA6Bu;  // split stream
+  unsigned int tau = (unsigned int)(test_frac * 4294967296.0);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL(split_flag_kernel, dim3(n_blocks), dim3(DG_BLOCK), 0,
+                     stream, (unsigned int*)counts.data_ptr(), n, key0, key1,
+                     tau);
+  hipLaunchKernelGGL(scan_level1_kernel, dim3(n_chunks), dim3(256), 0, stream,
+                     (const unsigned int*)counts.data_ptr(),
+                     (unsigned int*)offsets.data_ptr(),
+                     (unsigned int*)chunk_sums.data_ptr(), (long long)n_blocks);
+  hipLaunchKernelGGL(scan_level2_kernel,
+                     dim3(std::min<long long>(2048, n_chunks * 4)), dim3(256),
+                     0, stream, (unsigned int*)offsets.data_ptr(),
+                     (unsigned int*)chunk_sums.data_ptr(),
+                     (unsigned int*)total.data_ptr(), (long long)n_blocks,
+                     (int)n_chunks);
+  auto opts = X.options();
+  auto X_tr = at::empty({n}, opts);
+  auto y_tr = at::empty({n}, opts);
+  auto X_te = at::empty({n}, opts);
+  auto y_te = at::empty({n}, opts);
+  hipLaunchKernelGGL(split_scatter_kernel, dim3(n_blocks), dim3(DG_BLOCK), 0,
+                     stream, X.data_ptr<float>(), y.data_ptr<float>(),
+                     (const unsigned int*)offsets.data_ptr(),
+                     X_tr.data_ptr<float>(), y_tr.data_ptr<float>(),
+                     X_te.data_ptr<float>(), y_te.data_ptr<float>(), n, key0,
+                     key1, tau);
+  int64_t n_test = total.to(at::kLong).item<int64_t>();
+  int64_t n_train = n - n_test;
+  return {X_tr.narrow(0, 0, n_train), y_tr.narrow(0, 0, n_train),
+          X_te.narrow(0, 0, n_test), y_te.narrow(0, 0, n_test)};
+}

@@ -88,7 +88,8 @@ at::Tensor expand1d_bf16_hip(const at::Tensor& x, const at::Tensor& w,
 __global__ void rowdot_kernel(const bf16_t* __restrict__ h,
                               const bf16_t* __restrict__ w,
                               float* __restrict__ out, long long n, int H,
-                              float bias) {
+                              const float* __restrict__ bias_p) {
+  const float bias = *bias_p;  // device read: keeps predict capture-safe
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
   const int waves_per_block = blockDim.x >> 6;
@@ -108,10 +109,12 @@ __global__ void rowdot_kernel(const bf16_t* __restrict__ h,
 }
 
 at::Tensor rowdot_bf16_hip(const at::Tensor& h, const at::Tensor& w,
-                           double bias) {
+                           const at::Tensor& bias) {
   TORCH_CHECK(h.is_cuda() && h.dim() == 2);
   TORCH_CHECK(h.scalar_type() == at::kBFloat16 &&
               w.scalar_type() == at::kBFloat16);
+  TORCH_CHECK(bias.is_cuda() && bias.numel() == 1 &&
+              bias.scalar_type() == at::kFloat);
   const long long n = h.size(0);
   const int H = (int)h.size(1);
   TORCH_CHECK(H % 8 == 0, "rowdot: H must be a multiple of 8");
@@ -122,7 +125,7 @@ at::Tensor rowdot_bf16_hip(const at::Tensor& h, const at::Tensor& w,
   hipLaunchKernelGGL(rowdot_kernel, dim3(grid), dim3(64 * waves_per_block), 0,
                      stream, (const bf16_t*)h.data_ptr(),
                      (const bf16_t*)w.data_ptr(), out.data_ptr<float>(), n, H,
-                     (float)bias);
+                     bias.data_ptr<float>());
   return out;
 }
 

@@ -1,0 +1,107 @@
+// Fused optimizer + weight-copy kernels for the MLP training loop.
+//
+//   adam_step:        in-place fp32 Adam update (bias-corrected), fused
+//                     with the bf16 shadow-weight write the next forward
+//                     needs — one pass over the parameter instead of the
+//                     ~6 eager elementwise launches of a torch Adam step.
+//   transpose_to_bf16: 32x32 LDS-tiled fp32->bf16 transpose producing the
+//                     [out,in] / [in,out] weight pair the NT MFMA GEMM
+//                     consumes (both copies K-contiguous, gemm.hip).
+#include <ATen/cuda/CUDAContext.h>
+#include <hip/hip_runtime.h>
+#include <torch/extension.h>
+
+#include "bf16_utils.h"
+
+template <bool WRITE_BF16>
+__global__ void adam_step_kernel(float* __restrict__ p, const float* __restrict__ g,
+                                 float* __restrict__ m, float* __restrict__ v,
+                                 bf16_t* __restrict__ p_bf16, long long n,
+                                 float lr, float beta1, float beta2, float eps,
+                                 float inv_bc1, float inv_bc2) {
+  const long long stride = (long long)gridDim.x * blockDim.x;
+  for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    float gv = g[i];
+    float mv = beta1 * m[i] + (1.0f - beta1) * gv;
+    float vv = beta2 * v[i] + (1.0f - beta2) * gv * gv;
+    m[i] = mv;
+    v[i] = vv;
+    float mhat = mv * inv_bc1;
+    float vhat = vv * inv_bc2;
+    float pv = p[i] - lr * mhat / (sqrtf(vhat) + eps);
+    p[i] = pv;
+    if (WRITE_BF16) p_bf16[i] = f32_to_bf16(pv);
+  }
+}
+
+void adam_step_hip(at::Tensor p, const at::Tensor& g, at::Tensor m,
+                   at::Tensor v, const c10::optional<at::Tensor>& p_bf16,
+                   double lr, double beta1, double beta2, double eps,
+                   int64_t t) {
+  TORCH_CHECK(p.is_cuda() && p.scalar_type() == at::kFloat);
+  TORCH_CHECK(g.numel() == p.numel() && m.numel() == p.numel() &&
+              v.numel() == p.numel());
+  long long n = p.numel();
+  float inv_bc1 = 1.0f / (1.0f - powf((float)beta1, (float)t));
+  float inv_bc2 = 1.0f / (1.0f - powf((float)beta2, (float)t));
+  auto stream = at::cuda::getCurrentCUDAStream();
+  int grid = (int)std::min<long long>((n + 1023) / 1024, 2048);
+  if (p_bf16.has_value()) {
+    TORCH_CHECK(p_bf16->numel() == n &&
+                p_bf16->scalar_type() == at::kBFloat16);
+    hipLaunchKernelGGL((adam_step_kernel<true>), dim3(grid), dim3(256), 0,
+                       stream, p.data_ptr<float>(), g.data_ptr<float>(),
+                       m.data_ptr<float>(), v.data_ptr<float>(),
+                       (bf16_t*)p_bf16->data_ptr(), n, (float)lr,
+                       (float)beta1, (float)beta2, (float)eps, inv_bc1,
+                       inv_bc2);
+  } else {
+    hipLaunchKernelGGL((adam_step_kernel<false>), dim3(grid), dim3(256), 0,
+                       stream, p.data_ptr<float>(), g.data_ptr<float>(),
+                       m.data_ptr<float>(), v.data_ptr<float>(), nullptr, n,
+                       (float)lr, (float)beta1, (float)beta2, (float)eps,
+                       inv_bc1, inv_bc2);
+  }
+}
+
+// 32x32 LDS-tiled transpose, fp32 source -> bf16 transposed destination.
+// +1 padding column breaks the write-side bank conflict (Guideline 4).
+__global__ void transpose_to_bf16_kernel(const float* __restrict__ src,
+                                         bf16_t* __restrict__ dst,
+                                         int rows, int cols) {
+  __shared__ float tile[32][33];
+  int c0 = blockIdx.x * 32;
+  int r0 = blockIdx.y * 32;
+  // 256 threads: 8 rows of 32 per pass, 4 passes
+  int tc = threadIdx.x & 31;
+  int tr = threadIdx.x >> 5;
+#pragma unroll
+  for (int p = 0; p < 4; ++p) {
+    int r = r0 + tr + p * 8;
+    int c = c0 + tc;
+    tile[tr + p * 8][tc] = (r < rows && c < cols) ? src[(long long)r * cols + c]
+                                                  : 0.0f;
+  }
+  __syncthreads();
+#pragma unroll
+  for (int p = 0; p < 4; ++p) {
+    int r = c0 + tr + p * 8;  // destination row = source col
+    int c = r0 + tc;          // destination col = source row
+    if (r < cols && c < rows)
+      dst[(long long)r * rows + c] = f32_to_bf16(tile[tc][tr + p * 8]);
+  }
+}
+
+at::Tensor transpose_to_bf16_hip(const at::Tensor& src) {
+  TORCH_CHECK(src.is_cuda() && src.dim() == 2 &&
+              src.scalar_type() == at::kFloat);
+  int rows = (int)src.size(0), cols = (int)src.size(1);
+  auto dst = at::empty({cols, rows}, src.options().dtype(at::kBFloat16));
+  auto stream = at::cuda::getCurrentCUDAStream();
+  dim3 grid((cols + 31) / 32, (rows + 31) / 32);
+  hipLaunchKernelGGL(transpose_to_bf16_kernel, grid, dim3(256), 0, stream,
+                     src.data_ptr<float>(), (bf16_t*)dst.data_ptr(), rows,
+                     cols);
+  return dst;
+}

@@ -91,6 +91,23 @@ def datagen_cpu(
     return torch.from_numpy(y[keep].copy()), torch.from_numpy(X[keep].copy())
 
 
+_SPLIT_KEY1 = 0x85EBCA6B  # distinct philox stream from datagen's
+
+
+def random_split_cpu(
+    X: torch.Tensor, y: torch.Tensor, test_frac: float = 0.2, seed: int = 42
+):
+    """CPU oracle of the GPU philox partition (ops/hip/datagen.hip)."""
+    n = X.numel()
+    key1 = (seed >> 32) if seed > 0xFFFFFFFF else _SPLIT_KEY1
+    r = philox4x32(np.arange(n, dtype=np.uint64), seed, key1)
+    tau = np.uint32(int(test_frac * 4294967296.0))
+    is_test = r[:, 0] < tau
+    Xf, yf = X.float(), y.float()
+    te = torch.from_numpy(is_test)
+    return Xf[~te], yf[~te], Xf[te], yf[te]
+
+
 # --------------------------------------------------------------------------
 # OLS statistics / scoring / metrics oracles (fp64 accumulation)
 # --------------------------------------------------------------------------
@@ -202,3 +219,14 @@ def coldot_cpu(m: torch.Tensor, v: torch.Tensor, also_colsum: bool = False):
 
 def colsum_cpu(m: torch.Tensor) -> torch.Tensor:
     return m.float().sum(dim=0)
+
+
+def adam_step_cpu(p, g, m, v, p_bf16, lr, t, beta1, beta2, eps):
+    g = g.reshape(p.shape).to(p.dtype)
+    m.mul_(beta1).add_(g, alpha=1 - beta1)
+    v.mul_(beta2).addcmul_(g, g, value=1 - beta2)
+    mhat = m / (1 - beta1**t)
+    vhat = v / (1 - beta2**t)
+    p.sub_(lr * mhat / (vhat.sqrt() + eps))
+    if p_bf16 is not None:
+        p_bf16.copy_(p.bfloat16())

@@ -64,7 +64,7 @@ def run_cycle(
     n_rows: int,
     model_type: str = "linear",
     process_group=None,
-    persist_fmt: str = "npz",
+    persist_fmt: str = "npy",
     mlp_steps: int = 50,
     mlp_batch_size: int = 65536,
     use_graphs: bool = True,
@@ -196,5 +196,5 @@ class _NullStore:
     def put_metrics_csv(self, key, header, row):
         return None
 
-    def put_dataset(self, d, y, X, fmt="npz"):
+    def put_dataset(self, d, y, X, fmt="npy"):
         return None

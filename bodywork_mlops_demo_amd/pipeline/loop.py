@@ -90,7 +90,7 @@ def main(argv=None) -> None:
     p.add_argument("--model", default="linear", choices=["linear", "mlp"])
     p.add_argument("--device", default=None)
     p.add_argument("--start-date", default="2026-01-01")
-    p.add_argument("--format", default="csv", choices=["csv", "npz"])
+    p.add_argument("--format", default="csv", choices=["csv", "npy"])
     p.add_argument("--json-out", default=None)
     args = p.parse_args(argv)
     results = run_loop(

@@ -60,7 +60,7 @@ def main(argv=None) -> None:
     p.add_argument("--n", type=int, default=N_DEFAULT)
     p.add_argument("--date", default=None, help="YYYY-MM-DD (default: clock)")
     p.add_argument("--device", default=None)
-    p.add_argument("--format", default="csv", choices=["csv", "npz"])
+    p.add_argument("--format", default="csv", choices=["csv", "npy"])
     args = p.parse_args(argv)
     d = date_t.fromisoformat(args.date) if args.date else None
     with stage_guard(STAGE_NAME, exit_on_error=True):

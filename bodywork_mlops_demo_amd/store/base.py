@@ -72,10 +72,14 @@ class ArtefactStore(ABC):
             buf.write("\n".join(lines.tolist()))
             buf.write("\n")
             self.put_bytes(key, buf.getvalue().encode())
-        elif fmt == "npz":
-            key = contract.dataset_key(d, "npz")
+        elif fmt in ("npy", "npz", "bin"):
+            # raw .npy of [2, n] float32 (row 0 = y, row 1 = X): no zip
+            # container / CRC pass, so a 10M-row day persists at disk
+            # bandwidth (the .npz CRC alone cost ~as much as the write)
+            key = contract.dataset_key(d, "npy")
             bio = io.BytesIO()
-            np.savez(bio, y=y.astype(np.float32), X=X.astype(np.float32))
+            np.save(bio, np.stack([y.astype(np.float32),
+                                   X.astype(np.float32)]))
             self.put_bytes(key, bio.getvalue())
         else:
             raise ValueError(f"unknown dataset format {fmt!r}")
@@ -84,7 +88,10 @@ class ArtefactStore(ABC):
     def get_dataset(self, key: str) -> tuple[np.ndarray, np.ndarray]:
         """Load (y, X) float32 arrays from a dataset artefact."""
         raw = self.get_bytes(key)
-        if key.endswith(".npz"):
+        if key.endswith(".npy"):
+            arr = np.load(io.BytesIO(raw))
+            return arr[0], arr[1]
+        if key.endswith(".npz"):  # legacy binary artefacts
             z = np.load(io.BytesIO(raw))
             return z["y"], z["X"]
         # CSV: date,y,X header (stage_3:42); np.loadtxt is plenty here and

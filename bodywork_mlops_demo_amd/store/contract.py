@@ -36,7 +36,7 @@ def date_from_key(key: str) -> date:
 
 
 def dataset_key(d: date, fmt: str = "csv") -> str:
-    suffix = "csv" if fmt == "csv" else "npz"
+    suffix = "csv" if fmt == "csv" else "npy"
     return f"{DATASETS_PREFIX}regression-dataset-{d}.{suffix}"
 
 

@@ -47,9 +47,9 @@ def test_mlp_gpu_training_step():
     X = torch.rand(16384, device=DEV) * 100
     y = 1.0 + 0.5 * X + torch.randn_like(X) * 10
     mse0 = float(((m.predict(X) - y) ** 2).mean())
-    m.fit(X, y, steps=20, batch_size=8192, lr=1e-3)
+    m.fit(X, y, steps=40, batch_size=8192, lr=3e-3)
     mse = float(((m.predict(X) - y) ** 2).mean())
-    assert mse < mse0
+    assert mse < 0.5 * mse0, (mse, mse0)
 
 
 def test_full_cycle_gpu():
@@ -69,7 +69,7 @@ def test_cycle_with_store_gpu(tmp_path):
     from bodywork_mlops_demo_amd.store import LocalStore, contract
 
     store = LocalStore(str(tmp_path / "store"))
-    run_loop(store, days=2, n_rows=100_000, device=DEV, persist_fmt="npz",
+    run_loop(store, days=2, n_rows=100_000, device=DEV, persist_fmt="npy",
              start_date="2026-05-01")
     assert len(store.list_keys(contract.MODELS_PREFIX)) == 2
     artefact, _ = store.get_latest_model()

@@ -104,6 +104,23 @@ class TestFitAndMetrics:
         out = ops.linear_score(X, 1.0, 0.5)
         assert out.tolist() == [1.0, 26.0, 51.0]
 
+    def test_random_split_semantics(self):
+        X = torch.rand(100_000) * 100
+        y = 1 + 0.5 * X
+        Xtr, ytr, Xte, yte = ops.random_split(X, y, 0.2, seed=42)
+        n_te = Xte.shape[0]
+        assert Xtr.shape[0] + n_te == 100_000
+        assert abs(n_te - 20_000) < 600  # binomial(1e5, 0.2), ~4.7 sigma
+        # deterministic
+        _, _, Xte2, _ = ops.random_split(X, y, 0.2, seed=42)
+        assert torch.equal(Xte, Xte2)
+        # pairing preserved
+        assert torch.allclose(yte, 1 + 0.5 * Xte)
+        assert torch.allclose(ytr, 1 + 0.5 * Xtr)
+        # different seed -> different split
+        _, _, Xte3, _ = ops.random_split(X, y, 0.2, seed=43)
+        assert Xte3.shape != Xte.shape or not torch.equal(Xte3, Xte)
+
     def test_split_is_seeded_and_disjoint(self):
         tr1, te1 = ops.train_test_split_indices(100, 0.2, seed=42)
         tr2, te2 = ops.train_test_split_indices(100, 0.2, seed=42)

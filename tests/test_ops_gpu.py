@@ -60,6 +60,18 @@ class TestDatagenGPU:
         assert X.min() >= 0 and X.max() <= 100
 
 
+class TestSplitGPU:
+    def test_random_split_matches_cpu(self):
+        X = torch.rand(500_000, device=DEV) * 100
+        y = 1 + 0.5 * X
+        Xtr, ytr, Xte, yte = ops.random_split(X, y, 0.2, seed=42)
+        Xtr_c, ytr_c, Xte_c, yte_c = ops.random_split(
+            X.cpu(), y.cpu(), 0.2, seed=42)
+        assert torch.equal(Xte.cpu(), Xte_c)
+        assert torch.equal(Xtr.cpu(), Xtr_c)
+        assert torch.equal(yte.cpu(), yte_c)
+
+
 class TestLinregGPU:
     def test_stats_match_fp64_oracle(self):
         X = torch.rand(1_000_000, device=DEV) * 100

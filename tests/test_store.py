@@ -52,11 +52,11 @@ def test_dataset_roundtrip_csv(tmp_store):
     np.testing.assert_allclose(X2, X, rtol=1e-6)
 
 
-def test_dataset_roundtrip_npz(tmp_store):
+def test_dataset_roundtrip_npy(tmp_store):
     d = date(2026, 1, 2)
     y = np.random.randn(100).astype(np.float32)
     X = np.random.rand(100).astype(np.float32) * 100
-    key = tmp_store.put_dataset(d, y, X, fmt="npz")
+    key = tmp_store.put_dataset(d, y, X, fmt="npy")
     y2, X2 = tmp_store.get_dataset(key)
     np.testing.assert_array_equal(y2, y)
     np.testing.assert_array_equal(X2, X)
