@@ -25,6 +25,28 @@ class GPULinearRegressor:
         self.intercept_ = float(intercept)
         self.coef_ = float(coef)
         self.device = torch.device(device)
+        self._ab: torch.Tensor | None = None  # device-resident [a, b]
+
+    def _ab_tensor(self) -> torch.Tensor:
+        """2-element device tensor read by the scoring kernel — captured
+        graphs follow weight updates through this buffer (no recapture)."""
+        if self._ab is None or self._ab.device != self.device:
+            self._ab = torch.tensor([self.intercept_, self.coef_],
+                                    device=self.device, dtype=torch.float32)
+        return self._ab
+
+    def _sync_ab(self) -> None:
+        if self._ab is not None:
+            self._ab.copy_(torch.tensor([self.intercept_, self.coef_],
+                                        dtype=torch.float32))
+
+    def copy_weights_from(self, other: "GPULinearRegressor") -> bool:
+        """In-place weight swap (serving hot-redeploy path)."""
+        if not isinstance(other, GPULinearRegressor):
+            return False
+        self.intercept_, self.coef_ = other.intercept_, other.coef_
+        self._sync_ab()
+        return True
 
     # -- training ----------------------------------------------------------
     def fit(self, X: torch.Tensor, y: torch.Tensor, process_group=None):
@@ -41,14 +63,18 @@ class GPULinearRegressor:
 
             dist.all_reduce(stats, group=process_group)
         self.intercept_, self.coef_ = ops.solve_ols(stats.cpu())
+        self._sync_ab()
         return self
 
     # -- inference ---------------------------------------------------------
     def predict(self, X: torch.Tensor) -> torch.Tensor:
+        if X.device.type == "cuda":
+            return ops.linear_score(X, ab=self._ab_tensor())
         return ops.linear_score(X, self.intercept_, self.coef_)
 
     def to(self, device):
         self.device = torch.device(device)
+        self._ab = None
         return self
 
     # -- artefact compatibility --------------------------------------------

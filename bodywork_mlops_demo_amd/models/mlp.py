@@ -75,6 +75,18 @@ class GPUMLPRegressor:
         self._to_device()
         return self
 
+    def copy_weights_from(self, other: "GPUMLPRegressor") -> bool:
+        """In-place weight swap (serving hot-redeploy): masters and bf16
+        shadows are copied into the EXISTING tensors so captured serving
+        graphs keep valid pointers."""
+        if not isinstance(other, GPUMLPRegressor) or other.hidden != self.hidden:
+            return False
+        for name in ("w1", "b1", "W2", "b2", "w3", "b3"):
+            getattr(self, name).copy_(getattr(other, name).to(self.device))
+        for name in ("w1_bf", "b1_bf", "W2w_bf", "W2wt_bf", "b2_bf", "w3_bf"):
+            getattr(self, name).copy_(getattr(other, name).to(self.device))
+        return True
+
     def parameters(self) -> list[torch.Tensor]:
         return [self.w1, self.b1, self.W2, self.b2, self.w3, self.b3]
 

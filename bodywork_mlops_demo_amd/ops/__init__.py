@@ -147,11 +147,26 @@ def solve_ols(stats: torch.Tensor) -> tuple[float, float]:
 # scoring — reference stage_2:78 (model.predict)
 # --------------------------------------------------------------------------
 
-def linear_score(X: torch.Tensor, intercept: float, coef: float) -> torch.Tensor:
-    """Batched linear scoring ``yhat = intercept + coef * X`` (fp32)."""
+def linear_score(
+    X: torch.Tensor,
+    intercept: float | None = None,
+    coef: float | None = None,
+    ab: torch.Tensor | None = None,
+) -> torch.Tensor:
+    """Batched linear scoring ``yhat = intercept + coef * X`` (fp32).
+
+    On GPU, coefficients live in a 2-element device tensor ``ab`` read by
+    the kernel, so captured serving graphs track redeployed weights
+    without recapture; pass either ``ab`` or the python floats.
+    """
     if X.device.type == "cuda":
         core = _core(X.device)
-        return core.linear_score(X.contiguous(), intercept, coef)
+        if ab is None:
+            ab = torch.tensor([intercept, coef], device=X.device,
+                              dtype=torch.float32)
+        return core.linear_score(X.contiguous(), ab)
+    if ab is not None:
+        intercept, coef = float(ab[0]), float(ab[1])
     return reference.linear_score_cpu(X, intercept, coef)
 
 

@@ -13,8 +13,7 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor> random_split_hip(
     const at::Tensor& X, const at::Tensor& y, double test_frac, int64_t seed);
 // linreg.hip
 at::Tensor linreg_stats_hip(const at::Tensor& x, const at::Tensor& y);
-at::Tensor linear_score_hip(const at::Tensor& x, double intercept,
-                            double coef);
+at::Tensor linear_score_hip(const at::Tensor& x, const at::Tensor& ab);
 at::Tensor regression_metrics_hip(const at::Tensor& y, const at::Tensor& yhat);
 at::Tensor score_label_metrics_hip(const at::Tensor& s, const at::Tensor& l);
 // mlp_small.hip
@@ -46,7 +45,7 @@ TORCH_LIBRARY(bodywork_hip, m) {
   m.def("random_split(Tensor X, Tensor y, float test_frac, int seed) -> "
         "(Tensor, Tensor, Tensor, Tensor)");
   m.def("linreg_stats(Tensor x, Tensor y) -> Tensor");
-  m.def("linear_score(Tensor x, float intercept, float coef) -> Tensor");
+  m.def("linear_score(Tensor x, Tensor ab) -> Tensor");
   m.def("regression_metrics(Tensor y, Tensor yhat) -> Tensor");
   m.def("score_label_metrics(Tensor s, Tensor l) -> Tensor");
   m.def("expand1d_bf16(Tensor x, Tensor w, Tensor? b, bool relu, "

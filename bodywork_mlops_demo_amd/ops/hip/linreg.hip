@@ -82,8 +82,12 @@ at::Tensor linreg_stats_hip(const at::Tensor& x, const at::Tensor& y) {
 // ---- linear_score ---------------------------------------------------------
 
 __global__ void linear_score_kernel(const float* __restrict__ x,
-                                    float* __restrict__ out, float a, float b,
+                                    float* __restrict__ out,
+                                    const float* __restrict__ ab,
                                     long long n) {
+  // coefficients read from device memory so a captured hipGraph picks up
+  // redeployed models without recapture (serving hot-swap)
+  const float a = ab[0], b = ab[1];
   const long long stride = (long long)gridDim.x * blockDim.x;
   long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
   const long long n4 = n >> 2;
@@ -98,15 +102,16 @@ __global__ void linear_score_kernel(const float* __restrict__ x,
     out[j] = fmaf(b, x[j], a);
 }
 
-at::Tensor linear_score_hip(const at::Tensor& x, double intercept,
-                            double coef) {
+at::Tensor linear_score_hip(const at::Tensor& x, const at::Tensor& ab) {
   TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kFloat);
+  TORCH_CHECK(ab.is_cuda() && ab.numel() == 2 &&
+              ab.scalar_type() == at::kFloat);
   long long n = x.numel();
   auto out = at::empty_like(x);
   auto stream = at::cuda::getCurrentCUDAStream();
   hipLaunchKernelGGL(linear_score_kernel, dim3(reduce_grid(n)),
                      dim3(RED_BLOCK), 0, stream, x.data_ptr<float>(),
-                     out.data_ptr<float>(), (float)intercept, (float)coef, n);
+                     out.data_ptr<float>(), ab.data_ptr<float>(), n);
   return out;
 }
 

@@ -46,6 +46,24 @@ class CycleState:
         self.y = torch.empty(0, device=device)
         self.X = torch.empty(0, device=device)
         self._day_sizes: list[int] = []
+        self._io_pool = None
+        self._io_futures: list = []
+
+    # -- async artefact I/O: dataset writes overlap the next cycle's
+    #    compute (the D2H copy is synchronous; only the file write is
+    #    deferred).  drain before reading artefacts back or stopping a
+    #    benchmark clock.
+    def submit_io(self, fn, *args) -> None:
+        if self._io_pool is None:
+            from concurrent.futures import ThreadPoolExecutor
+
+            self._io_pool = ThreadPoolExecutor(max_workers=1)
+        self._io_futures.append(self._io_pool.submit(fn, *args))
+
+    def drain_io(self) -> None:
+        for f in self._io_futures:
+            f.result()  # re-raises write errors
+        self._io_futures.clear()
 
     def append_day(self, y: torch.Tensor, X: torch.Tensor) -> None:
         self.y = torch.cat([self.y, y])
@@ -92,8 +110,8 @@ def run_cycle(
         )
         state.append_day(y, X)
         if store is not None and state.rank == 0:
-            store.put_dataset(state.date, y.cpu().numpy(), X.cpu().numpy(),
-                              fmt=persist_fmt)
+            state.submit_io(store.put_dataset, state.date, y.cpu().numpy(),
+                            X.cpu().numpy(), persist_fmt)
 
     # -- stage 1: train on all accumulated data -----------------------------
     sync()
@@ -136,9 +154,17 @@ def run_cycle(
         joblib.dump(trained.to_sklearn(), bio)
         bio.seek(0)
         model = regressor_from_artifact(joblib.load(bio), device)
-    scorer = BatchedScorer(model, device, use_graphs=use_graphs)
+    # reuse the resident scorer across cycles when possible: weights are
+    # copied into the captured graphs' tensors (serving hot-redeploy)
+    scorer = None
     if scorer_cache is not None:
-        scorer_cache["scorer"] = scorer
+        cached = scorer_cache.get("scorer")
+        if cached is not None and cached.update_model(model):
+            scorer = cached
+    if scorer is None:
+        scorer = BatchedScorer(model, device, use_graphs=use_graphs)
+        if scorer_cache is not None:
+            scorer_cache["scorer"] = scorer
     sync()
     timings["deploy_s"] = perf_counter() - t0
 
@@ -151,8 +177,8 @@ def run_cycle(
         stream_offset=state.rank * n_rows,
     )
     if store is not None and state.rank == 0:
-        store.put_dataset(next_date, y_next.cpu().numpy(),
-                          X_next.cpu().numpy(), fmt=persist_fmt)
+        state.submit_io(store.put_dataset, next_date, y_next.cpu().numpy(),
+                        X_next.cpu().numpy(), persist_fmt)
     sync()
     timings["datagen_s"] = perf_counter() - t0
 

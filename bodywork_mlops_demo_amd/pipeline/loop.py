@@ -65,10 +65,12 @@ def run_loop(
             pass
 
     results = []
+    scorer_cache: dict = {}
     for day in range(days):
         r = run_cycle(
             state, store, n_rows, model_type=model_type,
             process_group=process_group, persist_fmt=persist_fmt,
+            scorer_cache=scorer_cache,
         )
         results.append(r)
         t = r["timings"]
@@ -79,6 +81,7 @@ def run_loop(
             f"test {t['test_s']:.3f}) online MAPE "
             f"{r['online']['MAPE']:.4f}"
         )
+    state.drain_io()
     return results
 
 

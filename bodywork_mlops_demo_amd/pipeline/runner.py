@@ -44,6 +44,9 @@ class ServiceHandle:
     stage: str
     procs: list[subprocess.Popen]
     ports: list[int]
+    cmds: list[list[str]] = field(default_factory=list)
+    envs: list[dict] = field(default_factory=list)
+    respawns: int = 0
 
     @property
     def urls(self) -> list[str]:
@@ -172,6 +175,7 @@ class PipelineRunner:
         svc = spec.service
         env_base = self._stage_env(spec)
         procs, ports = [], []
+        cmds, envs = [], []
         for r in range(svc.replicas):
             port = self.base_port + r if svc.replicas > 1 else svc.port
             env = dict(env_base)
@@ -184,7 +188,9 @@ class PipelineRunner:
                      f"(gpu={env.get('HIP_VISIBLE_DEVICES', 'cpu')})")
             procs.append(subprocess.Popen(cmd, env=env))
             ports.append(port)
-        handle = ServiceHandle(spec.name, procs, ports)
+            cmds.append(cmd)
+            envs.append(env)
+        handle = ServiceHandle(spec.name, procs, ports, cmds, envs)
         if not self._await_healthy(handle, svc.max_startup_time_seconds):
             self._stop_service(handle)
             get_error_monitor().capture_message(
@@ -217,6 +223,47 @@ class PipelineRunner:
         log.info(f"service {handle.stage}: {len(handle.ports)} replica(s) healthy "
                  f"on ports {handle.ports}")
         return True
+
+    # -- failure detection / elastic recovery ---------------
+    def watchdog_pass(self, max_respawns_per_replica: int = 3) -> int:
+        """One health sweep over running services: respawn dead replicas
+        (the k8s Deployment self-healing the reference delegates to the
+        cluster).  Returns the number of respawns performed.  Call
+        periodically (the drift loop calls it once per cycle) or from a
+        background thread."""
+        respawned = 0
+        for handle in self.services.values():
+            for i, proc in enumerate(handle.procs):
+                if proc.poll() is None:
+                    continue
+                if handle.respawns >= max_respawns_per_replica * len(handle.procs):
+                    log.error(
+                        f"service {handle.stage} replica {i} dead "
+                        f"(rc={proc.returncode}) and respawn budget spent"
+                    )
+                    continue
+                log.warning(
+                    f"service {handle.stage} replica {i} died "
+                    f"(rc={proc.returncode}); respawning"
+                )
+                get_error_monitor().capture_message(
+                    f"respawning {handle.stage} replica {i}", "warning"
+                )
+                handle.procs[i] = subprocess.Popen(handle.cmds[i],
+                                                   env=handle.envs[i])
+                handle.respawns += 1
+                respawned += 1
+        return respawned
+
+    def inject_replica_failure(self, stage: str, replica: int = 0) -> None:
+        """Fault-injection hook: kill one live replica (used by tests and
+        chaos drills to verify the watchdog + client retry path)."""
+        handle = self.services[stage]
+        proc = handle.procs[replica]
+        if proc.poll() is None:
+            proc.kill()
+            proc.wait(timeout=10)
+            log.info(f"injected failure: killed {stage} replica {replica}")
 
     @staticmethod
     def _stop_service(handle: ServiceHandle) -> None:

@@ -38,14 +38,25 @@ class BatchedScorer:
                 return b
         return n
 
+    def update_model(self, new_model) -> bool:
+        """Hot-redeploy: copy new weights into the resident model's
+        tensors so every captured graph stays valid (no recapture, no
+        warmup forwards).  Returns False when incompatible (caller should
+        build a fresh scorer)."""
+        copy = getattr(self.model, "copy_weights_from", None)
+        if copy is not None and copy(new_model.to(self.device)):
+            return True
+        return False
+
     def _capture(self, b: int):
         x_static = torch.zeros(b, device=self.device, dtype=torch.float32)
-        # warm up the kernels on a side stream before capture
+        # warm up kernel/launch state on a side stream before capture —
+        # a SMALL batch suffices (warmup exists for lazy init, not shape)
         s = torch.cuda.Stream()
         s.wait_stream(torch.cuda.current_stream())
         with torch.cuda.stream(s):
             for _ in range(2):
-                y_static = self.model.predict(x_static)
+                self.model.predict(x_static[: min(b, 1024)])
         torch.cuda.current_stream().wait_stream(s)
         graph = torch.cuda.CUDAGraph()
         with torch.cuda.graph(graph):

@@ -24,6 +24,7 @@ def test_cycle_semantics(tmp_store):
 
     r = run_cycle(state, tmp_store, n_rows=800, model_type="linear",
                   persist_fmt="csv")
+    state.drain_io()
     # model trained on day 1, tested on day 2's data
     assert tmp_store.exists(contract.model_key(date(2026, 1, 1)))
     assert tmp_store.exists(contract.dataset_key(date(2026, 1, 2)))
