@@ -103,6 +103,136 @@ __device__ __forceinline__ void stage_transposed(
 #define EPI_BIAS_RELU 1  // +bias then relu (bias may be null -> relu only)
 #define EPI_MASK 2       // multiply by (mask > 0)
 
+// shared epilogue: write the 4x4 fragment accumulator block
+template <int EPI, bool HAS_BIAS, bool OUT_FP32>
+__device__ __forceinline__ void write_epilogue(
+    f32x4 (&acc)[4][4], const float* __restrict__ bias,
+    const bf16_t* __restrict__ mask, void* __restrict__ C, long long M,
+    long long N, long long m0, long long n0, int wm, int wn, int fl, int kg) {
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      long long col = n0 + wn * 64 + j * 16 + fl;
+      if (col >= N) continue;
+      float bval = (EPI == EPI_BIAS_RELU && HAS_BIAS) ? bias[col] : 0.0f;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        long long row = m0 + wm * 64 + i * 16 + kg * 4 + r;
+        if (row >= M) continue;
+        float v = acc[i][j][r];
+        if (EPI == EPI_BIAS_RELU) {
+          v += bval;
+          v = fmaxf(v, 0.0f);
+        } else if (EPI == EPI_MASK) {
+          float mv = bf16_to_f32(mask[row * N + col]);
+          v = mv > 0.0f ? v : 0.0f;
+        }
+        if (OUT_FP32)
+          ((float*)C)[row * N + col] = v;
+        else
+          ((bf16_t*)C)[row * N + col] = f32_to_bf16(v);
+      }
+    }
+  }
+}
+
+// ---- glds-staged NT kernel (the hot path) --------------------------------
+// global_load_lds dwordx4 staging straight into a lane-linear LDS image;
+// the bank-conflict XOR swizzle lives on the SOURCE address and the
+// fragment-read address (both-sides rule): LDS chunk (row, c) holds the
+// logical 16-B chunk (row, c ^ (row & 7)).  Two LDS buffers, one
+// vmcnt(0)+barrier per K-tile (the guide's "step 3" structure).
+
+__device__ __forceinline__ void glds16(const void* gsrc, void* lds_dst) {
+  __builtin_amdgcn_global_load_lds(
+      (const __attribute__((address_space(1))) unsigned int*)gsrc,
+      (__attribute__((address_space(3))) unsigned int*)lds_dst, 16, 0, 0);
+}
+
+// stage a [128 x 64] bf16 tile: 16 wave-level 1-KiB glds per tile
+__device__ __forceinline__ void stage_tile_glds(
+    short* __restrict__ lds_tile, const bf16_t* __restrict__ src,
+    long long ld, long long row0, long long row_max, long long k0) {
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+#pragma unroll
+  for (int p = 0; p < 4; ++p) {
+    int ci0 = (p * 4 + wave) * 64;
+    int ci = ci0 + lane;
+    int row = ci >> 3;
+    int sc = (ci & 7) ^ (row & 7);  // inverse source swizzle
+    long long gr = row0 + row;
+    if (gr >= row_max) gr = row_max - 1;  // clamp: junk rows masked later
+    glds16(src + gr * ld + k0 + sc * 8, (char*)lds_tile + ci0 * 16);
+  }
+}
+
+template <int EPI, bool HAS_BIAS, bool OUT_FP32>
+__launch_bounds__(GEMM_THREADS)
+__global__ void gemm_nt_glds_kernel(
+    const bf16_t* __restrict__ A,  // [M,K]
+    const bf16_t* __restrict__ B,  // [N,K]
+    const float* __restrict__ bias, const bf16_t* __restrict__ mask,
+    void* __restrict__ C, long long M, long long N, long long K) {
+  __shared__ short lds_all[2 * 2 * 128 * 64];  // [buf][A/B][128][64]
+  const long long m0 = (long long)blockIdx.y * BM;
+  const long long n0 = (long long)blockIdx.x * BN;
+  const int wave = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int wm = wave >> 1, wn = wave & 1;
+  const int fl = lane & 15;
+  const int kg = lane >> 4;
+  const int swz = fl & 7;  // read-side XOR factor (row & 7 == fl & 7)
+
+  f32x4 acc[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  const int TILE = 128 * 64;  // shorts per tile
+
+  stage_tile_glds(lds_all, A, K, m0, M, 0);
+  stage_tile_glds(lds_all + 2 * TILE, B, K, n0, N, 0);
+  __syncthreads();  // carries the vmcnt(0) for the in-flight glds
+
+  int cur = 0;
+  for (long long k0 = 0; k0 < K; k0 += BK) {
+    short* as_cur = lds_all + cur * TILE;
+    short* bs_cur = lds_all + 2 * TILE + cur * TILE;
+    if (k0 + BK < K) {
+      stage_tile_glds(lds_all + (cur ^ 1) * TILE, A, K, m0, M, k0 + BK);
+      stage_tile_glds(lds_all + 2 * TILE + (cur ^ 1) * TILE, B, K, n0, N,
+                      k0 + BK);
+    }
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+      bf16x8_v a_frag[4], b_frag[4];
+      const int cidx = ((ks * 4 + kg) ^ swz) * 8;
+#pragma unroll
+      for (int f = 0; f < 4; ++f) {
+        a_frag[f] =
+            ((const lds_vec*)(as_cur + (wm * 64 + f * 16 + fl) * 64 + cidx))
+                ->v;
+        b_frag[f] =
+            ((const lds_vec*)(bs_cur + (wn * 64 + f * 16 + fl) * 64 + cidx))
+                ->v;
+      }
+#pragma unroll
+      for (int i = 0; i < 4; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a_frag[i], b_frag[j], acc[i][j], 0, 0, 0);
+    }
+    __syncthreads();  // drains next tile's glds; makes buffers reusable
+    cur ^= 1;
+  }
+  write_epilogue<EPI, HAS_BIAS, OUT_FP32>(acc, bias, mask, C, M, N, m0, n0,
+                                          wm, wn, fl, kg);
+}
+
 template <bool TN, int EPI, bool HAS_BIAS, bool OUT_FP32>
 __launch_bounds__(GEMM_THREADS)
 __global__ void gemm_bf16_kernel(
@@ -199,6 +329,28 @@ static void launch_gemm(bool tn, int epi, bool has_bias, bool out_fp32,
   const bf16_t* ap = (const bf16_t*)a.data_ptr();
   const bf16_t* bp = (const bf16_t*)b.data_ptr();
   void* cp = c.data_ptr();
+
+  // hot path: NT with K a tile multiple -> glds-staged kernel
+  if (!tn && K % BK == 0 && K > 0) {
+#define G_GLDS(EPI_, HB_, OF_)                                              \
+  hipLaunchKernelGGL((gemm_nt_glds_kernel<EPI_, HB_, OF_>), grid,           \
+                     dim3(GEMM_THREADS), 0, stream, ap, bp, bias, mask, cp, \
+                     M, N, K)
+    if (epi == EPI_BIAS_RELU) {
+      if (has_bias) { if (out_fp32) G_GLDS(EPI_BIAS_RELU, true, true);
+                      else          G_GLDS(EPI_BIAS_RELU, true, false); }
+      else          { if (out_fp32) G_GLDS(EPI_BIAS_RELU, false, true);
+                      else          G_GLDS(EPI_BIAS_RELU, false, false); }
+    } else if (epi == EPI_MASK) {
+      if (out_fp32) G_GLDS(EPI_MASK, false, true);
+      else          G_GLDS(EPI_MASK, false, false);
+    } else {
+      if (out_fp32) G_GLDS(EPI_NONE, false, true);
+      else          G_GLDS(EPI_NONE, false, false);
+    }
+#undef G_GLDS
+    return;
+  }
 
 #define G_LAUNCH(TN_, EPI_, HB_, OF_)                                       \
   hipLaunchKernelGGL((gemm_bf16_kernel<TN_, EPI_, HB_, OF_>), grid,         \
