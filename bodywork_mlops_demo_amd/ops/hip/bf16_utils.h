@@ -36,6 +36,14 @@ __device__ __forceinline__ bf16x8 f32_to_bf16x8(const float* in) {
   return v;
 }
 
+// MFMA operand fragment: 8 bf16 in one 16-B vector
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8_v;
+
+union lds_vec {
+  bf16x8_v v;
+  short s[8];
+};
+
 __device__ __forceinline__ float bf16_to_f32(bf16_t v) {
   return __bfloat162float(v);
 }

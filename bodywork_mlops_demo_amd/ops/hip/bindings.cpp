@@ -38,6 +38,7 @@ void adam_step_hip(at::Tensor p, const at::Tensor& g, at::Tensor m,
                    double lr, double beta1, double beta2, double eps,
                    int64_t t);
 at::Tensor transpose_to_bf16_hip(const at::Tensor& src);
+at::Tensor transpose_bf16_hip(const at::Tensor& src);
 
 TORCH_LIBRARY(bodywork_hip, m) {
   m.def("datagen(int n, int seed, int stream_offset, float alpha, float beta, "
@@ -60,6 +61,7 @@ TORCH_LIBRARY(bodywork_hip, m) {
         "Tensor(d!)? p_bf16, float lr, float beta1, float beta2, float eps, "
         "int t) -> ()");
   m.def("transpose_to_bf16(Tensor src) -> Tensor");
+  m.def("transpose_bf16(Tensor src) -> Tensor");
 }
 
 TORCH_LIBRARY_IMPL(bodywork_hip, CUDA, m) {
@@ -77,6 +79,7 @@ TORCH_LIBRARY_IMPL(bodywork_hip, CUDA, m) {
   m.impl("gemm_tn_bf16", gemm_tn_bf16_hip);
   m.impl("adam_step", adam_step_hip);
   m.impl("transpose_to_bf16", transpose_to_bf16_hip);
+  m.impl("transpose_bf16", transpose_bf16_hip);
 }
 
 // datagen takes no tensor argument, so the dispatcher cannot route it by

@@ -31,13 +31,7 @@
 #define LDS_STRIDE 72  // BK + 8 shorts pad
 #define GEMM_THREADS 256
 
-typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8_v;
 typedef __attribute__((ext_vector_type(4))) float f32x4;
-
-union lds_vec {
-  bf16x8_v v;
-  short s[8];
-};
 
 // stage a [ROWS x BK] K-major tile (row-major source, K contiguous):
 // thread t loads 16 B chunks; zero-fill outside (r1, k1) bounds.

@@ -93,6 +93,69 @@ __global__ void transpose_to_bf16_kernel(const float* __restrict__ src,
   }
 }
 
+// 64x64 LDS-tiled bf16->bf16 transpose, 16-B loads and stores.
+// Feeds the dW = dY^T @ X shape to the fast NT GEMM: two bandwidth-bound
+// transposes (~0.5 ms for a 65536x4096 operand pair) instead of the
+// scalar transpose-staged TN kernel (7x slower at H=4096).
+__global__ void transpose_bf16_kernel(const bf16_t* __restrict__ src,
+                                      bf16_t* __restrict__ dst,
+                                      long long rows, long long cols) {
+  __shared__ short tile[64][72];  // +8 pad: conflict-free column reads
+  const long long r0 = (long long)blockIdx.y * 64;
+  const long long c0 = (long long)blockIdx.x * 64;
+  const int tc = threadIdx.x & 7;   // 16-B chunk within the row
+  const int tr = threadIdx.x >> 3;  // row within the tile (32 per pass)
+#pragma unroll
+  for (int p = 0; p < 2; ++p) {
+    long long r = r0 + tr + p * 32;
+    long long c = c0 + tc * 8;
+    lds_vec v;
+    if (r < rows && c + 7 < cols) {
+      v = *(const lds_vec*)(src + r * cols + c);
+    } else {
+#pragma unroll
+      for (int e = 0; e < 8; ++e)
+        v.s[e] = (r < rows && c + e < cols)
+                     ? ((const short*)src)[r * cols + c + e]
+                     : (short)0;
+    }
+    *(lds_vec*)&tile[tr + p * 32][tc * 8] = v;
+  }
+  __syncthreads();
+  // write transposed: thread handles out row (= src col) c0+tr(+32p),
+  // 8 consecutive out cols (= src rows) r0+tc*8..
+#pragma unroll
+  for (int p = 0; p < 2; ++p) {
+    long long oc = c0 + tr + p * 32;   // out row = src col
+    long long orow0 = r0 + tc * 8;     // out col base = src row
+    if (oc >= cols) continue;
+    lds_vec v;
+#pragma unroll
+    for (int e = 0; e < 8; ++e) v.s[e] = tile[tc * 8 + e][tr + p * 32];
+    if (orow0 + 7 < rows) {
+      *(lds_vec*)(dst + oc * rows + orow0) = v;
+    } else {
+#pragma unroll
+      for (int e = 0; e < 8; ++e)
+        if (orow0 + e < rows)
+          ((short*)dst)[oc * rows + orow0 + e] = v.s[e];
+    }
+  }
+}
+
+at::Tensor transpose_bf16_hip(const at::Tensor& src) {
+  TORCH_CHECK(src.is_cuda() && src.dim() == 2 &&
+              src.scalar_type() == at::kBFloat16);
+  long long rows = src.size(0), cols = src.size(1);
+  auto dst = at::empty({cols, rows}, src.options());
+  auto stream = at::cuda::getCurrentCUDAStream();
+  dim3 grid((cols + 63) / 64, (rows + 63) / 64);
+  hipLaunchKernelGGL(transpose_bf16_kernel, grid, dim3(256), 0, stream,
+                     (const bf16_t*)src.data_ptr(), (bf16_t*)dst.data_ptr(),
+                     rows, cols);
+  return dst;
+}
+
 at::Tensor transpose_to_bf16_hip(const at::Tensor& src) {
   TORCH_CHECK(src.is_cuda() && src.dim() == 2 &&
               src.scalar_type() == at::kFloat);

@@ -65,6 +65,25 @@ class CycleState:
             f.result()  # re-raises write errors
         self._io_futures.clear()
 
+    def fast_cpu(self, name: str, t: torch.Tensor):
+        """Device->host via a cached pinned staging buffer (~10x the
+        pageable-copy rate).  The returned numpy view aliases the pinned
+        buffer, which is only reused after drain_io() — submit_io callers
+        drain before the next copy."""
+        if t.device.type != "cuda":
+            return t.numpy()
+        buf = getattr(self, "_pin", None)
+        if buf is None:
+            self._pin = {}
+        cur = self._pin.get(name)
+        if cur is None or cur.numel() < t.numel():
+            cur = torch.empty(t.numel(), dtype=t.dtype, pin_memory=True)
+            self._pin[name] = cur
+        view = cur[: t.numel()]
+        view.copy_(t, non_blocking=True)
+        torch.cuda.synchronize()
+        return view.numpy()
+
     def append_day(self, y: torch.Tensor, X: torch.Tensor) -> None:
         self.y = torch.cat([self.y, y])
         self.X = torch.cat([self.X, X])
@@ -110,8 +129,10 @@ def run_cycle(
         )
         state.append_day(y, X)
         if store is not None and state.rank == 0:
-            state.submit_io(store.put_dataset, state.date, y.cpu().numpy(),
-                            X.cpu().numpy(), persist_fmt)
+            state.drain_io()  # pinned staging buffers about to be reused
+            state.submit_io(store.put_dataset, state.date,
+                            state.fast_cpu("y", y), state.fast_cpu("X", X),
+                            persist_fmt)
 
     # -- stage 1: train on all accumulated data -----------------------------
     sync()
@@ -177,8 +198,10 @@ def run_cycle(
         stream_offset=state.rank * n_rows,
     )
     if store is not None and state.rank == 0:
-        state.submit_io(store.put_dataset, next_date, y_next.cpu().numpy(),
-                        X_next.cpu().numpy(), persist_fmt)
+        state.drain_io()  # pinned staging buffers about to be reused
+        state.submit_io(store.put_dataset, next_date,
+                        state.fast_cpu("y", y_next),
+                        state.fast_cpu("X", X_next), persist_fmt)
     sync()
     timings["datagen_s"] = perf_counter() - t0
 

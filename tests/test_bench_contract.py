@@ -1,0 +1,62 @@
+"""bench.py driver-contract tests: single-rank and torchrun world=2 (CPU).
+
+The driver launches bench.py via `python -m torch.distributed.run
+--nnodes=1 --nproc-per-node N --master-addr 127.0.0.1 ...` — this test
+protects that exact invocation on the gloo/CPU path.
+"""
+import json
+import os
+import subprocess
+import sys
+
+import pytest
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def _parse_json_line(stdout: str) -> dict:
+    for line in reversed(stdout.strip().splitlines()):
+        if line.startswith("{"):
+            return json.loads(line)
+    raise AssertionError(f"no JSON line in bench output:\n{stdout[-2000:]}")
+
+
+REQUIRED_KEYS = {
+    "metric", "value", "unit", "n_gpus", "steps", "warmup", "ms_per_step",
+    "higher_is_better", "scaling", "vs_baseline", "dtype", "data", "config",
+}
+
+
+@pytest.mark.timeout(600)
+def test_bench_single_rank_cpu(tmp_path):
+    proc = subprocess.run(
+        [sys.executable, "bench.py", "--rows", "20000", "--steps", "1",
+         "--warmup", "0", "--store", str(tmp_path / "store")],
+        cwd=REPO, capture_output=True, text=True, timeout=540,
+    )
+    assert proc.returncode == 0, proc.stderr[-2000:]
+    d = _parse_json_line(proc.stdout)
+    assert REQUIRED_KEYS <= set(d)
+    assert d["n_gpus"] == 1 and d["steps"] == 1
+    assert d["value"] > 0 and d["ms_per_step"] > 0
+    assert d["scaling"] == "weak"
+    assert d["config"]["parallelism"] == "dp1"
+
+
+@pytest.mark.timeout(600)
+def test_bench_torchrun_world2_cpu(tmp_path):
+    env = dict(os.environ, MASTER_ADDR="127.0.0.1")
+    proc = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29631", "bench.py", "--gpus", "2",
+         "--rows", "20000", "--steps", "1", "--warmup", "0",
+         "--store", str(tmp_path / "store")],
+        cwd=REPO, capture_output=True, text=True, timeout=540, env=env,
+    )
+    assert proc.returncode == 0, proc.stderr[-3000:]
+    d = _parse_json_line(proc.stdout)
+    assert d["n_gpus"] == 2
+    assert d["config"]["parallelism"] == "dp2"
+    # whole-job rows/sec over 2 ranks of 20k rows each
+    assert d["value"] > 0
