@@ -199,6 +199,27 @@ class TestGemmGPU:
         want = a.float().t() @ b.float()
         assert self._relerr(got.cpu(), want.cpu()) < 2e-2
 
+    @pytest.mark.parametrize("r,c", [(256, 512), (4096, 4096), (1000, 130)])
+    def test_transpose_bf16(self, r, c):
+        src = torch.randn(r, c, device=DEV).bfloat16()
+        got = ops.transpose_bf16(src)
+        assert torch.equal(got.cpu(), src.cpu().t().contiguous())
+
+    def test_adam_step_matches_cpu(self):
+        torch.manual_seed(0)
+        p = torch.randn(10000, device=DEV)
+        g = torch.randn(10000, device=DEV)
+        m = torch.zeros_like(p)
+        v = torch.zeros_like(p)
+        shadow = torch.zeros(10000, device=DEV).bfloat16()
+        pc, gc, mc, vc = p.cpu().clone(), g.cpu(), m.cpu().clone(), v.cpu().clone()
+        ops.adam_step(p, g, m, v, shadow, lr=1e-3, t=3)
+        reference.adam_step_cpu(pc, gc, mc, vc, None, 1e-3, 3, 0.9, 0.999,
+                                1e-8)
+        torch.testing.assert_close(p.cpu(), pc, rtol=1e-5, atol=1e-7)
+        torch.testing.assert_close(shadow.cpu().float(), pc.bfloat16().float(),
+                                   rtol=1e-2, atol=1e-2)
+
     def test_out_fp32(self):
         x = (torch.randn(128, 64, device=DEV)).bfloat16()
         w = (torch.randn(128, 64, device=DEV)).bfloat16()
