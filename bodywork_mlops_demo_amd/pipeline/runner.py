@@ -124,6 +124,32 @@ class PipelineRunner:
                 )
         return env
 
+    # -- per-stage requirements validation ---------------------------------
+    # The reference installs a per-stage pip list at container start
+    # (bodywork.yaml:10-16).  This runner executes in one pinned
+    # environment, so declared requirements are VALIDATED instead:
+    # missing packages or version mismatches are logged (and counted)
+    # before the stage runs.
+    @staticmethod
+    def check_requirements(spec: StageSpec) -> list[str]:
+        import importlib.metadata as md
+        import re as _re
+
+        problems = []
+        for req in spec.requirements:
+            m = _re.match(r"^([A-Za-z0-9_.-]+)\s*(==|>=|<=|~=)?\s*(.*)$", req)
+            if not m:
+                continue
+            name, op, want = m.group(1), m.group(2), m.group(3).strip()
+            try:
+                have = md.version(name)
+            except md.PackageNotFoundError:
+                problems.append(f"{req}: package not installed")
+                continue
+            if op == "==" and want and have != want:
+                problems.append(f"{req}: installed {have}")
+        return problems
+
     # -- stage execution ---------------------------------------------------
     def _module_cmd(self, spec: StageSpec) -> list[str]:
         path = spec.executable_module_path
@@ -135,6 +161,8 @@ class PipelineRunner:
         assert spec.batch is not None
         retries = spec.batch.retries
         timeout = spec.batch.max_completion_time_seconds
+        for problem in self.check_requirements(spec):
+            log.warning(f"stage {spec.name} requirement check: {problem}")
         env = self._stage_env(spec)
         cmd = self._module_cmd(spec)
         for attempt in range(retries + 1):

@@ -72,6 +72,20 @@ def create_app(store: ArtefactStore, device: str = "cpu",
             }
         )
 
+    @app.post("/score/v1/binary")
+    async def score_binary(request: Request) -> Response:
+        """High-throughput wire: raw little-endian float32 X in the body,
+        raw float32 predictions back (no JSON float formatting/parsing on
+        the hot path — the batch-JSON endpoint spends more time in number
+        parsing than the GPU does scoring)."""
+        body = await request.body()
+        X = np.frombuffer(body, dtype=np.float32)
+        preds = state["scorer"].score(X)
+        return Response(content=preds.astype(np.float32).tobytes(),
+                        media_type="application/octet-stream",
+                        headers={"X-Model-Info": state["model_info"],
+                                 "X-N": str(preds.shape[0])})
+
     @app.get("/healthz")
     async def healthz() -> Response:
         ok = "scorer" in state

@@ -85,6 +85,33 @@ def _score_batch(
     return scores, times
 
 
+def _score_binary(
+    url: str, X: np.ndarray, chunk: int = 1 << 20
+) -> tuple[np.ndarray, np.ndarray]:
+    """Binary batch client: raw float32 in/out against /score/v1/binary."""
+    import requests
+
+    bin_url = url.rstrip("/") + "/binary"
+    session = requests.Session()
+    session.mount(bin_url, requests.adapters.HTTPAdapter(max_retries=3))
+    scores = np.empty(X.shape[0], dtype=np.float32)
+    times = np.empty(X.shape[0])
+    for lo in range(0, X.shape[0], chunk):
+        hi = min(lo + chunk, X.shape[0])
+        t0 = perf_counter()
+        resp = session.post(
+            bin_url, data=X[lo:hi].astype(np.float32).tobytes(),
+            headers={"Content-Type": "application/octet-stream"}, timeout=300,
+        )
+        dt = perf_counter() - t0
+        if resp.ok:
+            scores[lo:hi] = np.frombuffer(resp.content, dtype=np.float32)
+        else:
+            scores[lo:hi] = -1
+        times[lo:hi] = dt / (hi - lo)
+    return scores, times
+
+
 def run(
     store: ArtefactStore,
     url: str = DEFAULT_URL,
@@ -123,6 +150,8 @@ def run(
             scores_np, times = _score_serial(url, X_np)
         elif mode == "batch":
             scores_np, times = _score_batch(url, X_np)
+        elif mode == "binary":
+            scores_np, times = _score_binary(url, X_np)
         else:
             raise ValueError(f"unknown client mode {mode!r}")
         mean_rt = float(times.mean())
@@ -156,7 +185,8 @@ def main(argv=None) -> None:
         "--url",
         default=os.environ.get("BODYWORK_AMD_SERVICE_URL", DEFAULT_URL),
     )
-    p.add_argument("--mode", default="batch", choices=["serial", "batch"])
+    p.add_argument("--mode", default="batch",
+                   choices=["serial", "batch", "binary"])
     p.add_argument("--device", default=None)
     args = p.parse_args(argv)
     with stage_guard(STAGE_NAME, exit_on_error=True):

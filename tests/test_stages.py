@@ -80,6 +80,15 @@ def test_serving_http_wire_format(seeded_store):
         r = client.post("/score/v1", json={"X": [10.0, 20.0]})
         assert len(r.json()["prediction"]) == 2
 
+        # binary wire: raw float32 in/out
+        X = np.array([10.0, 20.0, 30.0], dtype=np.float32)
+        r = client.post("/score/v1/binary", content=X.tobytes(),
+                        headers={"Content-Type": "application/octet-stream"})
+        assert r.status_code == 200
+        preds = np.frombuffer(r.content, dtype=np.float32)
+        assert preds.shape == (3,)
+        assert r.headers["X-N"] == "3"
+
         r = client.get("/healthz")
         assert r.json()["status"] == "ok"
 
