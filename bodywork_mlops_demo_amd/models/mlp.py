@@ -128,6 +128,17 @@ class GPUMLPRegressor:
             import torch.distributed as dist
 
             world = dist.get_world_size(process_group)
+
+        # single-GPU path: the ENTIRE Adam step (philox batch sampling,
+        # gather, forward, backward, fused Adam + shadow refresh) is one
+        # captured hipGraph, replayed per step — host dispatch of the ~30
+        # kernels was costing as much as the kernels themselves
+        if (process_group is None and self.device.type == "cuda"
+                and steps >= 4 and ops.hip_available()):
+            self._fit_captured(X.to(self.device), y.to(self.device), steps,
+                               min(batch_size, n), lr, seed)
+            return self
+
         for _ in range(steps):
             idx = torch.randint(0, n, (min(batch_size, n),), generator=g)
             xb = X[idx.to(X.device)]
@@ -146,6 +157,43 @@ class GPUMLPRegressor:
             t += 1
             self._adam_update(grads, lr, t)
         return self
+
+    def _fit_captured(self, X, y, steps: int, bs: int, lr: float, seed: int):
+        """hipGraph-captured Adam steps (see fit()).  Bias correction is a
+        device [2]-tensor the graph reads; the philox batch counter
+        advances on-device, so each replay trains on a fresh minibatch."""
+        n = X.shape[0]
+        ctr = torch.zeros(1, dtype=torch.int64, device=self.device)
+        bc = torch.ones(2, dtype=torch.float32, device=self.device)
+        beta1, beta2 = 0.9, 0.999
+
+        def one_step():
+            idx = ops.batch_indices(ctr, n, bs, seed)
+            xb = X.index_select(0, idx)
+            yb = y.index_select(0, idx)
+            grads = self._step_grads(xb, yb)
+            shadows = [self.w1_bf, self.b1_bf, self.W2wt_bf, self.b2_bf,
+                       self.w3_bf, None]
+            for p, gr, (m, v), s in zip(self.parameters(), grads,
+                                        self._opt_state, shadows):
+                ops.adam_step(p, gr, m, v, s, lr, t=1, bc=bc)
+            # refresh the transposed forward copy in place (graph-safe)
+            self.W2w_bf.copy_(ops.transpose_bf16(self.W2wt_bf))
+
+        # warmup on a side stream, then capture
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            bc.copy_(torch.tensor([1 / (1 - beta1), 1 / (1 - beta2)]))
+            one_step()
+        torch.cuda.current_stream().wait_stream(s)
+        graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(graph):
+            one_step()
+        for t in range(2, steps + 1):
+            bc.copy_(torch.tensor([1 / (1 - beta1**t), 1 / (1 - beta2**t)],
+                                  dtype=torch.float32), non_blocking=True)
+            graph.replay()
 
     def _step_grads(self, xb: torch.Tensor, yb: torch.Tensor) -> list[torch.Tensor]:
         nb = xb.shape[0]

@@ -394,17 +394,30 @@ def adam_step(
     beta1: float = 0.9,
     beta2: float = 0.999,
     eps: float = 1e-8,
+    bc: torch.Tensor | None = None,
 ) -> None:
     """Fused in-place Adam update, optionally writing the bf16 shadow
-    weight in the same pass (one HBM traversal per parameter per step)."""
+    weight in the same pass (one HBM traversal per parameter per step).
+    ``bc`` (device tensor [1/(1-b1^t), 1/(1-b2^t)]) overrides the host
+    bias correction so a captured training-step graph stays valid across
+    steps."""
     if p.device.type == "cuda":
         core = _core(p.device)
         core.adam_step(p.view(-1), g.reshape(-1).float(), m.view(-1),
                        v.view(-1),
                        p_bf16.view(-1) if p_bf16 is not None else None,
-                       lr, beta1, beta2, eps, t)
+                       lr, beta1, beta2, eps, t, bc)
         return
     reference.adam_step_cpu(p, g, m, v, p_bf16, lr, t, beta1, beta2, eps)
+
+
+def batch_indices(
+    ctr: torch.Tensor, n_data: int, bs: int, seed: int
+) -> torch.Tensor:
+    """Device-side philox minibatch sampling; ``ctr`` (int64 [1], device)
+    advances on-device so a captured step resamples on every replay."""
+    core = _core(ctr.device)
+    return core.batch_indices(ctr, n_data, bs, seed)
 
 
 def transpose_to_bf16(src: torch.Tensor) -> torch.Tensor:

@@ -36,7 +36,9 @@ at::Tensor gemm_tn_bf16_hip(const at::Tensor& a, const at::Tensor& b,
 void adam_step_hip(at::Tensor p, const at::Tensor& g, at::Tensor m,
                    at::Tensor v, const c10::optional<at::Tensor>& p_bf16,
                    double lr, double beta1, double beta2, double eps,
-                   int64_t t);
+                   int64_t t, const c10::optional<at::Tensor>& bc);
+at::Tensor batch_indices_hip(at::Tensor ctr, int64_t n_data, int64_t bs,
+                             int64_t seed);
 at::Tensor transpose_to_bf16_hip(const at::Tensor& src);
 at::Tensor transpose_bf16_hip(const at::Tensor& src);
 
@@ -59,7 +61,9 @@ TORCH_LIBRARY(bodywork_hip, m) {
   m.def("gemm_tn_bf16(Tensor a, Tensor b, bool out_fp32) -> Tensor");
   m.def("adam_step(Tensor(a!) p, Tensor g, Tensor(b!) m, Tensor(c!) v, "
         "Tensor(d!)? p_bf16, float lr, float beta1, float beta2, float eps, "
-        "int t) -> ()");
+        "int t, Tensor? bc) -> ()");
+  m.def("batch_indices(Tensor(a!) ctr, int n_data, int bs, int seed) "
+        "-> Tensor");
   m.def("transpose_to_bf16(Tensor src) -> Tensor");
   m.def("transpose_bf16(Tensor src) -> Tensor");
 }
@@ -78,6 +82,7 @@ TORCH_LIBRARY_IMPL(bodywork_hip, CUDA, m) {
   m.impl("linear_bf16", linear_bf16_hip);
   m.impl("gemm_tn_bf16", gemm_tn_bf16_hip);
   m.impl("adam_step", adam_step_hip);
+  m.impl("batch_indices", batch_indices_hip);
   m.impl("transpose_to_bf16", transpose_to_bf16_hip);
   m.impl("transpose_bf16", transpose_bf16_hip);
 }

@@ -12,13 +12,19 @@
 #include <torch/extension.h>
 
 #include "bf16_utils.h"
+#include "philox.h"
 
 template <bool WRITE_BF16>
 __global__ void adam_step_kernel(float* __restrict__ p, const float* __restrict__ g,
                                  float* __restrict__ m, float* __restrict__ v,
                                  bf16_t* __restrict__ p_bf16, long long n,
                                  float lr, float beta1, float beta2, float eps,
-                                 float inv_bc1, float inv_bc2) {
+                                 float inv_bc1, float inv_bc2,
+                                 const float* __restrict__ bc /* nullable */) {
+  if (bc) {  // device-side bias correction: hipGraph-replayable steps
+    inv_bc1 = bc[0];
+    inv_bc2 = bc[1];
+  }
   const long long stride = (long long)gridDim.x * blockDim.x;
   for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += stride) {
@@ -35,16 +41,58 @@ __global__ void adam_step_kernel(float* __restrict__ p, const float* __restrict_
   }
 }
 
+// device-side minibatch sampling: philox-keyed indices + a step counter
+// that advances on device, so a captured training step resamples on every
+// graph replay (no host RNG, no H2D index copy)
+__global__ void batch_indices_kernel(int64_t* __restrict__ out,
+                                     unsigned long long* __restrict__ ctr,
+                                     long long n_data, long long bs,
+                                     unsigned int key0, unsigned int key1) {
+  const unsigned long long step = *ctr;
+  const long long stride = (long long)gridDim.x * blockDim.x;
+  for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+       i < bs; i += stride) {
+    Philox4 r = philox4x32(step * (unsigned long long)bs + i, key0, key1);
+    out[i] = (int64_t)(((unsigned long long)r.x << 32 | r.y) %
+                       (unsigned long long)n_data);
+  }
+}
+
+__global__ void bump_counter_kernel(unsigned long long* ctr) { ++*ctr; }
+
+at::Tensor batch_indices_hip(at::Tensor ctr, int64_t n_data, int64_t bs,
+                             int64_t seed) {
+  TORCH_CHECK(ctr.is_cuda() && ctr.numel() == 1 &&
+              ctr.scalar_type() == at::kLong);
+  auto out = at::empty({bs}, ctr.options());
+  auto stream = at::cuda::getCurrentCUDAStream();
+  unsigned int key0 = (unsigned int)(seed & 0xFFFFFFFFll);
+  int grid = (int)std::min<int64_t>((bs + 255) / 256, 1024);
+  hipLaunchKernelGGL(batch_indices_kernel, dim3(grid), dim3(256), 0, stream,
+                     out.data_ptr<int64_t>(),
+                     (unsigned long long*)ctr.data_ptr(), n_data, bs, key0,
+                     0xB5297A4Du);
+  hipLaunchKernelGGL(bump_counter_kernel, dim3(1), dim3(1), 0, stream,
+                     (unsigned long long*)ctr.data_ptr());
+  return out;
+}
+
 void adam_step_hip(at::Tensor p, const at::Tensor& g, at::Tensor m,
                    at::Tensor v, const c10::optional<at::Tensor>& p_bf16,
                    double lr, double beta1, double beta2, double eps,
-                   int64_t t) {
+                   int64_t t, const c10::optional<at::Tensor>& bc) {
   TORCH_CHECK(p.is_cuda() && p.scalar_type() == at::kFloat);
   TORCH_CHECK(g.numel() == p.numel() && m.numel() == p.numel() &&
               v.numel() == p.numel());
   long long n = p.numel();
   float inv_bc1 = 1.0f / (1.0f - powf((float)beta1, (float)t));
   float inv_bc2 = 1.0f / (1.0f - powf((float)beta2, (float)t));
+  const float* bcp = nullptr;
+  if (bc.has_value()) {
+    TORCH_CHECK(bc->numel() == 2 && bc->scalar_type() == at::kFloat &&
+                bc->is_cuda());
+    bcp = bc->data_ptr<float>();
+  }
   auto stream = at::cuda::getCurrentCUDAStream();
   int grid = (int)std::min<long long>((n + 1023) / 1024, 2048);
   if (p_bf16.has_value()) {
@@ -55,13 +103,13 @@ void adam_step_hip(at::Tensor p, const at::Tensor& g, at::Tensor m,
                        m.data_ptr<float>(), v.data_ptr<float>(),
                        (bf16_t*)p_bf16->data_ptr(), n, (float)lr,
                        (float)beta1, (float)beta2, (float)eps, inv_bc1,
-                       inv_bc2);
+                       inv_bc2, bcp);
   } else {
     hipLaunchKernelGGL((adam_step_kernel<false>), dim3(grid), dim3(256), 0,
                        stream, p.data_ptr<float>(), g.data_ptr<float>(),
                        m.data_ptr<float>(), v.data_ptr<float>(), nullptr, n,
                        (float)lr, (float)beta1, (float)beta2, (float)eps,
-                       inv_bc1, inv_bc2);
+                       inv_bc1, inv_bc2, bcp);
   }
 }
 

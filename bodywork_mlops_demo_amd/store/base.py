@@ -38,6 +38,17 @@ class ArtefactStore(ABC):
     def delete(self, key: str) -> None:  # optional
         raise NotImplementedError
 
+    from contextlib import contextmanager as _cm
+
+    @_cm
+    def put_stream(self, key: str):
+        """Streaming put: yields a writable binary file object.  Default
+        buffers in memory; LocalStore overrides with a direct
+        tmpfile+rename (no intermediate copy of large artefacts)."""
+        bio = io.BytesIO()
+        yield bio
+        self.put_bytes(key, bio.getvalue())
+
     # -- contract-level helpers --------------------------------------------
     def latest(self, prefix: str) -> tuple[str, date]:
         """Key + date of the newest object under a prefix (stage_2:57-63)."""
@@ -73,14 +84,14 @@ class ArtefactStore(ABC):
             buf.write("\n")
             self.put_bytes(key, buf.getvalue().encode())
         elif fmt in ("npy", "npz", "bin"):
-            # raw .npy of [2, n] float32 (row 0 = y, row 1 = X): no zip
-            # container / CRC pass, so a 10M-row day persists at disk
-            # bandwidth (the .npz CRC alone cost ~as much as the write)
+            # two concatenated raw .npy records (y then X) streamed
+            # straight to the destination: no zip/CRC pass and no
+            # stack/BytesIO copies, so a 10M-row day persists at disk
+            # bandwidth
             key = contract.dataset_key(d, "npy")
-            bio = io.BytesIO()
-            np.save(bio, np.stack([y.astype(np.float32),
-                                   X.astype(np.float32)]))
-            self.put_bytes(key, bio.getvalue())
+            with self.put_stream(key) as f:
+                np.save(f, np.ascontiguousarray(y, dtype=np.float32))
+                np.save(f, np.ascontiguousarray(X, dtype=np.float32))
         else:
             raise ValueError(f"unknown dataset format {fmt!r}")
         return key
@@ -89,8 +100,11 @@ class ArtefactStore(ABC):
         """Load (y, X) float32 arrays from a dataset artefact."""
         raw = self.get_bytes(key)
         if key.endswith(".npy"):
-            arr = np.load(io.BytesIO(raw))
-            return arr[0], arr[1]
+            bio = io.BytesIO(raw)
+            first = np.load(bio)
+            if first.ndim == 2:  # legacy single stacked [2, n] record
+                return first[0], first[1]
+            return first, np.load(bio)
         if key.endswith(".npz"):  # legacy binary artefacts
             z = np.load(io.BytesIO(raw))
             return z["y"], z["X"]

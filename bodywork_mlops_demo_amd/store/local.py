@@ -55,6 +55,23 @@ class LocalStore(ArtefactStore):
                 os.unlink(tmp)
             raise
 
+    from contextlib import contextmanager as _cm
+
+    @_cm
+    def put_stream(self, key: str):
+        """Direct streaming write with the same atomicity (tmp+rename)."""
+        path = self._path(key)
+        os.makedirs(os.path.dirname(path), exist_ok=True)
+        fd, tmp = tempfile.mkstemp(dir=os.path.dirname(path), prefix=".tmp-")
+        try:
+            with os.fdopen(fd, "wb", buffering=1 << 20) as f:
+                yield f
+            os.replace(tmp, path)
+        except BaseException:
+            if os.path.exists(tmp):
+                os.unlink(tmp)
+            raise
+
     def exists(self, key: str) -> bool:
         return os.path.isfile(self._path(key))
 
