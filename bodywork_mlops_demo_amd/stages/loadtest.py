@@ -59,22 +59,37 @@ def _score_serial(url: str, X: np.ndarray) -> tuple[np.ndarray, np.ndarray]:
     return scores, times
 
 
+def _urls(url: str | list[str]) -> list[str]:
+    """Resolve the replica URL set: an explicit list, a comma-separated
+    env value (the runner's service-discovery channel, standing in for
+    the k8s ClusterIP load balancer), or the single URL given."""
+    if isinstance(url, list):
+        return url
+    if "," in url:
+        return [u.strip() for u in url.split(",") if u.strip()]
+    return [url]
+
+
 def _score_batch(
-    url: str, X: np.ndarray, chunk: int = 65536
+    url: str | list[str], X: np.ndarray, chunk: int = 65536
 ) -> tuple[np.ndarray, np.ndarray]:
-    """Chunked batch client against /score/v1/batch."""
+    """Chunked batch client against /score/v1/batch, chunks round-robined
+    across serving replicas."""
     import requests
 
-    batch_url = url.rstrip("/") + "/batch" if not url.endswith("/batch") else url
+    urls = [u.rstrip("/") + ("" if u.endswith("/batch") else "/batch")
+            for u in _urls(url)]
     session = requests.Session()
-    session.mount(batch_url, requests.adapters.HTTPAdapter(max_retries=3))
+    for u in urls:
+        session.mount(u, requests.adapters.HTTPAdapter(max_retries=3))
     scores = np.empty(X.shape[0])
     times = np.empty(X.shape[0])
-    for lo in range(0, X.shape[0], chunk):
+    for ci, lo in enumerate(range(0, X.shape[0], chunk)):
         hi = min(lo + chunk, X.shape[0])
         t0 = perf_counter()
         resp = session.post(
-            batch_url, json={"X": [float(v) for v in X[lo:hi]]}, timeout=300
+            urls[ci % len(urls)],
+            json={"X": [float(v) for v in X[lo:hi]]}, timeout=300,
         )
         dt = perf_counter() - t0
         if resp.ok:
@@ -86,21 +101,24 @@ def _score_batch(
 
 
 def _score_binary(
-    url: str, X: np.ndarray, chunk: int = 1 << 20
+    url: str | list[str], X: np.ndarray, chunk: int = 1 << 20
 ) -> tuple[np.ndarray, np.ndarray]:
-    """Binary batch client: raw float32 in/out against /score/v1/binary."""
+    """Binary batch client (raw float32 in/out), round-robined across
+    replicas; with >1 replica the chunks go out concurrently (one worker
+    per replica — the shared-nothing GPU fan-out)."""
     import requests
 
-    bin_url = url.rstrip("/") + "/binary"
+    urls = [u.rstrip("/") + "/binary" for u in _urls(url)]
     session = requests.Session()
-    session.mount(bin_url, requests.adapters.HTTPAdapter(max_retries=3))
+    for u in urls:
+        session.mount(u, requests.adapters.HTTPAdapter(max_retries=3))
     scores = np.empty(X.shape[0], dtype=np.float32)
     times = np.empty(X.shape[0])
-    for lo in range(0, X.shape[0], chunk):
-        hi = min(lo + chunk, X.shape[0])
+
+    def _one(ci: int, lo: int, hi: int):
         t0 = perf_counter()
         resp = session.post(
-            bin_url, data=X[lo:hi].astype(np.float32).tobytes(),
+            urls[ci % len(urls)], data=X[lo:hi].astype(np.float32).tobytes(),
             headers={"Content-Type": "application/octet-stream"}, timeout=300,
         )
         dt = perf_counter() - t0
@@ -109,6 +127,17 @@ def _score_binary(
         else:
             scores[lo:hi] = -1
         times[lo:hi] = dt / (hi - lo)
+
+    chunks = [(ci, lo, min(lo + chunk, X.shape[0]))
+              for ci, lo in enumerate(range(0, X.shape[0], chunk))]
+    if len(urls) > 1:
+        from concurrent.futures import ThreadPoolExecutor
+
+        with ThreadPoolExecutor(len(urls)) as pool:
+            list(pool.map(lambda c: _one(*c), chunks))
+    else:
+        for c in chunks:
+            _one(*c)
     return scores, times
 
 
@@ -183,7 +212,10 @@ def main(argv=None) -> None:
     p.add_argument("--store", default=None)
     p.add_argument(
         "--url",
-        default=os.environ.get("BODYWORK_AMD_SERVICE_URL", DEFAULT_URL),
+        default=os.environ.get(
+            "BODYWORK_AMD_SERVICE_URLS",  # all replicas (comma-separated)
+            os.environ.get("BODYWORK_AMD_SERVICE_URL", DEFAULT_URL),
+        ),
     )
     p.add_argument("--mode", default="batch",
                    choices=["serial", "batch", "binary"])
