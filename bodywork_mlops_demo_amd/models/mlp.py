@@ -158,39 +158,94 @@ class GPUMLPRegressor:
             self._adam_update(grads, lr, t)
         return self
 
+    def reinit_(self, seed: int = 7) -> "GPUMLPRegressor":
+        """Re-randomise weights and reset optimiser state IN PLACE, so a
+        cached training graph (whose nodes hold these tensor pointers)
+        stays valid for the next day's fresh fit."""
+        import math as _math
+
+        g = torch.Generator(device=self.device) if self.device.type == "cuda" \
+            else torch.Generator()
+        g.manual_seed(seed)
+        h = self.hidden
+        self.w1.normal_(0.0, _math.sqrt(2.0), generator=g)
+        self.b1.zero_()
+        self.W2.normal_(0.0, _math.sqrt(2.0 / h), generator=g)
+        self.b2.zero_()
+        self.w3.normal_(0.0, _math.sqrt(2.0 / h), generator=g)
+        self.b3.zero_()
+        if self._opt_state is not None:
+            for m, v in self._opt_state:
+                m.zero_()
+                v.zero_()
+        # refresh shadows in place (pointer-stable)
+        self.w1_bf.copy_(self.w1.bfloat16())
+        self.b1_bf.copy_(self.b1.bfloat16())
+        self.W2wt_bf.copy_(self.W2.bfloat16())
+        self.W2w_bf.copy_(ops.transpose_bf16(self.W2wt_bf)
+                          if self.device.type == "cuda"
+                          else self.W2.t().contiguous().bfloat16())
+        self.b2_bf.copy_(self.b2.bfloat16())
+        self.w3_bf.copy_(self.w3.bfloat16())
+        return self
+
     def _fit_captured(self, X, y, steps: int, bs: int, lr: float, seed: int):
         """hipGraph-captured Adam steps (see fit()).  Bias correction is a
-        device [2]-tensor the graph reads; the philox batch counter
-        advances on-device, so each replay trains on a fresh minibatch."""
+        device [2]-tensor the graph reads; the philox batch counter and
+        data size live on-device; the day's data is copied into persistent
+        capacity buffers — so ONE captured graph serves every retrain of
+        the drift loop (capture cost paid once, replay ~zero host cost)."""
         n = X.shape[0]
-        ctr = torch.zeros(1, dtype=torch.int64, device=self.device)
-        bc = torch.ones(2, dtype=torch.float32, device=self.device)
         beta1, beta2 = 0.9, 0.999
+        st = getattr(self, "_train_static", None)
+        if st is None or st["cap"] < n or st["bs"] != bs or st["lr"] != lr:
+            cap = max(n, int(st["cap"]) if st else 0)
+            st = {
+                "cap": cap, "bs": bs, "lr": lr,
+                "X": torch.empty(cap, device=self.device),
+                "y": torch.empty(cap, device=self.device),
+                "n_dev": torch.zeros(1, dtype=torch.int64, device=self.device),
+                "ctr": torch.zeros(1, dtype=torch.int64, device=self.device),
+                "bc": torch.ones(2, device=self.device),
+                "graph": None,
+            }
+            self._train_static = st
+        st["X"][:n].copy_(X)
+        st["y"][:n].copy_(y)
+        st["n_dev"].fill_(n)
+        st["ctr"].zero_()
+        bc = st["bc"]
 
         def one_step():
-            idx = ops.batch_indices(ctr, n, bs, seed)
-            xb = X.index_select(0, idx)
-            yb = y.index_select(0, idx)
+            idx = ops.batch_indices(st["ctr"], st["n_dev"], bs, seed)
+            xb = st["X"].index_select(0, idx)
+            yb = st["y"].index_select(0, idx)
             grads = self._step_grads(xb, yb)
             shadows = [self.w1_bf, self.b1_bf, self.W2wt_bf, self.b2_bf,
                        self.w3_bf, None]
-            for p, gr, (m, v), s in zip(self.parameters(), grads,
-                                        self._opt_state, shadows):
-                ops.adam_step(p, gr, m, v, s, lr, t=1, bc=bc)
+            for p, gr, (m, v), sh in zip(self.parameters(), grads,
+                                         self._opt_state, shadows):
+                ops.adam_step(p, gr, m, v, sh, lr, t=1, bc=bc)
             # refresh the transposed forward copy in place (graph-safe)
             self.W2w_bf.copy_(ops.transpose_bf16(self.W2wt_bf))
 
-        # warmup on a side stream, then capture
-        s = torch.cuda.Stream()
-        s.wait_stream(torch.cuda.current_stream())
-        with torch.cuda.stream(s):
-            bc.copy_(torch.tensor([1 / (1 - beta1), 1 / (1 - beta2)]))
-            one_step()
-        torch.cuda.current_stream().wait_stream(s)
-        graph = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(graph):
-            one_step()
-        for t in range(2, steps + 1):
+        t0 = 1
+        if st["graph"] is None:
+            # warmup on a side stream (this executes one real step), then
+            # capture (recording only — not an executed step)
+            stream = torch.cuda.Stream()
+            stream.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(stream):
+                bc.copy_(torch.tensor([1 / (1 - beta1), 1 / (1 - beta2)]))
+                one_step()
+            torch.cuda.current_stream().wait_stream(stream)
+            graph = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(graph):
+                one_step()
+            st["graph"] = graph
+            t0 = 2  # the warmup was step 1
+        graph = st["graph"]
+        for t in range(t0, steps + 1):
             bc.copy_(torch.tensor([1 / (1 - beta1**t), 1 / (1 - beta2**t)],
                                   dtype=torch.float32), non_blocking=True)
             graph.replay()

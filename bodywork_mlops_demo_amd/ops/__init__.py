@@ -412,12 +412,14 @@ def adam_step(
 
 
 def batch_indices(
-    ctr: torch.Tensor, n_data: int, bs: int, seed: int
+    ctr: torch.Tensor, n_dev: torch.Tensor, bs: int, seed: int
 ) -> torch.Tensor:
     """Device-side philox minibatch sampling; ``ctr`` (int64 [1], device)
-    advances on-device so a captured step resamples on every replay."""
+    advances on-device and the data size ``n_dev`` (int64 [1], device) is
+    read in-kernel, so a captured step stays valid across days whose row
+    counts differ."""
     core = _core(ctr.device)
-    return core.batch_indices(ctr, n_data, bs, seed)
+    return core.batch_indices(ctr, n_dev, bs, seed)
 
 
 def transpose_to_bf16(src: torch.Tensor) -> torch.Tensor:

@@ -37,8 +37,8 @@ void adam_step_hip(at::Tensor p, const at::Tensor& g, at::Tensor m,
                    at::Tensor v, const c10::optional<at::Tensor>& p_bf16,
                    double lr, double beta1, double beta2, double eps,
                    int64_t t, const c10::optional<at::Tensor>& bc);
-at::Tensor batch_indices_hip(at::Tensor ctr, int64_t n_data, int64_t bs,
-                             int64_t seed);
+at::Tensor batch_indices_hip(at::Tensor ctr, const at::Tensor& n_dev,
+                             int64_t bs, int64_t seed);
 at::Tensor transpose_to_bf16_hip(const at::Tensor& src);
 at::Tensor transpose_bf16_hip(const at::Tensor& src);
 
@@ -62,7 +62,7 @@ TORCH_LIBRARY(bodywork_hip, m) {
   m.def("adam_step(Tensor(a!) p, Tensor g, Tensor(b!) m, Tensor(c!) v, "
         "Tensor(d!)? p_bf16, float lr, float beta1, float beta2, float eps, "
         "int t, Tensor? bc) -> ()");
-  m.def("batch_indices(Tensor(a!) ctr, int n_data, int bs, int seed) "
+  m.def("batch_indices(Tensor(a!) ctr, Tensor n_dev, int bs, int seed) "
         "-> Tensor");
   m.def("transpose_to_bf16(Tensor src) -> Tensor");
   m.def("transpose_bf16(Tensor src) -> Tensor");

@@ -46,32 +46,35 @@ __global__ void adam_step_kernel(float* __restrict__ p, const float* __restrict_
 // graph replay (no host RNG, no H2D index copy)
 __global__ void batch_indices_kernel(int64_t* __restrict__ out,
                                      unsigned long long* __restrict__ ctr,
-                                     long long n_data, long long bs,
-                                     unsigned int key0, unsigned int key1) {
+                                     const long long* __restrict__ n_ptr,
+                                     long long bs, unsigned int key0,
+                                     unsigned int key1) {
   const unsigned long long step = *ctr;
+  const unsigned long long n_data = (unsigned long long)*n_ptr;
   const long long stride = (long long)gridDim.x * blockDim.x;
   for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
        i < bs; i += stride) {
     Philox4 r = philox4x32(step * (unsigned long long)bs + i, key0, key1);
-    out[i] = (int64_t)(((unsigned long long)r.x << 32 | r.y) %
-                       (unsigned long long)n_data);
+    out[i] = (int64_t)(((unsigned long long)r.x << 32 | r.y) % n_data);
   }
 }
 
 __global__ void bump_counter_kernel(unsigned long long* ctr) { ++*ctr; }
 
-at::Tensor batch_indices_hip(at::Tensor ctr, int64_t n_data, int64_t bs,
-                             int64_t seed) {
+at::Tensor batch_indices_hip(at::Tensor ctr, const at::Tensor& n_dev,
+                             int64_t bs, int64_t seed) {
   TORCH_CHECK(ctr.is_cuda() && ctr.numel() == 1 &&
               ctr.scalar_type() == at::kLong);
+  TORCH_CHECK(n_dev.is_cuda() && n_dev.numel() == 1 &&
+              n_dev.scalar_type() == at::kLong);
   auto out = at::empty({bs}, ctr.options());
   auto stream = at::cuda::getCurrentCUDAStream();
   unsigned int key0 = (unsigned int)(seed & 0xFFFFFFFFll);
   int grid = (int)std::min<int64_t>((bs + 255) / 256, 1024);
   hipLaunchKernelGGL(batch_indices_kernel, dim3(grid), dim3(256), 0, stream,
                      out.data_ptr<int64_t>(),
-                     (unsigned long long*)ctr.data_ptr(), n_data, bs, key0,
-                     0xB5297A4Du);
+                     (unsigned long long*)ctr.data_ptr(),
+                     n_dev.data_ptr<int64_t>(), bs, key0, 0xB5297A4Du);
   hipLaunchKernelGGL(bump_counter_kernel, dim3(1), dim3(1), 0, stream,
                      (unsigned long long*)ctr.data_ptr());
   return out;

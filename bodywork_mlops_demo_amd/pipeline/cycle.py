@@ -149,6 +149,7 @@ def run_cycle(
         mlp_batch_size=mlp_batch_size,
         data=(state.y, state.X, state.date),
         return_model=True,
+        model_cache=scorer_cache,
     )
     sync()
     timings["train_s"] = perf_counter() - t0

@@ -43,6 +43,7 @@ def run(
     mlp_batch_size: int = 65536,
     data: tuple[torch.Tensor, torch.Tensor, date_t] | None = None,
     return_model: bool = False,
+    model_cache: dict | None = None,
 ):
     """Train and persist; returns the offline metrics record
     (or ``(metrics, model)`` when ``return_model``).
@@ -74,7 +75,19 @@ def run(
             X_train, y_train, process_group=process_group
         )
     elif model_type == "mlp":
-        model = GPUMLPRegressor(device=device).fit(
+        # reuse the cached model object so its captured training graph
+        # survives across daily retrains (weights re-randomised in place)
+        model = None
+        if model_cache is not None:
+            cached = model_cache.get("mlp")
+            if (isinstance(cached, GPUMLPRegressor)
+                    and str(cached.device) == str(torch.device(device))):
+                model = cached.reinit_(seed=7 + data_date.toordinal())
+        if model is None:
+            model = GPUMLPRegressor(device=device)
+            if model_cache is not None:
+                model_cache["mlp"] = model
+        model.fit(
             X_train, y_train, steps=mlp_steps, batch_size=mlp_batch_size,
             process_group=process_group,
         )
