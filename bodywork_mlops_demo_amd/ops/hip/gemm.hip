@@ -170,8 +170,22 @@ __global__ void gemm_nt_glds_kernel(
     const float* __restrict__ bias, const bf16_t* __restrict__ mask,
     void* __restrict__ C, long long M, long long N, long long K) {
   __shared__ short lds_all[2 * 2 * 128 * 64];  // [buf][A/B][128][64]
-  const long long m0 = (long long)blockIdx.y * BM;
-  const long long n0 = (long long)blockIdx.x * BN;
+  // XCD-aware block remap (T1, bijective form): the dispatcher places
+  // block b on XCD b%8; remapping gives each XCD a contiguous run of
+  // tiles so neighbouring tiles' operand panels hit the same L2.
+  long long bx, by;
+  {
+    const long long nwg = (long long)gridDim.x * gridDim.y;
+    const long long orig = (long long)blockIdx.y * gridDim.x + blockIdx.x;
+    const long long q = nwg >> 3, r = nwg & 7;
+    const long long xcd = orig & 7;
+    const long long wgid =
+        (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + (orig >> 3);
+    bx = wgid % gridDim.x;
+    by = wgid / gridDim.x;
+  }
+  const long long m0 = by * BM;
+  const long long n0 = bx * BN;
   const int wave = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
   const int wm = wave >> 1, wn = wave & 1;

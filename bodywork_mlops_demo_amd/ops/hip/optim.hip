@@ -46,7 +46,7 @@ __global__ void adam_step_kernel(float* __restrict__ p, const float* __restrict_
 // graph replay (no host RNG, no H2D index copy)
 __global__ void batch_indices_kernel(int64_t* __restrict__ out,
                                      unsigned long long* __restrict__ ctr,
-                                     const long long* __restrict__ n_ptr,
+                                     const int64_t* __restrict__ n_ptr,
                                      long long bs, unsigned int key0,
                                      unsigned int key1) {
   const unsigned long long step = *ctr;
