@@ -203,17 +203,27 @@ __global__ void gemm_nt_glds_kernel(
 
   stage_tile_glds(lds_all, A, K, m0, M, 0);
   stage_tile_glds(lds_all + 2 * TILE, B, K, n0, N, 0);
-  __syncthreads();  // carries the vmcnt(0) for the in-flight glds
 
+  // Counted-vmcnt pipeline (guide §5 "Pipelining across barriers"): each
+  // iteration issues tile t+1's 8 wave-glds, then waits vmcnt(8) — tile
+  // t's staging has landed, t+1's stays IN FLIGHT across the compute —
+  // with raw s_barriers (a __syncthreads here would emit vmcnt(0) and
+  // drain the pipeline).  Every wave passes its own counted wait before
+  // the barrier, so all of tile t's LDS writes are visible to all waves.
   int cur = 0;
   for (long long k0 = 0; k0 < K; k0 += BK) {
     short* as_cur = lds_all + cur * TILE;
     short* bs_cur = lds_all + 2 * TILE + cur * TILE;
-    if (k0 + BK < K) {
+    const bool prefetching = k0 + BK < K;
+    if (prefetching) {
       stage_tile_glds(lds_all + (cur ^ 1) * TILE, A, K, m0, M, k0 + BK);
       stage_tile_glds(lds_all + 2 * TILE + (cur ^ 1) * TILE, B, K, n0, N,
                       k0 + BK);
+      asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+    } else {
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     }
+    __builtin_amdgcn_s_barrier();
 #pragma unroll
     for (int ks = 0; ks < 2; ++ks) {
       bf16x8_v a_frag[4], b_frag[4];
@@ -234,7 +244,9 @@ __global__ void gemm_nt_glds_kernel(
           acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               a_frag[i], b_frag[j], acc[i][j], 0, 0, 0);
     }
-    __syncthreads();  // drains next tile's glds; makes buffers reusable
+    // reads of buf[cur] complete before the next iteration's glds may
+    // overwrite it (writes are issued only after this barrier)
+    __builtin_amdgcn_s_barrier();
     cur ^= 1;
   }
   write_epilogue<EPI, HAS_BIAS, OUT_FP32>(acc, bias, mask, C, M, N, m0, n0,
