@@ -1,0 +1,94 @@
+"""Error-monitor + logging + CLI surface tests."""
+import json
+import logging
+import subprocess
+import sys
+
+import pytest
+
+from bodywork_mlops_demo_amd.monitoring.errors import ErrorMonitor, stage_guard
+from bodywork_mlops_demo_amd.utils.clock import VirtualClock
+from bodywork_mlops_demo_amd.utils.logging import LOG_FORMAT, configure_logger
+
+
+def test_error_monitor_event_log(tmp_path, monkeypatch):
+    monkeypatch.delenv("SENTRY_DSN", raising=False)
+    log_path = str(tmp_path / "events.jsonl")
+    mon = ErrorMonitor(dsn=None, event_log_path=log_path)
+    mon.set_tag("stage", "stage-1-train-model")
+    try:
+        raise ValueError("boom")
+    except ValueError as e:
+        mon.capture_exception(e)
+    mon.capture_message("drift detected", level="warning")
+
+    events = [json.loads(line) for line in open(log_path)]
+    assert len(events) == 2
+    assert events[0]["kind"] == "exception"
+    assert events[0]["type"] == "ValueError"
+    assert events[0]["tags"]["stage"] == "stage-1-train-model"
+    assert "boom" in events[0]["traceback"]
+    assert events[1]["kind"] == "message"
+    assert events[1]["level"] == "warning"
+
+
+def test_stage_guard_reraises_and_captures(tmp_path, monkeypatch):
+    monkeypatch.setenv("BODYWORK_AMD_EVENT_LOG",
+                       str(tmp_path / "ev.jsonl"))
+    import bodywork_mlops_demo_amd.monitoring.errors as errors
+
+    monkeypatch.setattr(errors, "_MONITOR", None)  # fresh monitor
+    with pytest.raises(RuntimeError):
+        with stage_guard("stage-x"):
+            raise RuntimeError("stage failed")
+    events = [json.loads(line) for line in open(tmp_path / "ev.jsonl")]
+    assert events[0]["tags"]["stage"] == "stage-x"
+
+
+def test_log_format_matches_reference():
+    # byte-identical record format (stage_1_train_model.py:148-153)
+    assert LOG_FORMAT == ("%(asctime)s - %(levelname)s - "
+                         "%(module)s.%(funcName)s - %(message)s")
+    log = configure_logger("parity-test", level="DEBUG")
+    assert log.level == logging.DEBUG
+
+
+def test_virtual_clock():
+    c = VirtualClock("2026-03-01")
+    assert str(c.today()) == "2026-03-01"
+    c.advance(2)
+    assert str(c.today()) == "2026-03-03"
+    assert c.day_of_year() == 62
+    c.set("2026-12-31")
+    assert c.day_of_year() == 365
+
+
+@pytest.mark.timeout(120)
+def test_package_cli_help_and_unknown():
+    out = subprocess.run([sys.executable, "-m", "bodywork_mlops_demo_amd"],
+                         capture_output=True, text=True)
+    assert out.returncode == 0
+    assert "run" in out.stdout and "analytics" in out.stdout
+    bad = subprocess.run(
+        [sys.executable, "-m", "bodywork_mlops_demo_amd", "nope"],
+        capture_output=True, text=True)
+    assert bad.returncode == 2
+
+
+@pytest.mark.timeout(180)
+def test_analytics_cli(tmp_path):
+    from datetime import date
+
+    from bodywork_mlops_demo_amd.pipeline.loop import run_loop
+    from bodywork_mlops_demo_amd.store import LocalStore
+
+    store_dir = str(tmp_path / "store")
+    run_loop(LocalStore(store_dir), days=2, n_rows=400, device="cpu",
+             start_date="2026-04-01")
+    out = subprocess.run(
+        [sys.executable, "-m", "bodywork_mlops_demo_amd", "analytics",
+         "--store", store_dir, "--csv-out", str(tmp_path / "joined.csv")],
+        capture_output=True, text=True)
+    assert out.returncode == 0, out.stderr[-1500:]
+    assert "summary" in out.stdout
+    assert (tmp_path / "joined.csv").exists()
