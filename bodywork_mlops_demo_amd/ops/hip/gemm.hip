@@ -351,9 +351,13 @@ static void launch_gemm(bool tn, int epi, bool has_bias, bool out_fp32,
 
   // hot path: NT with K a tile multiple -> glds-staged kernel
   if (!tn && K % BK == 0 && K > 0) {
+    // A/B-measured on MI355X: the T1 remap LOSES 4-14% on these NT
+    // shapes (887 vs 1029 TF @8192^3 — the bx-major walk trades away the
+    // default round-robin's B-panel spread), so default OFF; env
+    // BODYWORK_GEMM_XCD=1 re-enables for experiments.
     static int xcd_swz = [] {
       const char* e = getenv("BODYWORK_GEMM_XCD");
-      return (e && e[0] == '0') ? 0 : 1;
+      return (e && e[0] == '1') ? 1 : 0;
     }();
 #define G_GLDS(EPI_, HB_, OF_)                                              \
   hipLaunchKernelGGL((gemm_nt_glds_kernel<EPI_, HB_, OF_>), grid,           \
