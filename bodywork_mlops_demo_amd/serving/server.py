@@ -25,6 +25,7 @@ from fastapi import FastAPI, Request
 from fastapi.responses import JSONResponse, Response
 
 from bodywork_mlops_demo_amd.models import regressor_from_artifact
+from bodywork_mlops_demo_amd.monitoring.tracing import RequestTracer, timed
 from bodywork_mlops_demo_amd.serving.scorer import BatchedScorer
 from bodywork_mlops_demo_amd.store import ArtefactStore
 from bodywork_mlops_demo_amd.utils.logging import configure_logger
@@ -35,6 +36,7 @@ log = configure_logger(__name__)
 def create_app(store: ArtefactStore, device: str = "cpu",
                use_graphs: bool = True) -> FastAPI:
     state: dict = {}
+    tracer = RequestTracer()
 
     @asynccontextmanager
     async def lifespan(app: FastAPI):
@@ -54,7 +56,8 @@ def create_app(store: ArtefactStore, device: str = "cpu",
         payload = await request.json()
         features = payload["X"]
         scalar = np.isscalar(features)
-        preds = state["scorer"].score(features)
+        with timed(tracer, rows=1 if scalar else len(features)):
+            preds = state["scorer"].score(features)
         prediction = float(preds[0]) if scalar else [float(p) for p in preds]
         return JSONResponse(
             {"prediction": prediction, "model_info": state["model_info"]}
@@ -63,7 +66,8 @@ def create_app(store: ArtefactStore, device: str = "cpu",
     @app.post("/score/v1/batch")
     async def score_batch(request: Request) -> Response:
         payload = await request.json()
-        preds = state["scorer"].score(payload["X"])
+        with timed(tracer, rows=len(payload["X"])):
+            preds = state["scorer"].score(payload["X"])
         return JSONResponse(
             {
                 "predictions": [float(p) for p in preds],
@@ -80,11 +84,24 @@ def create_app(store: ArtefactStore, device: str = "cpu",
         parsing than the GPU does scoring)."""
         body = await request.body()
         X = np.frombuffer(body, dtype=np.float32)
-        preds = state["scorer"].score(X)
+        with timed(tracer, rows=X.shape[0]):
+            preds = state["scorer"].score(X)
         return Response(content=preds.astype(np.float32).tobytes(),
                         media_type="application/octet-stream",
                         headers={"X-Model-Info": state["model_info"],
                                  "X-N": str(preds.shape[0])})
+
+    @app.get("/stats")
+    async def stats() -> Response:
+        return JSONResponse(tracer.snapshot())
+
+    @app.get("/metrics")
+    async def metrics() -> Response:
+        prom = RequestTracer.prometheus()
+        if prom is None:
+            return JSONResponse(tracer.snapshot())
+        data, content_type = prom
+        return Response(content=data, media_type=content_type)
 
     @app.get("/healthz")
     async def healthz() -> Response:

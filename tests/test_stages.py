@@ -92,6 +92,16 @@ def test_serving_http_wire_format(seeded_store):
         r = client.get("/healthz")
         assert r.json()["status"] == "ok"
 
+        # tracing: latency histogram + prometheus exposition
+        r = client.get("/stats")
+        snap = r.json()
+        assert snap["requests"] >= 3
+        assert snap["rows_scored"] >= 8
+        assert snap["mean_latency_s"] > 0
+        r = client.get("/metrics")
+        assert r.status_code == 200
+        assert b"scoring_requests_total" in r.content or "requests" in r.text
+
 
 def test_mlp_stage1(seeded_store):
     metrics = train.run(
