@@ -10,7 +10,7 @@ every scoring request is timed in-process and exported:
 """
 from __future__ import annotations
 
-import math
+
 import threading
 import time
 

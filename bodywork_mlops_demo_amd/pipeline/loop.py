@@ -13,7 +13,7 @@ from __future__ import annotations
 
 import argparse
 import json
-from datetime import date as date_t, timedelta
+from datetime import date as date_t
 
 import torch
 
