@@ -91,16 +91,24 @@ class GPUMLPRegressor:
         return [self.w1, self.b1, self.W2, self.b2, self.w3, self.b3]
 
     # -- forward -----------------------------------------------------------
-    def _forward(self, x: torch.Tensor):
+    def _forward(self, x: torch.Tensor, want_masks: bool = False):
+        """Forward pass; with ``want_masks`` the relu layers also emit
+        their 1-bit activation masks (consumed by the masked backward —
+        16x less mask traffic than re-reading the activations)."""
         xn = (x.float() - self.X_MU) / self.X_SIGMA
-        h1 = ops.expand1d_bf16(xn, self.w1_bf, self.b1_bf, relu=True)
-        h2 = ops.linear_bf16(h1, self.W2w_bf, bias=self.b2_bf, relu=True)
+        if want_masks:
+            h1, m1 = ops.expand1d_bf16(xn, self.w1_bf, self.b1_bf, relu=True,
+                                       emit_mask=True)
+            h2, m2 = ops.linear_relu_mask_bf16(h1, self.W2w_bf, self.b2_bf)
+        else:
+            h1 = ops.expand1d_bf16(xn, self.w1_bf, self.b1_bf, relu=True)
+            h2 = ops.linear_bf16(h1, self.W2w_bf, bias=self.b2_bf, relu=True)
+            m1 = m2 = None
         yhat = ops.rowdot_bf16(h2, self.w3_bf, self.b3)
-        return yhat, h1, h2, xn
+        return yhat, h1, h2, xn, m1, m2
 
     def predict(self, X: torch.Tensor) -> torch.Tensor:
-        yhat, _, _, _ = self._forward(X.to(self.device))
-        return yhat
+        return self._forward(X.to(self.device))[0]
 
     # -- training ----------------------------------------------------------
     def fit(
@@ -252,17 +260,17 @@ class GPUMLPRegressor:
 
     def _step_grads(self, xb: torch.Tensor, yb: torch.Tensor) -> list[torch.Tensor]:
         nb = xb.shape[0]
-        yhat, h1, h2, xn = self._forward(xb)
+        yhat, h1, h2, xn, m1, m2 = self._forward(xb, want_masks=True)
         dy = (2.0 / nb) * (yhat - yb.float())                   # (n,)
         # layer 3: yhat = h2 @ w3 + b3
         dw3 = ops.coldot_bf16(h2, dy)                           # (H,)
         db3 = dy.sum().reshape(1)
-        # dh2 = outer(dy, w3) * relu'(h2): fused expand with mask
-        dz2 = ops.expand1d_bf16(dy, self.w3_bf, None, relu=False, mask=h2)
+        # dh2 = outer(dy, w3) * relu'(h2): fused expand with 1-bit mask
+        dz2 = ops.expand1d_bf16(dy, self.w3_bf, None, relu=False, mask=m2)
         # layer 2: h2 = relu(h1 @ W2 + b2), W2 [in,out]
         dW2 = ops.gemm_tn_bf16(h1, dz2, out_fp32=True)          # (in,out)
         db2 = ops.colsum_bf16(dz2)
-        dz1 = ops.linear_bf16(dz2, self.W2wt_bf, mask=h1)       # dz2 @ W2^T
+        dz1 = ops.linear_bf16(dz2, self.W2wt_bf, mask=m1)       # dz2 @ W2^T
         # layer 1: h1 = relu(xn w1 + b1)
         dw1, db1 = ops.coldot_bf16(dz1, xn, also_colsum=True)
         return [dw1, db1, dW2, db2, dw3, db3]

@@ -287,6 +287,19 @@ def linear_bf16(
     return reference.linear_bf16_cpu(x, w, bias, relu, mask, out_fp32)
 
 
+def linear_relu_mask_bf16(
+    x: torch.Tensor, w: torch.Tensor, bias: torch.Tensor | None = None
+) -> tuple[torch.Tensor, torch.Tensor]:
+    """relu(x @ w^T + bias) AND its 1-bit activation mask in one pass
+    (the MLP forward hot path; the mask feeds the masked backward GEMM)."""
+    if x.device.type == "cuda":
+        core = _core(x.device)
+        return core.linear_relu_mask_bf16(
+            x.contiguous(), w.contiguous(),
+            bias.contiguous() if bias is not None else None)
+    return reference.linear_relu_mask_cpu(x, w, bias)
+
+
 def gemm_tn_bf16(
     a: torch.Tensor, b: torch.Tensor, out_fp32: bool = False
 ) -> torch.Tensor:
@@ -323,23 +336,29 @@ def expand1d_bf16(
     b: torch.Tensor | None = None,
     relu: bool = False,
     mask: torch.Tensor | None = None,
-) -> torch.Tensor:
+    emit_mask: bool = False,
+):
     """Fused rank-1 expansion: out[i,j] = act(x[i]*w[j] + b[j]).
 
     The MLP's 1-feature input layer (and its backward dh = outer(dy, w))
     as a single bandwidth-bound kernel instead of a degenerate K=1 MFMA
-    GEMM.  ``mask`` multiplies by (mask>0) — ReLU backward.  Returns bf16
-    (n, H).
+    GEMM.  ``mask`` is a 1-BIT relu mask (uint8 [n, H/8], bit e of byte c
+    = column 8c+e active) — 16x less backward mask traffic than re-reading
+    activations; ``emit_mask`` additionally returns that bitmask for the
+    produced activations.  Returns bf16 (n, H) (or a (out, maskbits)
+    pair when ``emit_mask``).
     """
     if x.device.type == "cuda":
         core = _core(x.device)
-        return core.expand1d_bf16(
+        out, mbits = core.expand1d_bf16(
             x.contiguous().float(), w.contiguous(),
             b.contiguous() if b is not None else None,
             relu,
             mask.contiguous() if mask is not None else None,
+            emit_mask,
         )
-    return reference.expand1d_cpu(x, w, b, relu, mask)
+        return (out, mbits) if emit_mask else out
+    return reference.expand1d_cpu(x, w, b, relu, mask, emit_mask)
 
 
 def rowdot_bf16(

@@ -17,9 +17,10 @@ at::Tensor linear_score_hip(const at::Tensor& x, const at::Tensor& ab);
 at::Tensor regression_metrics_hip(const at::Tensor& y, const at::Tensor& yhat);
 at::Tensor score_label_metrics_hip(const at::Tensor& s, const at::Tensor& l);
 // mlp_small.hip
-at::Tensor expand1d_bf16_hip(const at::Tensor& x, const at::Tensor& w,
-                             const c10::optional<at::Tensor>& b, bool relu,
-                             const c10::optional<at::Tensor>& mask);
+std::tuple<at::Tensor, at::Tensor> expand1d_bf16_hip(
+    const at::Tensor& x, const at::Tensor& w,
+    const c10::optional<at::Tensor>& b, bool relu,
+    const c10::optional<at::Tensor>& mask, bool emit_mask);
 at::Tensor rowdot_bf16_hip(const at::Tensor& h, const at::Tensor& w,
                            const at::Tensor& bias);
 at::Tensor coldot_bf16_hip(const at::Tensor& m, const at::Tensor& v,
@@ -30,6 +31,9 @@ at::Tensor linear_bf16_hip(const at::Tensor& x, const at::Tensor& w,
                            const c10::optional<at::Tensor>& bias, bool relu,
                            const c10::optional<at::Tensor>& mask,
                            bool out_fp32);
+std::tuple<at::Tensor, at::Tensor> linear_relu_mask_bf16_hip(
+    const at::Tensor& x, const at::Tensor& w,
+    const c10::optional<at::Tensor>& bias);
 at::Tensor gemm_tn_bf16_hip(const at::Tensor& a, const at::Tensor& b,
                             bool out_fp32);
 // optim.hip
@@ -52,12 +56,14 @@ TORCH_LIBRARY(bodywork_hip, m) {
   m.def("regression_metrics(Tensor y, Tensor yhat) -> Tensor");
   m.def("score_label_metrics(Tensor s, Tensor l) -> Tensor");
   m.def("expand1d_bf16(Tensor x, Tensor w, Tensor? b, bool relu, "
-        "Tensor? mask) -> Tensor");
+        "Tensor? mask, bool emit_mask) -> (Tensor, Tensor)");
   m.def("rowdot_bf16(Tensor h, Tensor w, Tensor bias) -> Tensor");
   m.def("coldot_bf16(Tensor m, Tensor v, bool also_colsum) -> Tensor");
   m.def("colsum_bf16(Tensor m) -> Tensor");
   m.def("linear_bf16(Tensor x, Tensor w, Tensor? bias, bool relu, "
         "Tensor? mask, bool out_fp32) -> Tensor");
+  m.def("linear_relu_mask_bf16(Tensor x, Tensor w, Tensor? bias) -> "
+        "(Tensor, Tensor)");
   m.def("gemm_tn_bf16(Tensor a, Tensor b, bool out_fp32) -> Tensor");
   m.def("adam_step(Tensor(a!) p, Tensor g, Tensor(b!) m, Tensor(c!) v, "
         "Tensor(d!)? p_bf16, float lr, float beta1, float beta2, float eps, "
@@ -80,6 +86,7 @@ TORCH_LIBRARY_IMPL(bodywork_hip, CUDA, m) {
   m.impl("coldot_bf16", coldot_bf16_hip);
   m.impl("colsum_bf16", colsum_bf16_hip);
   m.impl("linear_bf16", linear_bf16_hip);
+  m.impl("linear_relu_mask_bf16", linear_relu_mask_bf16_hip);
   m.impl("gemm_tn_bf16", gemm_tn_bf16_hip);
   m.impl("adam_step", adam_step_hip);
   m.impl("batch_indices", batch_indices_hip);

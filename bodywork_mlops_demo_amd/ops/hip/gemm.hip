@@ -97,35 +97,53 @@ __device__ __forceinline__ void stage_transposed(
 #define EPI_BIAS_RELU 1  // +bias then relu (bias may be null -> relu only)
 #define EPI_MASK 2       // multiply by (mask > 0)
 
-// shared epilogue: write the 4x4 fragment accumulator block
-template <int EPI, bool HAS_BIAS, bool OUT_FP32>
+// shared epilogue: write the 4x4 fragment accumulator block.
+// EPI_MASK consumes a 1-BIT ReLU mask (uint8 [M, N/8], bit = col&7);
+// EMIT_MASK writes that bitmask for relu outputs (wave-ballot: the 16
+// lanes of a row-group supply the 16 bits, lane fl==0 stores a ushort).
+template <int EPI, bool HAS_BIAS, bool OUT_FP32, bool EMIT_MASK = false>
 __device__ __forceinline__ void write_epilogue(
     f32x4 (&acc)[4][4], const float* __restrict__ bias,
-    const bf16_t* __restrict__ mask, void* __restrict__ C, long long M,
+    const unsigned char* __restrict__ mask,
+    unsigned char* __restrict__ mask_out, void* __restrict__ C, long long M,
     long long N, long long m0, long long n0, int wm, int wn, int fl, int kg) {
 #pragma unroll
   for (int i = 0; i < 4; ++i) {
 #pragma unroll
     for (int j = 0; j < 4; ++j) {
       long long col = n0 + wn * 64 + j * 16 + fl;
-      if (col >= N) continue;
-      float bval = (EPI == EPI_BIAS_RELU && HAS_BIAS) ? bias[col] : 0.0f;
+      bool col_ok = col < N;
+      float bval =
+          (EPI == EPI_BIAS_RELU && HAS_BIAS && col_ok) ? bias[col] : 0.0f;
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         long long row = m0 + wm * 64 + i * 16 + kg * 4 + r;
-        if (row >= M) continue;
+        bool row_ok = row < M;
         float v = acc[i][j][r];
         if (EPI == EPI_BIAS_RELU) {
           v += bval;
           v = fmaxf(v, 0.0f);
         } else if (EPI == EPI_MASK) {
-          float mv = bf16_to_f32(mask[row * N + col]);
-          v = mv > 0.0f ? v : 0.0f;
+          unsigned char mb = (row_ok && col_ok)
+                                 ? mask[row * (N >> 3) + (col >> 3)]
+                                 : (unsigned char)0;
+          v = (mb >> (fl & 7)) & 1 ? v : 0.0f;
         }
-        if (OUT_FP32)
-          ((float*)C)[row * N + col] = v;
-        else
-          ((bf16_t*)C)[row * N + col] = f32_to_bf16(v);
+        if (EMIT_MASK) {
+          // ballot over the wave: bits l of b = lane l's v>0; this
+          // row-group's 16 cols live at bits [kg*16, kg*16+16)
+          unsigned long long b = __ballot(v > 0.0f);
+          if (fl == 0 && row_ok && col < N) {
+            unsigned short bits = (unsigned short)((b >> (kg * 16)) & 0xFFFF);
+            *(unsigned short*)(mask_out + row * (N >> 3) + (col >> 3)) = bits;
+          }
+        }
+        if (row_ok && col_ok) {
+          if (OUT_FP32)
+            ((float*)C)[row * N + col] = v;
+          else
+            ((bf16_t*)C)[row * N + col] = f32_to_bf16(v);
+        }
       }
     }
   }
@@ -162,12 +180,13 @@ __device__ __forceinline__ void stage_tile_glds(
   }
 }
 
-template <int EPI, bool HAS_BIAS, bool OUT_FP32>
+template <int EPI, bool HAS_BIAS, bool OUT_FP32, bool EMIT_MASK = false>
 __launch_bounds__(GEMM_THREADS)
 __global__ void gemm_nt_glds_kernel(
     const bf16_t* __restrict__ A,  // [M,K]
     const bf16_t* __restrict__ B,  // [N,K]
-    const float* __restrict__ bias, const bf16_t* __restrict__ mask,
+    const float* __restrict__ bias, const unsigned char* __restrict__ mask,
+    unsigned char* __restrict__ mask_out,
     void* __restrict__ C, long long M, long long N, long long K,
     int xcd_swz) {
   __shared__ short lds_all[2 * 2 * 128 * 64];  // [buf][A/B][128][64]
@@ -248,8 +267,8 @@ __global__ void gemm_nt_glds_kernel(
     __syncthreads();  // drains next tile's glds; makes buffers reusable
     cur ^= 1;
   }
-  write_epilogue<EPI, HAS_BIAS, OUT_FP32>(acc, bias, mask, C, M, N, m0, n0,
-                                          wm, wn, fl, kg);
+  write_epilogue<EPI, HAS_BIAS, OUT_FP32, EMIT_MASK>(
+      acc, bias, mask, mask_out, C, M, N, m0, n0, wm, wn, fl, kg);
 }
 
 template <bool TN, int EPI, bool HAS_BIAS, bool OUT_FP32>
@@ -257,7 +276,7 @@ __launch_bounds__(GEMM_THREADS)
 __global__ void gemm_bf16_kernel(
     const bf16_t* __restrict__ A,  // NT: [M,K]; TN: [R=K, M]
     const bf16_t* __restrict__ B,  // NT: [N,K]; TN: [R=K, N]
-    const float* __restrict__ bias, const bf16_t* __restrict__ mask,
+    const float* __restrict__ bias, const unsigned char* __restrict__ mask,
     void* __restrict__ C, long long M, long long N, long long K) {
   __shared__ short lds_all[2 * BM * LDS_STRIDE];
   short* As = lds_all;
@@ -307,41 +326,16 @@ __global__ void gemm_bf16_kernel(
     __syncthreads();
   }
 
-  // epilogue: C[row][col], row = tile_m + wm*64 + i*16 + kg*4 + r,
-  //           col = tile_n + wn*64 + j*16 + fl
-#pragma unroll
-  for (int i = 0; i < 4; ++i) {
-#pragma unroll
-    for (int j = 0; j < 4; ++j) {
-      long long col = n0 + wn * 64 + j * 16 + fl;
-      if (col >= N) continue;
-      float bval = (EPI == EPI_BIAS_RELU && HAS_BIAS) ? bias[col] : 0.0f;
-#pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        long long row = m0 + wm * 64 + i * 16 + kg * 4 + r;
-        if (row >= M) continue;
-        float v = acc[i][j][r];
-        if (EPI == EPI_BIAS_RELU) {
-          v += bval;
-          v = fmaxf(v, 0.0f);
-        } else if (EPI == EPI_MASK) {
-          float mv = bf16_to_f32(mask[row * N + col]);
-          v = mv > 0.0f ? v : 0.0f;
-        }
-        if (OUT_FP32)
-          ((float*)C)[row * N + col] = v;
-        else
-          ((bf16_t*)C)[row * N + col] = f32_to_bf16(v);
-      }
-    }
-  }
+  write_epilogue<EPI, HAS_BIAS, OUT_FP32>(acc, bias, mask, nullptr, C, M, N,
+                                          m0, n0, wm, wn, fl, kg);
 }
 
 // ---- launchers ------------------------------------------------------------
 
 static void launch_gemm(bool tn, int epi, bool has_bias, bool out_fp32,
                         const at::Tensor& a, const at::Tensor& b,
-                        const float* bias, const bf16_t* mask, at::Tensor& c,
+                        const float* bias, const unsigned char* mask,
+                        unsigned char* mask_out, at::Tensor& c,
                         long long M, long long N, long long K) {
   dim3 grid((N + BN - 1) / BN, (M + BM - 1) / BM);
   auto stream = at::cuda::getCurrentCUDAStream();
@@ -360,10 +354,19 @@ static void launch_gemm(bool tn, int epi, bool has_bias, bool out_fp32,
       return (e && e[0] == '1') ? 1 : 0;
     }();
 #define G_GLDS(EPI_, HB_, OF_)                                              \
-  hipLaunchKernelGGL((gemm_nt_glds_kernel<EPI_, HB_, OF_>), grid,           \
-                     dim3(GEMM_THREADS), 0, stream, ap, bp, bias, mask, cp, \
-                     M, N, K, xcd_swz)
-    if (epi == EPI_BIAS_RELU) {
+  hipLaunchKernelGGL((gemm_nt_glds_kernel<EPI_, HB_, OF_, false>), grid,    \
+                     dim3(GEMM_THREADS), 0, stream, ap, bp, bias, mask,     \
+                     nullptr, cp, M, N, K, xcd_swz)
+#define G_GLDS_EM(EPI_, HB_, OF_)                                           \
+  hipLaunchKernelGGL((gemm_nt_glds_kernel<EPI_, HB_, OF_, true>), grid,     \
+                     dim3(GEMM_THREADS), 0, stream, ap, bp, bias, mask,     \
+                     mask_out, cp, M, N, K, xcd_swz)
+    if (epi == EPI_BIAS_RELU && mask_out != nullptr) {
+      if (has_bias) { if (out_fp32) G_GLDS_EM(EPI_BIAS_RELU, true, true);
+                      else          G_GLDS_EM(EPI_BIAS_RELU, true, false); }
+      else          { if (out_fp32) G_GLDS_EM(EPI_BIAS_RELU, false, true);
+                      else          G_GLDS_EM(EPI_BIAS_RELU, false, false); }
+    } else if (epi == EPI_BIAS_RELU) {
       if (has_bias) { if (out_fp32) G_GLDS(EPI_BIAS_RELU, true, true);
                       else          G_GLDS(EPI_BIAS_RELU, true, false); }
       else          { if (out_fp32) G_GLDS(EPI_BIAS_RELU, false, true);
@@ -375,6 +378,7 @@ static void launch_gemm(bool tn, int epi, bool has_bias, bool out_fp32,
       if (out_fp32) G_GLDS(EPI_NONE, false, true);
       else          G_GLDS(EPI_NONE, false, false);
     }
+#undef G_GLDS_EM
 #undef G_GLDS
     return;
   }
@@ -383,6 +387,7 @@ static void launch_gemm(bool tn, int epi, bool has_bias, bool out_fp32,
   hipLaunchKernelGGL((gemm_bf16_kernel<TN_, EPI_, HB_, OF_>), grid,         \
                      dim3(GEMM_THREADS), 0, stream, ap, bp, bias, mask, cp, \
                      M, N, K)
+  /* mask_out unsupported on the fallback kernel (checked in hosts) */
 #define G_EPI(TN_)                                                          \
   do {                                                                      \
     if (epi == EPI_BIAS_RELU) {                                             \
@@ -430,11 +435,40 @@ at::Tensor linear_bf16_hip(const at::Tensor& x, const at::Tensor& w,
   // add would cost a pass, so clamp-free bias is handled here:
   TORCH_CHECK(!(bias.has_value() && !relu),
               "linear_bf16: bias currently requires relu epilogue");
-  const bf16_t* mp =
-      mask.has_value() ? (const bf16_t*)mask->data_ptr() : nullptr;
-  launch_gemm(false, epi, bias.has_value(), out_fp32, x, w, bp, mp, c, M, N,
-              K);
+  const unsigned char* mp = nullptr;
+  if (mask.has_value()) {
+    TORCH_CHECK(mask->scalar_type() == at::kByte &&
+                    mask->numel() == M * (N / 8) && N % 8 == 0,
+                "linear_bf16: mask must be a uint8 [M, N/8] relu bitmask");
+    mp = (const unsigned char*)mask->data_ptr();
+  }
+  launch_gemm(false, epi, bias.has_value(), out_fp32, x, w, bp, mp, nullptr,
+              c, M, N, K);
   return c;
+}
+
+// C, maskbits = relu(x @ w^T + bias): the forward hot path also emitting
+// the 1-bit relu mask the backward consumes (uint8 [M, N/8]).
+std::tuple<at::Tensor, at::Tensor> linear_relu_mask_bf16_hip(
+    const at::Tensor& x, const at::Tensor& w,
+    const c10::optional<at::Tensor>& bias) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 2 && w.dim() == 2);
+  TORCH_CHECK(x.scalar_type() == at::kBFloat16 &&
+              w.scalar_type() == at::kBFloat16);
+  long long M = x.size(0), K = x.size(1), N = w.size(0);
+  TORCH_CHECK(w.size(1) == K && N % 16 == 0 && K % BK == 0,
+              "linear_relu_mask: w [N,K], N%16==0, K%64==0");
+  auto c = at::empty({M, N}, x.options());
+  auto mask_out = at::empty({M, N / 8}, x.options().dtype(at::kByte));
+  at::Tensor bias_f;
+  const float* bp = nullptr;
+  if (bias.has_value()) {
+    bias_f = bias->to(at::kFloat);
+    bp = bias_f.data_ptr<float>();
+  }
+  launch_gemm(false, EPI_BIAS_RELU, bias.has_value(), false, x, w, bp,
+              nullptr, (unsigned char*)mask_out.data_ptr(), c, M, N, K);
+  return {c, mask_out};
 }
 
 // C = a^T @ b; a [R,M], b [R,N] (the dW backward shape)
@@ -447,7 +481,7 @@ at::Tensor gemm_tn_bf16_hip(const at::Tensor& a, const at::Tensor& b,
   TORCH_CHECK(b.size(0) == R, "gemm_tn: row counts must match");
   auto c = at::empty({M, N},
                      a.options().dtype(out_fp32 ? at::kFloat : at::kBFloat16));
-  launch_gemm(true, EPI_NONE, false, out_fp32, a, b, nullptr, nullptr, c, M,
-              N, R);
+  launch_gemm(true, EPI_NONE, false, out_fp32, a, b, nullptr, nullptr,
+              nullptr, c, M, N, R);
   return c;
 }

@@ -17,11 +17,16 @@
 
 // ---- expand1d -------------------------------------------------------------
 
-template <bool RELU, bool HAS_BIAS, bool HAS_MASK>
+// HAS_MASK: multiply by a 1-BIT ReLU mask (uint8 [n, H/8], bit e of byte
+// c = column 8c+e > 0) — 16x less mask traffic than re-reading the bf16
+// activations, and the mask stays L2-resident at MLP scale.
+// EMIT_MASK: write that bitmask for the activations this kernel produces.
+template <bool RELU, bool HAS_BIAS, bool HAS_MASK, bool EMIT_MASK>
 __global__ void expand1d_kernel(const float* __restrict__ x,
                                 const bf16_t* __restrict__ w,
                                 const bf16_t* __restrict__ b,
-                                const bf16_t* __restrict__ mask,
+                                const unsigned char* __restrict__ mask,
+                                unsigned char* __restrict__ mask_out,
                                 bf16_t* __restrict__ out, long long n,
                                 int h8 /* H/8 */) {
   const long long total = (long long)n * h8;
@@ -40,18 +45,24 @@ __global__ void expand1d_kernel(const float* __restrict__ x,
       if (RELU) acc[e] = fmaxf(acc[e], 0.0f);
     }
     if (HAS_MASK) {
-      float mv[8];
-      bf16x8_to_f32(load_bf16x8(mask + row * (long long)h8 * 8 + c8 * 8), mv);
+      unsigned char mb = mask[row * h8 + c8];
 #pragma unroll
-      for (int e = 0; e < 8; ++e) acc[e] = mv[e] > 0.0f ? acc[e] : 0.0f;
+      for (int e = 0; e < 8; ++e) acc[e] = (mb >> e) & 1 ? acc[e] : 0.0f;
+    }
+    if (EMIT_MASK) {
+      unsigned char mb = 0;
+#pragma unroll
+      for (int e = 0; e < 8; ++e) mb |= (acc[e] > 0.0f ? 1u : 0u) << e;
+      mask_out[row * h8 + c8] = mb;
     }
     store_bf16x8(out + row * (long long)h8 * 8 + c8 * 8, f32_to_bf16x8(acc));
   }
 }
 
-at::Tensor expand1d_bf16_hip(const at::Tensor& x, const at::Tensor& w,
-                             const c10::optional<at::Tensor>& b, bool relu,
-                             const c10::optional<at::Tensor>& mask) {
+std::tuple<at::Tensor, at::Tensor> expand1d_bf16_hip(
+    const at::Tensor& x, const at::Tensor& w,
+    const c10::optional<at::Tensor>& b, bool relu,
+    const c10::optional<at::Tensor>& mask, bool emit_mask) {
   TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kFloat);
   TORCH_CHECK(w.scalar_type() == at::kBFloat16);
   const long long n = x.numel();
@@ -60,26 +71,42 @@ at::Tensor expand1d_bf16_hip(const at::Tensor& x, const at::Tensor& w,
   auto out = at::empty({n, H}, w.options());
   const bool has_bias = b.has_value();
   const bool has_mask = mask.has_value();
+  at::Tensor mask_out;
+  if (emit_mask)
+    mask_out = at::empty({n, H / 8}, w.options().dtype(at::kByte));
+  else
+    mask_out = at::empty({0}, w.options().dtype(at::kByte));
+  if (has_mask)
+    TORCH_CHECK(mask->scalar_type() == at::kByte &&
+                mask->numel() == n * (H / 8),
+                "expand1d: mask must be uint8 [n, H/8] bitmask");
   auto stream = at::cuda::getCurrentCUDAStream();
   long long total = n * (H / 8);
   int grid = (int)std::min<long long>((total + 255) / 256, 2048);
   const bf16_t* wp = (const bf16_t*)w.data_ptr();
   const bf16_t* bp = has_bias ? (const bf16_t*)b->data_ptr() : nullptr;
-  const bf16_t* mp = has_mask ? (const bf16_t*)mask->data_ptr() : nullptr;
+  const unsigned char* mp =
+      has_mask ? (const unsigned char*)mask->data_ptr() : nullptr;
+  unsigned char* mop =
+      emit_mask ? (unsigned char*)mask_out.data_ptr() : nullptr;
 
-#define LAUNCH_E1D(R, B_, M_)                                                \
-  hipLaunchKernelGGL((expand1d_kernel<R, B_, M_>), dim3(grid), dim3(256), 0, \
-                     stream, x.data_ptr<float>(), wp, bp, mp,                \
-                     (bf16_t*)out.data_ptr(), n, (int)(H / 8))
+#define LAUNCH_E1D(R, B_, M_, E_)                                         \
+  hipLaunchKernelGGL((expand1d_kernel<R, B_, M_, E_>), dim3(grid),        \
+                     dim3(256), 0, stream, x.data_ptr<float>(), wp, bp,   \
+                     mp, mop, (bf16_t*)out.data_ptr(), n, (int)(H / 8))
+#define LAUNCH_E1D_E(R, B_, M_)                                           \
+  do { if (emit_mask) LAUNCH_E1D(R, B_, M_, true);                        \
+       else LAUNCH_E1D(R, B_, M_, false); } while (0)
   if (relu) {
-    if (has_bias) { if (has_mask) LAUNCH_E1D(true, true, true); else LAUNCH_E1D(true, true, false); }
-    else          { if (has_mask) LAUNCH_E1D(true, false, true); else LAUNCH_E1D(true, false, false); }
+    if (has_bias) { if (has_mask) LAUNCH_E1D_E(true, true, true); else LAUNCH_E1D_E(true, true, false); }
+    else          { if (has_mask) LAUNCH_E1D_E(true, false, true); else LAUNCH_E1D_E(true, false, false); }
   } else {
-    if (has_bias) { if (has_mask) LAUNCH_E1D(false, true, true); else LAUNCH_E1D(false, true, false); }
-    else          { if (has_mask) LAUNCH_E1D(false, false, true); else LAUNCH_E1D(false, false, false); }
+    if (has_bias) { if (has_mask) LAUNCH_E1D_E(false, true, true); else LAUNCH_E1D_E(false, true, false); }
+    else          { if (has_mask) LAUNCH_E1D_E(false, false, true); else LAUNCH_E1D_E(false, false, false); }
   }
+#undef LAUNCH_E1D_E
 #undef LAUNCH_E1D
-  return out;
+  return {out, mask_out};
 }
 
 // ---- rowdot ---------------------------------------------------------------

@@ -164,6 +164,22 @@ def score_label_sums_cpu(scores: torch.Tensor, labels: torch.Tensor):
 # GEMM oracle
 # --------------------------------------------------------------------------
 
+def pack_relu_mask(active: torch.Tensor) -> torch.Tensor:
+    """bool [n, H] -> uint8 [n, H/8] bitmask (bit e of byte c = col 8c+e),
+    matching the device kernels' layout."""
+    n, H = active.shape
+    bits = active.to(torch.uint8).reshape(n, H // 8, 8)
+    weights = (1 << torch.arange(8, dtype=torch.uint8))
+    return (bits * weights).sum(dim=2).to(torch.uint8)
+
+
+def unpack_relu_mask(maskbits: torch.Tensor, H: int) -> torch.Tensor:
+    """uint8 [n, H/8] -> bool [n, H]."""
+    n = maskbits.shape[0]
+    exp = maskbits.unsqueeze(2) >> torch.arange(8, dtype=torch.uint8)
+    return (exp & 1).reshape(n, H).bool()
+
+
 def linear_bf16_cpu(
     x: torch.Tensor,
     w: torch.Tensor,
@@ -178,8 +194,19 @@ def linear_bf16_cpu(
     if relu:
         c = torch.relu(c)
     if mask is not None:
-        c = c * (mask.float() > 0)
+        c = c * unpack_relu_mask(mask, c.shape[1])
     return c if out_fp32 else c.bfloat16()
+
+
+def linear_relu_mask_cpu(
+    x: torch.Tensor, w: torch.Tensor, bias: torch.Tensor | None = None
+) -> tuple[torch.Tensor, torch.Tensor]:
+    c = x.float() @ w.float().t()
+    if bias is not None:
+        c = c + bias.float()
+    c = torch.relu(c)
+    # mask tests the fp32 epilogue value (> 0), matching the device kernel
+    return c.bfloat16(), pack_relu_mask(c > 0)
 
 
 def gemm_tn_bf16_cpu(
@@ -195,14 +222,17 @@ def expand1d_cpu(
     b: torch.Tensor | None = None,
     relu: bool = False,
     mask: torch.Tensor | None = None,
-) -> torch.Tensor:
+    emit_mask: bool = False,
+):
     out = torch.outer(x.float(), w.float())
     if b is not None:
         out = out + b.float()
     if relu:
         out = torch.relu(out)
     if mask is not None:
-        out = out * (mask.float() > 0)
+        out = out * unpack_relu_mask(mask, out.shape[1])
+    if emit_mask:
+        return out.bfloat16(), pack_relu_mask(out > 0)
     return out.bfloat16()
 
 

@@ -117,15 +117,24 @@ class TestMLPOpsGPU:
         torch.testing.assert_close(got.cpu().float(), want.float(),
                                    rtol=1e-2, atol=1e-2)
 
-    def test_expand1d_mask(self):
+    def test_expand1d_bitmask_roundtrip(self):
         x = torch.randn(256, device=DEV)
         w = torch.randn(512, device=DEV).bfloat16()
-        mask = torch.randn(256, 512, device=DEV).bfloat16()
+        mask = reference.pack_relu_mask(
+            torch.rand(256, 512) > 0.5).to(DEV)
         got = ops.expand1d_bf16(x, w, None, relu=False, mask=mask)
         want = reference.expand1d_cpu(x.cpu(), w.cpu(), None, False,
                                       mask.cpu())
         torch.testing.assert_close(got.cpu().float(), want.float(),
                                    rtol=1e-2, atol=1e-2)
+
+    def test_expand1d_emit_mask(self):
+        x = torch.randn(128, device=DEV)
+        w = torch.randn(256, device=DEV).bfloat16()
+        b = torch.randn(256, device=DEV).bfloat16()
+        out, mbits = ops.expand1d_bf16(x, w, b, relu=True, emit_mask=True)
+        want_mask = reference.pack_relu_mask(out.cpu().float() > 0)
+        assert torch.equal(mbits.cpu(), want_mask)
 
     def test_rowdot(self):
         h = torch.randn(777, 4096, device=DEV).bfloat16()
@@ -178,14 +187,31 @@ class TestGemmGPU:
         assert self._relerr(got.cpu(), want.cpu()) < 2e-2
         assert (got.float() >= 0).all()
 
-    def test_linear_mask(self):
+    def test_linear_bitmask(self):
+        from bodywork_mlops_demo_amd.ops import reference
+
         x = (torch.randn(256, 512, device=DEV) * 0.5).bfloat16()
         w = (torch.randn(256, 512, device=DEV) * 0.5).bfloat16()
-        mask = torch.randn(256, 256, device=DEV).bfloat16()
+        active = torch.rand(256, 256) > 0.5
+        mask = reference.pack_relu_mask(active).to(DEV)
         got = ops.linear_bf16(x, w, mask=mask)
-        want = (x.float() @ w.float().t()) * (mask.float() > 0)
-        assert self._relerr(got.cpu(), want.cpu()) < 2e-2
-        assert (got.float()[mask.float() <= 0] == 0).all()
+        want = (x.float() @ w.float().t()).cpu() * active
+        assert self._relerr(got.cpu(), want) < 2e-2
+        assert (got.float().cpu()[~active] == 0).all()
+
+    def test_linear_relu_emit_mask(self):
+        from bodywork_mlops_demo_amd.ops import reference
+
+        x = (torch.randn(200, 128, device=DEV) * 0.5).bfloat16()
+        w = (torch.randn(256, 128, device=DEV) * 0.5).bfloat16()
+        b = torch.randn(256, device=DEV).bfloat16()
+        out, mbits = ops.linear_relu_mask_bf16(x, w, b)
+        assert (out.float() >= 0).all()
+        want_mask = reference.pack_relu_mask(out.cpu().float() > 0)
+        assert torch.equal(mbits.cpu(), want_mask)
+        # values match the plain bias+relu kernel
+        ref = ops.linear_bf16(x, w, bias=b, relu=True)
+        assert torch.equal(out, ref)
 
     @pytest.mark.parametrize("r,m,n", [
         (256, 128, 128),
