@@ -98,6 +98,25 @@ def test_metrics_csv_roundtrip(tmp_store):
     assert list(rec) == ["date", "MAPE", "r_squared", "max_residual"]
 
 
+def test_mixed_format_history(tmp_store):
+    """CSV and binary days interleave transparently in read-all."""
+    tmp_store.put_dataset(date(2026, 1, 1), np.full(2, 1.0, np.float32),
+                          np.full(2, 1.0, np.float32), fmt="csv")
+    tmp_store.put_dataset(date(2026, 1, 2), np.full(3, 2.0, np.float32),
+                          np.full(3, 2.0, np.float32), fmt="npy")
+    y, X, latest = tmp_store.get_all_datasets()
+    assert latest == date(2026, 1, 2)
+    assert y.tolist() == [1, 1, 2, 2, 2]
+
+
+def test_single_row_csv(tmp_store):
+    """np.genfromtxt returns 1-D for single-row files — must stay 2-D."""
+    tmp_store.put_dataset(date(2026, 1, 5), np.array([7.0], np.float32),
+                          np.array([3.0], np.float32), fmt="csv")
+    y, X = tmp_store.get_dataset("datasets/regression-dataset-2026-01-05.csv")
+    assert y.shape == (1,) and float(y[0]) == 7.0 and float(X[0]) == 3.0
+
+
 def test_atomic_write_and_key_escape(tmp_store):
     with pytest.raises(ValueError):
         tmp_store.put_bytes("../escape.txt", b"x")
