@@ -109,6 +109,31 @@ __device__ __forceinline__ void write_epilogue(
     long long N, long long m0, long long n0, int wm, int wn, int fl, int kg) {
 #pragma unroll
   for (int i = 0; i < 4; ++i) {
+    // EPI_MASK: one aligned u64 per row covers this wave's whole 64-col
+    // stripe (bit b = col n0+wn*64+b) — replaces 16 scalar byte loads
+    unsigned long long mrow[4];
+    if (EPI == EPI_MASK) {
+      const long long stripe = n0 + wn * 64;
+      const bool full = stripe + 64 <= N;  // edge tile: byte-wise gather
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        long long row = m0 + wm * 64 + i * 16 + kg * 4 + r;
+        if (row >= M) {
+          mrow[r] = 0ull;
+        } else if (full) {
+          mrow[r] = *(const unsigned long long*)(mask + row * (N >> 3) +
+                                                 (stripe >> 3));
+        } else {
+          unsigned long long v = 0;
+          for (int b8 = 0; b8 < 8; ++b8)
+            if (stripe + b8 * 8 < N)
+              v |= (unsigned long long)
+                       mask[row * (N >> 3) + ((stripe >> 3) + b8)]
+                   << (8 * b8);
+          mrow[r] = v;
+        }
+      }
+    }
 #pragma unroll
     for (int j = 0; j < 4; ++j) {
       long long col = n0 + wn * 64 + j * 16 + fl;
@@ -124,10 +149,7 @@ __device__ __forceinline__ void write_epilogue(
           v += bval;
           v = fmaxf(v, 0.0f);
         } else if (EPI == EPI_MASK) {
-          unsigned char mb = (row_ok && col_ok)
-                                 ? mask[row * (N >> 3) + (col >> 3)]
-                                 : (unsigned char)0;
-          v = (mb >> (fl & 7)) & 1 ? v : 0.0f;
+          v = (mrow[r] >> (j * 16 + fl)) & 1 ? v : 0.0f;
         }
         if (EMIT_MASK) {
           // ballot over the wave: bits l of b = lane l's v>0; this
