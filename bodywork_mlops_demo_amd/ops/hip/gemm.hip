@@ -352,6 +352,96 @@ __global__ void gemm_bf16_kernel(
                                           m0, n0, wm, wn, fl, kg);
 }
 
+// ---- 3-buffer deep-pipelined variant (A/B: BODYWORK_GEMM_NBUF=3) ----------
+// One raw barrier per K-tile; tile t+2 is staged while t computes, so each
+// glds has ~2 MFMA phases (~1100 cyc) to land instead of ~1 (the HBM
+// latency the 2-buffer __syncthreads drain exposes).  Costs LDS 96 KiB ->
+// 1 block/CU.  Invariant at iteration t after issuing t+2's stage: a
+// wave's outstanding glds are (t+1, t+2) = 16, so `vmcnt(16)` + barrier
+// guarantees tile t is landed and globally visible.
+
+template <int EPI, bool HAS_BIAS, bool OUT_FP32, bool EMIT_MASK = false>
+__launch_bounds__(GEMM_THREADS)
+__global__ void gemm_nt_glds3_kernel(
+    const bf16_t* __restrict__ A, const bf16_t* __restrict__ B,
+    const float* __restrict__ bias, const unsigned char* __restrict__ mask,
+    unsigned char* __restrict__ mask_out, void* __restrict__ C, long long M,
+    long long N, long long K) {
+  __shared__ short lds_all[3 * 2 * 128 * 64];  // [buf0..2][A/B][128][64]
+  const long long m0 = (long long)blockIdx.y * BM;
+  const long long n0 = (long long)blockIdx.x * BN;
+  const int wave = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int wm = wave >> 1, wn = wave & 1;
+  const int fl = lane & 15;
+  const int kg = lane >> 4;
+  const int swz = fl & 7;
+
+  f32x4 acc[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  const int TILE = 128 * 64;
+  const long long nt = K / BK;
+
+  stage_tile_glds(lds_all, A, K, m0, M, 0);
+  stage_tile_glds(lds_all + TILE, B, K, n0, N, 0);
+  if (nt > 1) {
+    stage_tile_glds(lds_all + 2 * TILE, A, K, m0, M, BK);
+    stage_tile_glds(lds_all + 3 * TILE, B, K, n0, N, BK);
+  }
+
+  for (long long t = 0; t < nt; ++t) {
+    const int cur = (int)(t % 3);
+    // 1. wait: own glds older than tile t+1's are landed -> tile t landed
+    if (t + 1 < nt) {
+      asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+    } else {
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    }
+    // 2. rendezvous: every wave has (a) finished its tile t-1 ds_reads
+    //    (last iteration, lgkm-waited) and (b) passed its own vmcnt ->
+    //    tile t is visible to all, and buf[(t+2)%3] (= t-1's) is free.
+    __builtin_amdgcn_s_barrier();
+    // 3. prefetch tile t+2 into the freed buffer (2 compute phases to land)
+    if (t + 2 < nt) {
+      const int nxt = (int)((t + 2) % 3);
+      stage_tile_glds(lds_all + nxt * 2 * TILE, A, K, m0, M, (t + 2) * BK);
+      stage_tile_glds(lds_all + nxt * 2 * TILE + TILE, B, K, n0, N,
+                      (t + 2) * BK);
+    }
+    short* as_cur = lds_all + cur * 2 * TILE;
+    short* bs_cur = as_cur + TILE;
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+      bf16x8_v a_frag[4], b_frag[4];
+      const int cidx = ((ks * 4 + kg) ^ swz) * 8;
+#pragma unroll
+      for (int f = 0; f < 4; ++f) {
+        a_frag[f] =
+            ((const lds_vec*)(as_cur + (wm * 64 + f * 16 + fl) * 64 + cidx))
+                ->v;
+        b_frag[f] =
+            ((const lds_vec*)(bs_cur + (wn * 64 + f * 16 + fl) * 64 + cidx))
+                ->v;
+      }
+#pragma unroll
+      for (int i = 0; i < 4; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a_frag[i], b_frag[j], acc[i][j], 0, 0, 0);
+    }
+    // all this tile's ds_reads are complete (lgkm-waited before their
+    // MFMAs), so after the NEXT iteration's barrier its buffer may be
+    // overwritten; no trailing barrier needed.
+  }
+  write_epilogue<EPI, HAS_BIAS, OUT_FP32, EMIT_MASK>(
+      acc, bias, mask, mask_out, C, M, N, m0, n0, wm, wn, fl, kg);
+}
+
 // ---- launchers ------------------------------------------------------------
 
 static void launch_gemm(bool tn, int epi, bool has_bias, bool out_fp32,
@@ -375,14 +465,32 @@ static void launch_gemm(bool tn, int epi, bool has_bias, bool out_fp32,
       const char* e = getenv("BODYWORK_GEMM_XCD");
       return (e && e[0] == '1') ? 1 : 0;
     }();
+    static int nbuf3 = [] {  // A/B: 3-buffer deep pipeline (96 KiB LDS)
+      const char* e = getenv("BODYWORK_GEMM_NBUF");
+      return (e && e[0] == '3') ? 1 : 0;
+    }();
 #define G_GLDS(EPI_, HB_, OF_)                                              \
-  hipLaunchKernelGGL((gemm_nt_glds_kernel<EPI_, HB_, OF_, false>), grid,    \
-                     dim3(GEMM_THREADS), 0, stream, ap, bp, bias, mask,     \
-                     nullptr, cp, M, N, K, xcd_swz)
+  do {                                                                      \
+    if (nbuf3)                                                              \
+      hipLaunchKernelGGL((gemm_nt_glds3_kernel<EPI_, HB_, OF_, false>),     \
+                         grid, dim3(GEMM_THREADS), 0, stream, ap, bp, bias, \
+                         mask, nullptr, cp, M, N, K);                       \
+    else                                                                    \
+      hipLaunchKernelGGL((gemm_nt_glds_kernel<EPI_, HB_, OF_, false>),      \
+                         grid, dim3(GEMM_THREADS), 0, stream, ap, bp, bias, \
+                         mask, nullptr, cp, M, N, K, xcd_swz);              \
+  } while (0)
 #define G_GLDS_EM(EPI_, HB_, OF_)                                           \
-  hipLaunchKernelGGL((gemm_nt_glds_kernel<EPI_, HB_, OF_, true>), grid,     \
-                     dim3(GEMM_THREADS), 0, stream, ap, bp, bias, mask,     \
-                     mask_out, cp, M, N, K, xcd_swz)
+  do {                                                                      \
+    if (nbuf3)                                                              \
+      hipLaunchKernelGGL((gemm_nt_glds3_kernel<EPI_, HB_, OF_, true>),      \
+                         grid, dim3(GEMM_THREADS), 0, stream, ap, bp, bias, \
+                         mask, mask_out, cp, M, N, K);                      \
+    else                                                                    \
+      hipLaunchKernelGGL((gemm_nt_glds_kernel<EPI_, HB_, OF_, true>),       \
+                         grid, dim3(GEMM_THREADS), 0, stream, ap, bp, bias, \
+                         mask, mask_out, cp, M, N, K, xcd_swz);             \
+  } while (0)
     if (epi == EPI_BIAS_RELU && mask_out != nullptr) {
       if (has_bias) { if (out_fp32) G_GLDS_EM(EPI_BIAS_RELU, true, true);
                       else          G_GLDS_EM(EPI_BIAS_RELU, true, false); }
