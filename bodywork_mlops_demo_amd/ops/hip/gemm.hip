@@ -455,6 +455,23 @@ static void launch_gemm(bool tn, int epi, bool has_bias, bool out_fp32,
   const bf16_t* bp = (const bf16_t*)b.data_ptr();
   void* cp = c.data_ptr();
 
+  // deep-pipeline 256^2 8-phase kernel (gemm8.hip) for large NT shapes
+  if (!tn && K % 128 == 0 && K >= 256 && M >= 256 && K > 0) {
+    static int use8 = [] {
+      const char* e = getenv("BODYWORK_GEMM_8PHASE");
+      return (e && e[0] == '1') ? 1 : 0;
+    }();
+    if (use8) {
+      extern void launch_gemm8(int, bool, bool, bool, const void*,
+                               const void*, const float*,
+                               const unsigned char*, unsigned char*, void*,
+                               long long, long long, long long);
+      launch_gemm8(epi, has_bias, out_fp32, mask_out != nullptr, ap, bp,
+                   bias, mask, mask_out, cp, M, N, K);
+      return;
+    }
+  }
+
   // hot path: NT with K a tile multiple -> glds-staged kernel
   if (!tn && K % BK == 0 && K > 0) {
     // A/B-measured on MI355X: the T1 remap LOSES 4-14% on these NT
