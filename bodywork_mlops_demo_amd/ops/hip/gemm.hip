@@ -455,8 +455,9 @@ static void launch_gemm(bool tn, int epi, bool has_bias, bool out_fp32,
   const bf16_t* bp = (const bf16_t*)b.data_ptr();
   void* cp = c.data_ptr();
 
-  // deep-pipeline 256^2 8-phase kernel (gemm8.hip) for large NT shapes
-  if (!tn && K % 128 == 0 && K >= 256 && M >= 256 && K > 0) {
+  // deep-pipeline 256^2 8-phase kernel (gemm8.hip) for exact-tile shapes
+  if (!tn && K % 128 == 0 && K >= 256 && K <= (1 << 22) && M % 256 == 0 &&
+      N % 256 == 0 && M >= 256) {
     static int use8 = [] {
       const char* e = getenv("BODYWORK_GEMM_8PHASE");
       return (e && e[0] == '1') ? 1 : 0;

@@ -58,21 +58,17 @@ __device__ __forceinline__ void g8_glds16(const void* gsrc, void* lds_dst) {
 
 // stage one [128][64]-bf16 half-tile: 1024 16-B chunks, 512 threads -> 2
 // wave-level glds per wave.  Lane-linear LDS; XOR swizzle on the source.
-__device__ __forceinline__ void g8_stage_half(
-    short* __restrict__ slot, const bf16_t* __restrict__ src, long long ld,
-    long long row0, long long row_max, long long k0) {
-  const int lane = threadIdx.x & 63;
+// The per-lane source BYTE offsets (row*K + swizzled column chunk) are
+// tile-invariant — computed once at kernel entry (off0/off1, 32-bit) so
+// the hot loop's staging is {uniform base + int offset} with no 64-bit
+// per-lane math (this kernel is dispatched only for M,N % 256 == 0, so
+// no row clamping is needed).
+__device__ __forceinline__ void g8_stage_half(short* __restrict__ slot,
+                                              const char* __restrict__ base,
+                                              int off0, int off1) {
   const int wave = threadIdx.x >> 6;
-#pragma unroll
-  for (int s = 0; s < 2; ++s) {
-    int ci0 = (s * 8 + wave) * 64;
-    int ci = ci0 + lane;
-    int row = ci >> 3;
-    int sc = (ci & 7) ^ (row & 7);
-    long long gr = row0 + row;
-    if (gr >= row_max) gr = row_max - 1;  // clamp; junk masked in epilogue
-    g8_glds16(src + gr * ld + k0 + sc * 8, (char*)slot + ci0 * 16);
-  }
+  g8_glds16(base + off0, (char*)slot + (0 * 8 + wave) * 1024);
+  g8_glds16(base + off1, (char*)slot + (1 * 8 + wave) * 1024);
 }
 
 template <int EPI, bool HAS_BIAS, bool OUT_FP32, bool EMIT_MASK>
@@ -190,16 +186,35 @@ __global__ void gemm_nt_8phase_kernel(
   bf16x8_v a_set0[2][2], a_set1[2][2];  // [mi][ks] — two quad sets
   bf16x8_v bA[4][2], bB[4][2];          // [nfrag][ks] — even/odd tile B
 
+  // per-lane staging byte offsets (tile-invariant; see g8_stage_half)
+  int stg_off[2];
+  {
+    const int lane_ = threadIdx.x & 63;
+    const int wave_ = threadIdx.x >> 6;
+#pragma unroll
+    for (int s = 0; s < 2; ++s) {
+      int ci = (s * 8 + wave_) * 64 + lane_;
+      int row = ci >> 3;
+      int sc = (ci & 7) ^ (row & 7);
+      stg_off[s] = (int)((row * K + sc * 8) * 2);
+    }
+  }
+  // uniform base pointers: A halves at rows m0 / m0+128, B at n0 / n0+128
+  const char* Ah0 = (const char*)(A + m0 * K);
+  const char* Ah1 = (const char*)(A + (m0 + 128) * K);
+  const char* Bh0 = (const char*)(B + n0 * K);
+  const char* Bh1 = (const char*)(B + (n0 + 128) * K);
+#define G8_KOFF(T) ((long long)(T) * (G8_BK * 2))
+
   // ---- prologue: 7 half-tiles, then land tile 0 -------------------------
-  // A halves: rows m0+half*128; B halves: rows n0+half*128 (B is [N,K])
-  g8_stage_half(G8_ASLOT(0, 0), A, K, m0, M, 0);
-  g8_stage_half(G8_ASLOT(0, 1), A, K, m0 + 128, M, 0);
-  g8_stage_half(G8_BSLOT(0, 0), B, K, n0, N, 0);
-  g8_stage_half(G8_BSLOT(0, 1), B, K, n0 + 128, N, 0);
+  g8_stage_half(G8_ASLOT(0, 0), Ah0, stg_off[0], stg_off[1]);
+  g8_stage_half(G8_ASLOT(0, 1), Ah1, stg_off[0], stg_off[1]);
+  g8_stage_half(G8_BSLOT(0, 0), Bh0, stg_off[0], stg_off[1]);
+  g8_stage_half(G8_BSLOT(0, 1), Bh1, stg_off[0], stg_off[1]);
   if (nt > 1) {
-    g8_stage_half(G8_BSLOT(1, 0), B, K, n0, N, G8_BK);
-    g8_stage_half(G8_BSLOT(1, 1), B, K, n0 + 128, N, G8_BK);
-    g8_stage_half(G8_ASLOT(1, 0), A, K, m0, M, G8_BK);
+    g8_stage_half(G8_BSLOT(1, 0), Bh0 + G8_KOFF(1), stg_off[0], stg_off[1]);
+    g8_stage_half(G8_BSLOT(1, 1), Bh1 + G8_KOFF(1), stg_off[0], stg_off[1]);
+    g8_stage_half(G8_ASLOT(1, 0), Ah0 + G8_KOFF(1), stg_off[0], stg_off[1]);
   }
   asm volatile("s_waitcnt vmcnt(6)" ::: "memory");  // tile 0 landed
   __builtin_amdgcn_s_barrier();
@@ -227,18 +242,20 @@ __global__ void gemm_nt_8phase_kernel(
     /* issue schedule */                                                    \
     if (Q == 0) {                                                           \
       if ((T) + 1 < nt)                                                     \
-        g8_stage_half(G8_ASLOT(((TPAR) ^ 1), 1), A, K, m0 + 128, M,         \
-                      ((T) + 1) * G8_BK);                                   \
+        g8_stage_half(G8_ASLOT(((TPAR) ^ 1), 1), Ah1 + G8_KOFF((T) + 1),    \
+                      stg_off[0], stg_off[1]);                              \
     } else if (Q == 1) {                                                    \
       if ((T) + 2 < nt)                                                     \
-        g8_stage_half(G8_BSLOT(TPAR, 0), B, K, n0, N, ((T) + 2) * G8_BK);   \
+        g8_stage_half(G8_BSLOT(TPAR, 0), Bh0 + G8_KOFF((T) + 2),            \
+                      stg_off[0], stg_off[1]);                              \
     } else if (Q == 2) {                                                    \
       if ((T) + 2 < nt)                                                     \
-        g8_stage_half(G8_BSLOT(TPAR, 1), B, K, n0 + 128, N,                 \
-                      ((T) + 2) * G8_BK);                                   \
+        g8_stage_half(G8_BSLOT(TPAR, 1), Bh1 + G8_KOFF((T) + 2),            \
+                      stg_off[0], stg_off[1]);                              \
     } else {                                                                \
       if ((T) + 2 < nt)                                                     \
-        g8_stage_half(G8_ASLOT(TPAR, 0), A, K, m0, M, ((T) + 2) * G8_BK);   \
+        g8_stage_half(G8_ASLOT(TPAR, 0), Ah0 + G8_KOFF((T) + 2),            \
+                      stg_off[0], stg_off[1]);                              \
     }                                                                       \
     /* reads for the NEXT phase */                                          \
     if (Q < 3) {                                                            \
