@@ -103,6 +103,37 @@ def test_runner_end_to_end_subprocesses(tmp_path):
     assert not runner.services
 
 
+@pytest.mark.timeout(300)
+def test_runner_parallel_dag_step(tmp_path):
+    """Comma fan-out in the DAG string runs batch stages concurrently."""
+    cfg = load_config("""
+version: "1.0"
+project:
+  name: fanout
+  DAG: gen-a,gen-b >> stage-1-train-model
+stages:
+  gen-a:
+    executable_module_path: bodywork_mlops_demo_amd/stages/datagen.py
+    args: ["--n", "300", "--date", "2026-06-01"]
+    batch: {max_completion_time_seconds: 120, retries: 1}
+  gen-b:
+    executable_module_path: bodywork_mlops_demo_amd/stages/datagen.py
+    args: ["--n", "300", "--date", "2026-06-02"]
+    batch: {max_completion_time_seconds: 120, retries: 1}
+  stage-1-train-model:
+    executable_module_path: bodywork_mlops_demo_amd/stages/train.py
+    batch: {max_completion_time_seconds: 120, retries: 1}
+""")
+    store_dir = str(tmp_path / "store")
+    report = PipelineRunner(cfg, store_uri=store_dir, n_gpus=0).run()
+    assert report.ok, report.failed
+    from bodywork_mlops_demo_amd.store import LocalStore
+
+    store = LocalStore(store_dir)
+    assert len(store.list_keys(contract.DATASETS_PREFIX)) == 2
+    assert len(store.list_keys(contract.MODELS_PREFIX)) == 1
+
+
 def test_runner_retries_failing_stage(tmp_path):
     cfg = load_config("""
 version: "1.0"
