@@ -65,24 +65,42 @@ class CycleState:
             f.result()  # re-raises write errors
         self._io_futures.clear()
 
-    def fast_cpu(self, name: str, t: torch.Tensor):
-        """Device->host via a cached pinned staging buffer (~10x the
-        pageable-copy rate).  The returned numpy view aliases the pinned
-        buffer, which is only reused after drain_io() — submit_io callers
-        drain before the next copy."""
-        if t.device.type != "cuda":
-            return t.numpy()
-        buf = getattr(self, "_pin", None)
-        if buf is None:
+    def _pin_view(self, name: str, numel: int, dtype) -> torch.Tensor:
+        if getattr(self, "_pin", None) is None:
             self._pin = {}
         cur = self._pin.get(name)
-        if cur is None or cur.numel() < t.numel():
-            cur = torch.empty(t.numel(), dtype=t.dtype, pin_memory=True)
+        if cur is None or cur.numel() < numel:
+            cur = torch.empty(numel, dtype=dtype, pin_memory=True)
             self._pin[name] = cur
-        view = cur[: t.numel()]
-        view.copy_(t, non_blocking=True)
-        torch.cuda.synchronize()
-        return view.numpy()
+        return cur[:numel]
+
+    def persist_async(self, store, d, y: torch.Tensor, X: torch.Tensor,
+                      fmt: str) -> None:
+        """Persist a day's dataset without blocking the pipeline: the D2H
+        copies run on a side stream through cached pinned buffers
+        (overlapping the next stage's compute) and the file write runs on
+        the I/O thread after the copy event fires.  Pinned buffers are
+        reused only after drain_io()."""
+        if y.device.type != "cuda":
+            self.submit_io(store.put_dataset, d, y.numpy(), X.numpy(), fmt)
+            return
+        self.drain_io()  # pinned staging buffers about to be reused
+        yv = self._pin_view("y", y.numel(), y.dtype)
+        Xv = self._pin_view("X", X.numel(), X.dtype)
+        if getattr(self, "_io_stream", None) is None:
+            self._io_stream = torch.cuda.Stream()
+        ev = torch.cuda.Event()
+        self._io_stream.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(self._io_stream):
+            yv.copy_(y, non_blocking=True)
+            Xv.copy_(X, non_blocking=True)
+            ev.record()
+
+        def _write():
+            ev.synchronize()
+            store.put_dataset(d, yv.numpy(), Xv.numpy(), fmt)
+
+        self.submit_io(_write)
 
     def append_day(self, y: torch.Tensor, X: torch.Tensor) -> None:
         self.y = torch.cat([self.y, y])
@@ -129,10 +147,7 @@ def run_cycle(
         )
         state.append_day(y, X)
         if store is not None and state.rank == 0:
-            state.drain_io()  # pinned staging buffers about to be reused
-            state.submit_io(store.put_dataset, state.date,
-                            state.fast_cpu("y", y), state.fast_cpu("X", X),
-                            persist_fmt)
+            state.persist_async(store, state.date, y, X, persist_fmt)
 
     # -- stage 1: train on all accumulated data -----------------------------
     sync()
@@ -198,12 +213,9 @@ def run_cycle(
         n_rows, next_date.timetuple().tm_yday, seed, device=device,
         stream_offset=state.rank * n_rows,
     )
-    if store is not None and state.rank == 0:
-        state.drain_io()  # pinned staging buffers about to be reused
-        state.submit_io(store.put_dataset, next_date,
-                        state.fast_cpu("y", y_next),
-                        state.fast_cpu("X", X_next), persist_fmt)
     sync()
+    if store is not None and state.rank == 0:
+        state.persist_async(store, next_date, y_next, X_next, persist_fmt)
     timings["datagen_s"] = perf_counter() - t0
 
     # -- stage 4: test the deployed model on unseen t+1 data ----------------
