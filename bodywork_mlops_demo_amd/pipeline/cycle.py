@@ -84,9 +84,19 @@ class CycleState:
         if y.device.type != "cuda":
             self.submit_io(store.put_dataset, d, y.numpy(), X.numpy(), fmt)
             return
-        self.drain_io()  # pinned staging buffers about to be reused
-        yv = self._pin_view("y", y.numel(), y.dtype)
-        Xv = self._pin_view("X", X.numel(), X.dtype)
+        # double-buffered pinned staging: only the write that used THIS
+        # buffer pair (two cycles ago) must be awaited — with the serial
+        # I/O worker that future is normally long done, so the pipeline
+        # never stalls on its own artefact I/O
+        flip = getattr(self, "_pin_flip", 0) ^ 1
+        self._pin_flip = flip
+        if getattr(self, "_pin_futures", None) is None:
+            self._pin_futures = {}
+        prev = self._pin_futures.get(flip)
+        if prev is not None:
+            prev.result()
+        yv = self._pin_view(f"y{flip}", y.numel(), y.dtype)
+        Xv = self._pin_view(f"X{flip}", X.numel(), X.dtype)
         if getattr(self, "_io_stream", None) is None:
             self._io_stream = torch.cuda.Stream()
         ev = torch.cuda.Event()
@@ -100,7 +110,9 @@ class CycleState:
             ev.synchronize()
             store.put_dataset(d, yv.numpy(), Xv.numpy(), fmt)
 
+        fut_idx = len(self._io_futures)
         self.submit_io(_write)
+        self._pin_futures[flip] = self._io_futures[fut_idx]
 
     def append_day(self, y: torch.Tensor, X: torch.Tensor) -> None:
         self.y = torch.cat([self.y, y])
