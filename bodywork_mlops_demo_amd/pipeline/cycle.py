@@ -49,10 +49,11 @@ class CycleState:
         self._io_pool = None
         self._io_futures: list = []
 
-    # -- async artefact I/O: dataset writes overlap the next cycle's
-    #    compute (the D2H copy is synchronous; only the file write is
-    #    deferred).  drain before reading artefacts back or stopping a
-    #    benchmark clock.
+    # -- async artefact I/O: dataset D2H copies run on a side stream
+    #    through double-buffered pinned staging and the file write runs
+    #    on the I/O thread after the copy event fires, overlapping the
+    #    next stages' compute.  drain_io() before reading artefacts back
+    #    or stopping a benchmark clock.
     def submit_io(self, fn, *args) -> None:
         if self._io_pool is None:
             from concurrent.futures import ThreadPoolExecutor
@@ -105,6 +106,11 @@ class CycleState:
             yv.copy_(y, non_blocking=True)
             Xv.copy_(X, non_blocking=True)
             ev.record()
+        # the source tensors are consumed on the side stream: tell the
+        # caching allocator, or their blocks could be handed to a later
+        # default-stream allocation while the copy is still in flight
+        y.record_stream(self._io_stream)
+        X.record_stream(self._io_stream)
 
         def _write():
             ev.synchronize()
