@@ -54,6 +54,23 @@ class TestDatagenGPU:
         mismatch = (Xg[:n].cpu() != Xc[:n]).sum().item()
         assert mismatch <= 2  # only boundary-flip induced shifts allowed
 
+    @pytest.mark.parametrize("n", [1, 7, 255, 256, 257, 1023, 262145])
+    def test_edge_sizes(self, n):
+        """Scan/compaction boundaries: sub-block, block-exact, chunk+1."""
+        yg, Xg = ops.datagen(n, 30, seed=11, device=DEV)
+        yc, Xc = ops.datagen(n, 30, seed=11, device="cpu")
+        assert abs(yg.shape[0] - yc.shape[0]) <= 2
+        m = min(yg.shape[0], yc.shape[0])
+        assert (Xg[:m].cpu() != Xc[:m]).sum().item() <= 2
+
+        X = torch.rand(max(n, 1), device=DEV)
+        y = X * 2
+        Xtr, ytr, Xte, yte = ops.random_split(X, y, 0.2, seed=3)
+        assert Xtr.shape[0] + Xte.shape[0] == X.shape[0]
+        Xtr_c, _, Xte_c, _ = ops.random_split(X.cpu(), y.cpu(), 0.2, seed=3)
+        assert torch.equal(Xtr.cpu(), Xtr_c)
+        assert torch.equal(Xte.cpu(), Xte_c)
+
     def test_large_n(self):
         y, X = ops.datagen(50_000_000, 10, seed=5, device=DEV)
         assert y.shape[0] > 25_000_000
