@@ -48,6 +48,14 @@ class CycleState:
         self._day_sizes: list[int] = []
         self._io_pool = None
         self._io_futures: list = []
+        self.cycle_count = 0
+
+    @property
+    def persist_rank(self) -> int:
+        """Dataset-persist duty rotates across DP ranks per cycle so no
+        single rank is the artefact-I/O straggler in the weak-scaling
+        bench (any rank's shard is an equally representative artefact)."""
+        return self.cycle_count % max(self.world_size, 1)
 
     # -- async artefact I/O: dataset D2H copies run on a side stream
     #    through double-buffered pinned staging and the file write runs
@@ -232,7 +240,7 @@ def run_cycle(
         stream_offset=state.rank * n_rows,
     )
     sync()
-    if store is not None and state.rank == 0:
+    if store is not None and state.rank == state.persist_rank:
         state.persist_async(store, next_date, y_next, X_next, persist_fmt)
     timings["datagen_s"] = perf_counter() - t0
 
@@ -252,6 +260,7 @@ def run_cycle(
     # -- advance the clock ---------------------------------------------------
     state.append_day(y_next, X_next)
     state.date = next_date
+    state.cycle_count += 1
 
     timings["cycle_s"] = sum(
         v for k, v in timings.items() if k.endswith("_s")
