@@ -15,9 +15,11 @@
 //               into the same K-contiguous LDS images.
 //
 // Geometry: 128x128 C-tile per 256-thread block (4 waves as 2x2, each
-// wave a 64x64 sub-tile = 4x4 fragments of 16x16), BK=64, LDS rows padded
-// to 72 shorts (row stride 144 B = 36 banks: a 16-lane fragment-read group
-// touches 16 distinct banks - conflict-free without an XOR swizzle).
+// wave a 64x64 sub-tile = 4x4 fragments of 16x16), BK=64.
+// LDS layouts: the glds hot path uses lane-linear [128][64] tiles with a
+// 16-B-chunk XOR swizzle carried on the SOURCE address and the fragment
+// read (both-sides rule; PMC-measured 0 bank conflicts); the fallback
+// register-staged kernels pad rows to 72 shorts (144-B stride) instead.
 // fp32 C/D per guide §3: col = lane&15, row = (lane>>4)*4 + reg.
 #include <ATen/cuda/CUDAContext.h>
 #include <hip/hip_runtime.h>
