@@ -1,0 +1,52 @@
+#!/usr/bin/env bash
+# Regenerate every BASELINE.md measurement on an MI355X box.
+# Usage (on a GPU box, repo root):  bash tools/collect_baseline.sh [outdir]
+# Or through gpurun:
+#   gpurun --timeout 3000 -- 'bash tools/collect_baseline.sh gpurun_out/baseline'
+set -uo pipefail
+OUT=${1:-gpurun_out/baseline}
+mkdir -p "$OUT"
+log() { echo "== $*" | tee -a "$OUT/summary.txt"; }
+
+log "GPU test suite"
+python -m pytest tests -m gpu -q 2>&1 | tail -1 | tee -a "$OUT/summary.txt"
+
+log "config 2: linear 10M-row cycle (steady state)"
+timeout 400 python bench.py --rows 10000000 --steps 10 --warmup 3 \
+  > "$OUT/bench_linear_10m.json" 2>"$OUT/bench_linear_10m.log"
+tail -1 "$OUT/bench_linear_10m.json" | tee -a "$OUT/summary.txt"
+
+log "config 3 scale/GPU: 125M rows"
+timeout 700 python bench.py --rows 125000000 --steps 2 --warmup 1 \
+  > "$OUT/bench_linear_125m.json" 2>/dev/null
+tail -1 "$OUT/bench_linear_125m.json" | tee -a "$OUT/summary.txt"
+
+log "config 5: MLP-4096, 1M rows"
+timeout 700 python bench.py --model mlp --rows 1000000 --steps 3 --warmup 2 \
+  --mlp-steps 20 > "$OUT/bench_mlp_1m.json" 2>/dev/null
+tail -1 "$OUT/bench_mlp_1m.json" | tee -a "$OUT/summary.txt"
+
+log "config 4: 30-day drift loop (reference scale)"
+timeout 500 python -m bodywork_mlops_demo_amd loop --days 30 --rows 1440 \
+  --store "$OUT/loopstore" --device cuda:0 \
+  --json-out "$OUT/loop30.json" > "$OUT/loop30.log" 2>&1
+python -m bodywork_mlops_demo_amd analytics --store "$OUT/loopstore" \
+  2>/dev/null | tail -2 | tee -a "$OUT/summary.txt"
+
+log "GEMM microbench"
+python - 2>/dev/null <<'EOF' | tee -a "$OUT/summary.txt"
+import torch, time
+from bodywork_mlops_demo_amd import ops
+def bench(fn, n=20):
+    for _ in range(3): fn()
+    torch.cuda.synchronize(); t0 = time.perf_counter()
+    for _ in range(n): fn()
+    torch.cuda.synchronize(); return (time.perf_counter() - t0) / n
+for M, N, K in [(4096, 4096, 4096), (8192, 8192, 8192)]:
+    x = (torch.randn(M, K, device="cuda") * 0.5).bfloat16()
+    w = (torch.randn(N, K, device="cuda") * 0.5).bfloat16()
+    t = bench(lambda: ops.linear_bf16(x, w))
+    print(f"NT GEMM {M}x{N}x{K}: {2*M*N*K/t/1e12:.0f} TF")
+EOF
+
+log "done; see $OUT/"
