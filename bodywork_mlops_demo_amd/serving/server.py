@@ -15,6 +15,13 @@ Extensions beyond the reference (batched GPU serving):
 Served by uvicorn (ASGI) instead of the reference's Flask dev server —
 one replica process per GPU, model resident in HBM, hipGraph-captured
 batch scoring.
+
+Concurrency model (deliberate): handlers are async but call the scorer
+synchronously, so scoring requests SERIALIZE within a replica — the
+scorer's static graph buffers are single-stream and one GPU stream is
+already saturated by a single large batch.  Throughput scales by replica
+fan-out (one process per GPU, the reference's ``replicas`` semantics),
+not by intra-process concurrency.
 """
 from __future__ import annotations
 
