@@ -37,8 +37,9 @@ def main() -> None:
     p.add_argument("--warmup", type=int, default=2)
     p.add_argument("--rows", type=int, default=10_000_000,
                    help="synthetic rows per GPU per day (BASELINE config 2)")
-    p.add_argument("--model", default="linear", choices=["linear", "mlp"],
-                   help="mlp = BASELINE config 5 (4096-d MFMA GEMM path)")
+    p.add_argument("--model", default="linear",
+                   help="linear | poly[<degree>] | mlp "
+                        "(mlp = BASELINE config 5, 4096-d MFMA GEMM path)")
     p.add_argument("--mlp-steps", type=int, default=50)
     p.add_argument("--mlp-batch", type=int, default=65536)
     p.add_argument("--no-graphs", action="store_true")
@@ -158,8 +159,8 @@ def main() -> None:
             "data": "synthetic (on-GPU philox drift generator, "
                     f"{args.rows} rows/GPU/day, random-init weights)",
             "config": {
-                "model": ("mlp-4096x2" if args.model == "mlp"
-                          else "linear-ols"),
+                "model": {"linear": "linear-ols",
+                          "mlp": "mlp-4096x2"}.get(args.model, args.model),
                 "rows_per_gpu_per_day": args.rows,
                 "history_days": 1,
                 "parallelism": f"dp{args.gpus}",

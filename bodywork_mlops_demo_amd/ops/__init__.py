@@ -143,6 +143,71 @@ def solve_ols(stats: torch.Tensor) -> tuple[float, float]:
     return intercept, coef
 
 
+def poly_stats(
+    X: torch.Tensor,
+    y: torch.Tensor,
+    degree: int,
+    mu: float = 50.0,
+    s: float = 50.0,
+) -> torch.Tensor:
+    """Fused sufficient statistics for polynomial ridge/OLS.
+
+    The k-feature generalisation of :func:`linreg_stats` (SURVEY §2.2
+    mapping table): one pass over (X, y) producing fp64
+    ``[n, upper-tri(PhiT Phi), PhiT y]`` over the IMPLICIT normalised
+    basis ``phi_j = ((x-mu)/s)^j`` — the design matrix never touches
+    HBM.  These scalars are the whole DP all-reduce payload.
+    """
+    nf = degree + 1
+    if X.device.type == "cuda":
+        core = _core(X.device)
+        return core.poly_stats(X.contiguous().float(),
+                               y.contiguous().float(), nf, mu, s)
+    return reference.poly_stats_cpu(X, y, nf, mu, s)
+
+
+def solve_poly(
+    stats: torch.Tensor, degree: int, l2: float = 0.0
+) -> list[float]:
+    """Normalised-basis coefficients from fused stats (host-side solve;
+    the intercept term is not penalised)."""
+    import numpy as np
+
+    nf = degree + 1
+    tri = nf * (nf + 1) // 2
+    vals = stats.cpu().numpy()
+    A = np.zeros((nf, nf))
+    k = 1
+    for a in range(nf):
+        for b in range(a, nf):
+            A[a, b] = A[b, a] = vals[k]
+            k += 1
+    bvec = vals[1 + tri:1 + tri + nf]
+    if l2 > 0:
+        reg = np.eye(nf) * l2 * vals[0]
+        reg[0, 0] = 0.0  # do not penalise the intercept
+        A = A + reg
+    return np.linalg.solve(A, bvec).tolist()
+
+
+def poly_score(
+    X: torch.Tensor,
+    coef: torch.Tensor | list[float],
+    mu: float = 50.0,
+    s: float = 50.0,
+) -> torch.Tensor:
+    """Horner evaluation of the normalised-basis polynomial (fp32).
+
+    ``coef`` may be a device tensor (read in-kernel, so captured serving
+    graphs follow redeployed coefficients)."""
+    if X.device.type == "cuda":
+        core = _core(X.device)
+        if not torch.is_tensor(coef):
+            coef = torch.tensor(coef, device=X.device, dtype=torch.float32)
+        return core.poly_score(X.contiguous().float(), coef.float(), mu, s)
+    return reference.poly_score_cpu(X, coef, mu, s)
+
+
 # --------------------------------------------------------------------------
 # scoring — reference stage_2:78 (model.predict)
 # --------------------------------------------------------------------------

@@ -14,6 +14,10 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor> random_split_hip(
 // linreg.hip
 at::Tensor linreg_stats_hip(const at::Tensor& x, const at::Tensor& y);
 at::Tensor linear_score_hip(const at::Tensor& x, const at::Tensor& ab);
+at::Tensor poly_stats_hip(const at::Tensor& x, const at::Tensor& y,
+                          int64_t nf, double mu, double s);
+at::Tensor poly_score_hip(const at::Tensor& x, const at::Tensor& coef,
+                          double mu, double s);
 at::Tensor regression_metrics_hip(const at::Tensor& y, const at::Tensor& yhat);
 at::Tensor score_label_metrics_hip(const at::Tensor& s, const at::Tensor& l);
 // mlp_small.hip
@@ -53,6 +57,8 @@ TORCH_LIBRARY(bodywork_hip, m) {
         "(Tensor, Tensor, Tensor, Tensor)");
   m.def("linreg_stats(Tensor x, Tensor y) -> Tensor");
   m.def("linear_score(Tensor x, Tensor ab) -> Tensor");
+  m.def("poly_stats(Tensor x, Tensor y, int nf, float mu, float s) -> Tensor");
+  m.def("poly_score(Tensor x, Tensor coef, float mu, float s) -> Tensor");
   m.def("regression_metrics(Tensor y, Tensor yhat) -> Tensor");
   m.def("score_label_metrics(Tensor s, Tensor l) -> Tensor");
   m.def("expand1d_bf16(Tensor x, Tensor w, Tensor? b, bool relu, "
@@ -79,6 +85,8 @@ TORCH_LIBRARY_IMPL(bodywork_hip, CUDA, m) {
   m.impl("random_split", random_split_hip);
   m.impl("linreg_stats", linreg_stats_hip);
   m.impl("linear_score", linear_score_hip);
+  m.impl("poly_stats", poly_stats_hip);
+  m.impl("poly_score", poly_score_hip);
   m.impl("regression_metrics", regression_metrics_hip);
   m.impl("score_label_metrics", score_label_metrics_hip);
   m.impl("expand1d_bf16", expand1d_bf16_hip);

@@ -115,6 +115,113 @@ at::Tensor linear_score_hip(const at::Tensor& x, const at::Tensor& ab) {
   return out;
 }
 
+// ---- polynomial ridge: fused X^T X / X^T y over an IMPLICIT basis --------
+// The k-feature generalisation of linreg_stats promised by the SURVEY
+// mapping table: sufficient statistics for ridge/OLS over the normalised
+// polynomial basis phi_j(x) = t^j, t = (x - mu)/s, j = 0..NF-1.  The
+// design matrix is never materialised — each thread expands its rows'
+// basis on the fly and accumulates the upper-triangular X^T X plus X^T y
+// in fp64, so the pass reads exactly the 8 bytes/row the linear fit
+// reads regardless of degree.  Output layout:
+//   out[0]                    = n
+//   out[1 .. T]               = upper triangle of X^T X, row-major
+//                               (T = NF*(NF+1)/2)
+//   out[T+1 .. T+NF]          = X^T y
+// Host side solves the (ridge-regularised) normal equations.
+
+template <int NF>
+__global__ void poly_stats_kernel(const float* __restrict__ x,
+                                  const float* __restrict__ y,
+                                  double* __restrict__ out, long long n,
+                                  float mu, float inv_s) {
+  constexpr int TRI = NF * (NF + 1) / 2;
+  double acc[TRI + NF];
+#pragma unroll
+  for (int i = 0; i < TRI + NF; ++i) acc[i] = 0.0;
+  const long long stride = (long long)gridDim.x * RED_BLOCK;
+  for (long long j = (long long)blockIdx.x * RED_BLOCK + threadIdx.x; j < n;
+       j += stride) {
+    double t = (double)((x[j] - mu) * inv_s);
+    double yv = y[j];
+    double phi[NF];
+    phi[0] = 1.0;
+#pragma unroll
+    for (int p = 1; p < NF; ++p) phi[p] = phi[p - 1] * t;
+    int s = 0;
+#pragma unroll
+    for (int a = 0; a < NF; ++a) {
+#pragma unroll
+      for (int b = a; b < NF; ++b) acc[s++] += phi[a] * phi[b];
+    }
+#pragma unroll
+    for (int a = 0; a < NF; ++a) acc[TRI + a] += phi[a] * yv;
+  }
+  __shared__ double lds[RED_WAVES];
+#pragma unroll
+  for (int i = 0; i < TRI + NF; ++i) {
+    double v = block_sum_f64<RED_WAVES>(acc[i], lds);
+    if (threadIdx.x == 0) atomicAdd(&out[1 + i], v);
+  }
+}
+
+at::Tensor poly_stats_hip(const at::Tensor& x, const at::Tensor& y,
+                          int64_t nf, double mu, double s) {
+  TORCH_CHECK(x.is_cuda() && y.is_cuda() && x.numel() == y.numel());
+  TORCH_CHECK(x.scalar_type() == at::kFloat && y.scalar_type() == at::kFloat);
+  TORCH_CHECK(nf >= 2 && nf <= 6, "poly_stats: 2 <= degree+1 <= 6");
+  long long n = x.numel();
+  int tri = (int)(nf * (nf + 1) / 2);
+  auto out = at::zeros({1 + tri + nf}, x.options().dtype(at::kDouble));
+  auto stream = at::cuda::getCurrentCUDAStream();
+  dim3 grid(reduce_grid(n));
+  float inv_s = (float)(1.0 / s);
+#define PS_LAUNCH(NF_)                                                     \
+  hipLaunchKernelGGL((poly_stats_kernel<NF_>), grid, dim3(RED_BLOCK), 0,   \
+                     stream, x.data_ptr<float>(), y.data_ptr<float>(),     \
+                     out.data_ptr<double>(), n, (float)mu, inv_s)
+  switch (nf) {
+    case 2: PS_LAUNCH(2); break;
+    case 3: PS_LAUNCH(3); break;
+    case 4: PS_LAUNCH(4); break;
+    case 5: PS_LAUNCH(5); break;
+    default: PS_LAUNCH(6); break;
+  }
+#undef PS_LAUNCH
+  out[0].fill_((double)n);
+  return out;
+}
+
+// Horner-scheme polynomial scoring in the normalised basis; coefficients
+// read from device memory so captured serving graphs follow redeploys.
+__global__ void poly_score_kernel(const float* __restrict__ x,
+                                  float* __restrict__ outv,
+                                  const float* __restrict__ coef, int nf,
+                                  float mu, float inv_s, long long n) {
+  const long long stride = (long long)gridDim.x * blockDim.x;
+  for (long long j = (long long)blockIdx.x * blockDim.x + threadIdx.x; j < n;
+       j += stride) {
+    float t = (x[j] - mu) * inv_s;
+    float acc = coef[nf - 1];
+    for (int p = nf - 2; p >= 0; --p) acc = fmaf(acc, t, coef[p]);
+    outv[j] = acc;
+  }
+}
+
+at::Tensor poly_score_hip(const at::Tensor& x, const at::Tensor& coef,
+                          double mu, double s) {
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kFloat);
+  TORCH_CHECK(coef.is_cuda() && coef.scalar_type() == at::kFloat &&
+              coef.numel() >= 2);
+  long long n = x.numel();
+  auto out = at::empty_like(x);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL(poly_score_kernel, dim3(reduce_grid(n)),
+                     dim3(RED_BLOCK), 0, stream, x.data_ptr<float>(),
+                     out.data_ptr<float>(), coef.data_ptr<float>(),
+                     (int)coef.numel(), (float)mu, (float)(1.0 / s), n);
+  return out;
+}
+
 // ---- regression_metrics (offline: stage_1:79-90 semantics) ---------------
 // out = [n, sum_ape, ss_res, sum_y, sum_yy, max_resid]
 

@@ -128,6 +128,29 @@ def linear_score_cpu(X: torch.Tensor, intercept: float, coef: float) -> torch.Te
     return (intercept + coef * X.float()).float()
 
 
+def poly_stats_cpu(X, y, nf: int, mu: float, s: float) -> torch.Tensor:
+    t = ((X.double() - mu) / s).numpy()
+    phi = np.vander(t, nf, increasing=True)  # [n, nf]
+    A = phi.T @ phi
+    b = phi.T @ y.double().numpy()
+    out = [float(X.numel())]
+    for a in range(nf):
+        for bb in range(a, nf):
+            out.append(A[a, bb])
+    out.extend(b.tolist())
+    return torch.tensor(out, dtype=torch.float64)
+
+
+def poly_score_cpu(X, coef, mu: float, s: float) -> torch.Tensor:
+    if torch.is_tensor(coef):
+        coef = coef.cpu().tolist()
+    t = (X.float() - mu) / s
+    acc = torch.full_like(t, float(coef[-1]))
+    for c in reversed(coef[:-1]):
+        acc = acc * t + float(c)
+    return acc
+
+
 _MAPE_EPS = float(np.finfo(np.float64).eps)  # sklearn's epsilon
 
 

@@ -90,7 +90,8 @@ def main(argv=None) -> None:
     p.add_argument("--store", default=None)
     p.add_argument("--days", type=int, default=30)
     p.add_argument("--rows", type=int, default=24 * 60)
-    p.add_argument("--model", default="linear", choices=["linear", "mlp"])
+    p.add_argument("--model", default="linear",
+                   help="linear | poly[<degree>] | mlp")
     p.add_argument("--device", default=None)
     p.add_argument("--start-date", default="2026-01-01")
     p.add_argument("--format", default="csv", choices=["csv", "npy"])

@@ -18,7 +18,11 @@ from datetime import date as date_t
 import torch
 
 from bodywork_mlops_demo_amd import ops
-from bodywork_mlops_demo_amd.models import GPULinearRegressor, GPUMLPRegressor
+from bodywork_mlops_demo_amd.models import (
+    GPULinearRegressor,
+    GPUMLPRegressor,
+    GPUPolyRegressor,
+)
 from bodywork_mlops_demo_amd.monitoring import stage_guard
 from bodywork_mlops_demo_amd.store import ArtefactStore, contract, open_store
 from bodywork_mlops_demo_amd.utils.logging import configure_logger
@@ -74,6 +78,12 @@ def run(
         model = GPULinearRegressor(device=device).fit(
             X_train, y_train, process_group=process_group
         )
+    elif model_type.startswith("poly"):
+        # "poly" (degree 3) or "poly<d>", e.g. "poly2"
+        degree = int(model_type[4:]) if len(model_type) > 4 else 3
+        model = GPUPolyRegressor(degree=degree, device=device).fit(
+            X_train, y_train, process_group=process_group
+        )
     elif model_type == "mlp":
         # reuse the cached model object so its captured training graph
         # survives across daily retrains (weights re-randomised in place)
@@ -117,7 +127,8 @@ def run(
 def main(argv=None) -> None:
     p = argparse.ArgumentParser(description=__doc__)
     p.add_argument("--store", default=None, help="store URI (dir or s3://bucket)")
-    p.add_argument("--model", default="linear", choices=["linear", "mlp"])
+    p.add_argument("--model", default="linear",
+                   help="linear | poly[<degree>] | mlp")
     p.add_argument("--device", default=None)
     p.add_argument("--mlp-steps", type=int, default=50)
     args = p.parse_args(argv)

@@ -104,6 +104,17 @@ class TestLinregGPU:
         # same fmaf on both paths? oracle is torch mul+add; allow 1 ulp
         torch.testing.assert_close(got, want, rtol=1e-6, atol=1e-5)
 
+    def test_poly_stats_and_score_match_cpu(self):
+        X = torch.rand(500_000, device=DEV) * 100
+        y = 3 + 0.8 * X - 0.004 * X * X + torch.randn_like(X)
+        got = ops.poly_stats(X, y, degree=3).cpu()
+        want = reference.poly_stats_cpu(X.cpu(), y.cpu(), 4, 50.0, 50.0)
+        torch.testing.assert_close(got, want, rtol=1e-10, atol=1e-4)
+        coef = ops.solve_poly(got, 3)
+        sg = ops.poly_score(X, coef)
+        sc = reference.poly_score_cpu(X.cpu(), coef, 50.0, 50.0)
+        torch.testing.assert_close(sg.cpu(), sc, rtol=1e-5, atol=1e-4)
+
     def test_regression_metrics(self):
         y = torch.rand(500_000, device=DEV) * 100 + 1
         yhat = y + torch.randn_like(y) * 5
