@@ -129,7 +129,10 @@ def linear_score_cpu(X: torch.Tensor, intercept: float, coef: float) -> torch.Te
 
 
 def poly_stats_cpu(X, y, nf: int, mu: float, s: float) -> torch.Tensor:
-    t = ((X.double() - mu) / s).numpy()
+    # t is computed in fp32 exactly like the device kernel ((x-mu)*inv_s),
+    # then expanded/accumulated in fp64
+    t = ((X.float() - np.float32(mu)) * np.float32(1.0 / s)) \
+        .numpy().astype(np.float64)
     phi = np.vander(t, nf, increasing=True)  # [n, nf]
     A = phi.T @ phi
     b = phi.T @ y.double().numpy()

@@ -22,7 +22,9 @@ def _curved_data(n=30000, seed=0):
 def test_poly_stats_match_numpy_design_matrix():
     X, y = _curved_data(5000)
     stats = ops.poly_stats(X, y, degree=3)
-    t = ((X.double() - 50.0) / 50.0).numpy()
+    # fp32 t (kernel semantics), fp64 accumulation
+    t = ((X.float() - np.float32(50.0)) * np.float32(1.0 / 50.0)) \
+        .numpy().astype(np.float64)
     phi = np.vander(t, 4, increasing=True)
     A = phi.T @ phi
     k = 1
