@@ -373,13 +373,32 @@ def main(argv=None) -> None:
     p.add_argument("--store", default=None)
     p.add_argument("--secrets", default=None)
     p.add_argument("--keep-services", action="store_true")
+    p.add_argument("--repeat", type=int, default=1,
+                   help="run the DAG N times (the reference's daily k8s "
+                        "cronjob role, README.md:5); advances the virtual "
+                        "pipeline date between runs")
+    p.add_argument("--interval", type=float, default=0.0,
+                   help="seconds to sleep between repeated runs")
     args = p.parse_args(argv)
     runner = PipelineRunner(args.config, store_uri=args.store,
                             secrets_file=args.secrets)
-    report = runner.run(teardown_services=not args.keep_services)
-    log.info(f"pipeline finished: ok={report.ok} succeeded={report.succeeded} "
-             f"failed={report.failed} durations={report.durations}")
-    sys.exit(0 if report.ok else 1)
+    ok = True
+    for cycle in range(args.repeat):
+        report = runner.run(teardown_services=not args.keep_services)
+        log.info(
+            f"pipeline run {cycle + 1}/{args.repeat}: ok={report.ok} "
+            f"succeeded={report.succeeded} failed={report.failed} "
+            f"durations={report.durations}"
+        )
+        ok = ok and report.ok
+        if cycle + 1 < args.repeat:
+            from bodywork_mlops_demo_amd.utils.clock import CLOCK
+
+            CLOCK.advance(1)
+            os.environ["BODYWORK_AMD_DATE"] = str(CLOCK.today())
+            if args.interval > 0:
+                time.sleep(args.interval)
+    sys.exit(0 if ok else 1)
 
 
 if __name__ == "__main__":

@@ -134,6 +134,46 @@ stages:
     assert len(store.list_keys(contract.MODELS_PREFIX)) == 1
 
 
+@pytest.mark.timeout(420)
+def test_runner_repeat_cron_mode(tmp_path, monkeypatch):
+    """--repeat N = the reference's daily-cronjob role: the DAG runs N
+    times, the virtual date advancing between runs, so dated artefacts
+    accumulate like consecutive days."""
+    import bodywork_mlops_demo_amd.pipeline.runner as runner_mod
+
+    cfg_path = tmp_path / "p.yaml"
+    cfg_path.write_text("""
+version: "1.0"
+project:
+  name: cron-test
+  DAG: stage-3-generate-next-dataset >> stage-1-train-model
+stages:
+  stage-3-generate-next-dataset:
+    executable_module_path: bodywork_mlops_demo_amd/stages/datagen.py
+    args: ["--n", "200"]
+    batch: {max_completion_time_seconds: 120, retries: 1}
+  stage-1-train-model:
+    executable_module_path: bodywork_mlops_demo_amd/stages/train.py
+    batch: {max_completion_time_seconds: 120, retries: 1}
+""")
+    store_dir = str(tmp_path / "store")
+    monkeypatch.setenv("BODYWORK_AMD_DATE", "2026-07-01")
+    from bodywork_mlops_demo_amd.utils.clock import CLOCK
+
+    CLOCK.set("2026-07-01")
+    with pytest.raises(SystemExit) as e:
+        runner_mod.main([str(cfg_path), "--store", store_dir,
+                         "--repeat", "2"])
+    assert e.value.code == 0
+    from bodywork_mlops_demo_amd.store import LocalStore
+
+    store = LocalStore(store_dir)
+    keys = store.list_keys(contract.DATASETS_PREFIX)
+    assert contract.dataset_key(date(2026, 7, 1)) in keys
+    assert contract.dataset_key(date(2026, 7, 2)) in keys
+    assert len(store.list_keys(contract.MODELS_PREFIX)) == 2
+
+
 def test_runner_retries_failing_stage(tmp_path):
     cfg = load_config("""
 version: "1.0"
