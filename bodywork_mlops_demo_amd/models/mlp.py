@@ -107,8 +107,20 @@ class GPUMLPRegressor:
         yhat = ops.rowdot_bf16(h2, self.w3_bf, self.b3)
         return yhat, h1, h2, xn, m1, m2
 
+    #: rows per forward chunk: bounds transient activations to
+    #: 2 * chunk * H bf16 (= 16 GiB at H=4096) however large the batch
+    PREDICT_CHUNK = 1 << 20
+
     def predict(self, X: torch.Tensor) -> torch.Tensor:
-        return self._forward(X.to(self.device))[0]
+        X = X.to(self.device)
+        n = X.shape[0]
+        if n <= self.PREDICT_CHUNK:
+            return self._forward(X)[0]
+        out = torch.empty(n, dtype=torch.float32, device=self.device)
+        for lo in range(0, n, self.PREDICT_CHUNK):
+            hi = min(lo + self.PREDICT_CHUNK, n)
+            out[lo:hi] = self._forward(X[lo:hi])[0]
+        return out
 
     # -- training ----------------------------------------------------------
     def fit(
