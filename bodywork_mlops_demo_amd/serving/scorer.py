@@ -23,7 +23,7 @@ log = configure_logger(__name__)
 
 
 class BatchedScorer:
-    BUCKETS = [1, 16, 256, 4096, 65536, 1 << 20]
+    BUCKETS = [1, 16, 256, 4096, 65536, 1 << 20, 1 << 22, 1 << 24]
 
     def __init__(self, model, device: str | torch.device = "cpu",
                  use_graphs: bool = True):
@@ -31,9 +31,14 @@ class BatchedScorer:
         self.device = torch.device(device)
         self.use_graphs = use_graphs and self.device.type == "cuda"
         self._graphs: dict[int, tuple] = {}
+        # models with large per-row activations advertise a chunk bound
+        # (e.g. the MLP's 2^20 = ~16 GiB of transient bf16 activations);
+        # elementwise models batch up to 2^24 rows per graph replay
+        max_bucket = int(getattr(model, "PREDICT_CHUNK", 1 << 24))
+        self.buckets = [b for b in self.BUCKETS if b <= max_bucket]
 
     def _bucket(self, n: int) -> int:
-        for b in self.BUCKETS:
+        for b in self.buckets:
             if n <= b:
                 return b
         return n
@@ -74,7 +79,7 @@ class BatchedScorer:
         n = X.shape[0]
         if not self.use_graphs:
             return self.model.predict(X)
-        max_b = self.BUCKETS[-1]
+        max_b = self.buckets[-1]
         if n > max_b:
             out = torch.empty(n, device=self.device, dtype=torch.float32)
             for lo in range(0, n, max_b):
