@@ -150,15 +150,22 @@ def run_cycle(
     mlp_batch_size: int = 65536,
     use_graphs: bool = True,
     scorer_cache: dict | None = None,
+    skip_train: bool = False,
 ) -> dict:
     """Run one full cycle; returns per-phase timings + metrics.
 
     In a DP world each rank generates/holds/trains on its own ``n_rows``
     rows (weak scaling: per-GPU work fixed); rank 0 persists artefacts.
+
+    ``skip_train`` keeps the currently deployed model (drift-policy
+    loops retrain only when the live metrics degrade); it requires a
+    scorer already resident in ``scorer_cache``.
     """
     device = state.device
     dev_cuda = device.startswith("cuda")
     timings: dict[str, float] = {}
+    if skip_train and not (scorer_cache and scorer_cache.get("scorer")):
+        skip_train = False  # nothing deployed yet -> must train
 
     def sync():
         if dev_cuda:
@@ -179,6 +186,13 @@ def run_cycle(
     sync()
     t0 = perf_counter()
     eff_store = store if store is not None else _NullStore()
+    if skip_train:
+        timings["train_s"] = 0.0
+        timings["deploy_s"] = 0.0
+        metrics = scorer_cache.get("offline_metrics", {})
+        scorer = scorer_cache["scorer"]
+        return _finish_cycle(state, store, n_rows, persist_fmt, scorer,
+                             timings, metrics, sync)
     metrics, trained = stage1.run(
         eff_store,
         model_type=model_type,
@@ -228,8 +242,19 @@ def run_cycle(
         scorer = BatchedScorer(model, device, use_graphs=use_graphs)
         if scorer_cache is not None:
             scorer_cache["scorer"] = scorer
+    if scorer_cache is not None:
+        scorer_cache["offline_metrics"] = metrics
     sync()
     timings["deploy_s"] = perf_counter() - t0
+
+    return _finish_cycle(state, store, n_rows, persist_fmt, scorer, timings,
+                         metrics, sync)
+
+
+def _finish_cycle(state, store, n_rows, persist_fmt, scorer, timings,
+                  metrics, sync):
+    """Stages 3+4 and clock advance (shared by train and skip-train paths)."""
+    device = state.device
 
     # -- stage 3: generate day t+1 ------------------------------------------
     t0 = perf_counter()

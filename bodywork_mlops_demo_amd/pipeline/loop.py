@@ -36,7 +36,18 @@ def run_loop(
     rank: int = 0,
     world_size: int = 1,
     resume: bool = True,
+    retrain_policy: str = "always",
+    drift_threshold: float = 1.5,
 ) -> list[dict]:
+    """Run ``days`` cycles.
+
+    ``retrain_policy``:
+    - ``"always"`` — reference semantics: retrain + redeploy every day.
+    - ``"drift"``  — monitoring-driven MLOps: keep the deployed model
+      while its live MAPE stays below ``drift_threshold`` x its own
+      offline MAPE at training time; retrain only when the online
+      metrics show the concept has drifted away from it.
+    """
     device = device or ("cuda" if torch.cuda.is_available() else "cpu")
     if isinstance(start_date, str):
         start_date = date_t.fromisoformat(start_date)
@@ -66,13 +77,17 @@ def run_loop(
 
     results = []
     scorer_cache: dict = {}
+    drifted = True  # first day always trains
     for day in range(days):
+        skip = retrain_policy == "drift" and not drifted
         r = run_cycle(
             state, store, n_rows, model_type=model_type,
             process_group=process_group, persist_fmt=persist_fmt,
-            scorer_cache=scorer_cache,
+            scorer_cache=scorer_cache, skip_train=skip,
         )
         results.append(r)
+        offline_mape = (r["offline"] or {}).get("MAPE", float("inf"))
+        drifted = r["online"]["MAPE"] > drift_threshold * offline_mape
         t = r["timings"]
         log.info(
             f"cycle {day + 1}/{days} ({state.date}): "
@@ -80,6 +95,9 @@ def run_loop(
             f"{t['deploy_s']:.3f} datagen {t['datagen_s']:.3f} "
             f"test {t['test_s']:.3f}) online MAPE "
             f"{r['online']['MAPE']:.4f}"
+            + (" [retrain skipped: no drift]" if skip else "")
+            + (" [drift detected]" if drifted and retrain_policy == "drift"
+               else "")
         )
     state.drain_io()
     return results
@@ -96,11 +114,16 @@ def main(argv=None) -> None:
     p.add_argument("--start-date", default="2026-01-01")
     p.add_argument("--format", default="csv", choices=["csv", "npy"])
     p.add_argument("--json-out", default=None)
+    p.add_argument("--retrain-policy", default="always",
+                   choices=["always", "drift"])
+    p.add_argument("--drift-threshold", type=float, default=1.5)
     args = p.parse_args(argv)
     results = run_loop(
         open_store(args.store), days=args.days, n_rows=args.rows,
         model_type=args.model, device=args.device,
         start_date=args.start_date, persist_fmt=args.format,
+        retrain_policy=args.retrain_policy,
+        drift_threshold=args.drift_threshold,
     )
     if args.json_out:
         with open(args.json_out, "w") as f:

@@ -190,3 +190,20 @@ stages:
     report = runner.run()
     assert not report.ok
     assert report.attempts["boom"] == 2  # initial + 1 retry
+
+
+def test_drift_policy_skips_retrains(tmp_store):
+    """retrain_policy='drift' keeps the deployed model while its live
+    MAPE stays within threshold x offline MAPE, and retrains when the
+    sinusoidal concept moves away."""
+    results = run_loop(tmp_store, days=8, n_rows=800, model_type="linear",
+                       device="cpu", start_date="2026-02-01",
+                       retrain_policy="drift", drift_threshold=2.5)
+    trained = [r["timings"]["train_s"] > 0 for r in results]
+    assert trained[0]  # first day always trains
+    assert not all(trained), "policy never skipped a retrain"
+    # model artefacts only appear on trained days
+    n_models = len(tmp_store.list_keys(contract.MODELS_PREFIX))
+    assert n_models == sum(trained)
+    # every day still produced online metrics
+    assert len(tmp_store.list_keys(contract.TEST_METRICS_PREFIX)) == 8
