@@ -43,12 +43,25 @@ class CycleState:
         self.rank = rank
         self.world_size = world_size
         self.history_days = history_days
-        self.y = torch.empty(0, device=device)
-        self.X = torch.empty(0, device=device)
+        # geometric-capacity history buffers: appending a day never
+        # reallocates unless capacity is exceeded (a per-day torch.cat
+        # fragmented the caching allocator to ~240 GiB reserved over a
+        # 150-day run)
+        self._buf_y = torch.empty(0, device=device)
+        self._buf_X = torch.empty(0, device=device)
+        self._n = 0
         self._day_sizes: list[int] = []
         self._io_pool = None
         self._io_futures: list = []
         self.cycle_count = 0
+
+    @property
+    def y(self) -> torch.Tensor:
+        return self._buf_y[: self._n]
+
+    @property
+    def X(self) -> torch.Tensor:
+        return self._buf_X[: self._n]
 
     @property
     def persist_rank(self) -> int:
@@ -129,14 +142,31 @@ class CycleState:
         self._pin_futures[flip] = self._io_futures[fut_idx]
 
     def append_day(self, y: torch.Tensor, X: torch.Tensor) -> None:
-        self.y = torch.cat([self.y, y])
-        self.X = torch.cat([self.X, X])
-        self._day_sizes.append(int(y.shape[0]))
+        add = int(y.shape[0])
+        need = self._n + add
+        if need > self._buf_y.numel():
+            cap = max(need, 2 * self._buf_y.numel())
+            new_y = torch.empty(cap, device=self.device, dtype=torch.float32)
+            new_X = torch.empty(cap, device=self.device, dtype=torch.float32)
+            if self._n:
+                new_y[: self._n].copy_(self._buf_y[: self._n])
+                new_X[: self._n].copy_(self._buf_X[: self._n])
+            self._buf_y, self._buf_X = new_y, new_X
+        self._buf_y[self._n:need].copy_(y)
+        self._buf_X[self._n:need].copy_(X)
+        self._n = need
+        self._day_sizes.append(add)
         if self.history_days is not None:
+            drop = 0
             while len(self._day_sizes) > self.history_days:
-                drop = self._day_sizes.pop(0)
-                self.y = self.y[drop:].contiguous()
-                self.X = self.X[drop:].contiguous()
+                drop += self._day_sizes.pop(0)
+            if drop:
+                keep = self._n - drop
+                # overlapping region: stage through a clone (fixed-size
+                # transient the allocator reuses)
+                self._buf_y[:keep].copy_(self._buf_y[drop:self._n].clone())
+                self._buf_X[:keep].copy_(self._buf_X[drop:self._n].clone())
+                self._n = keep
 
 
 def run_cycle(
