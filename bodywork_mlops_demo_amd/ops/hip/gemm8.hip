@@ -1,37 +1,45 @@
 // 256x256-tile 8-phase software-pipelined NT MFMA GEMM (gfx950).
 //
-// The deep-pipeline upgrade over gemm.hip's 128^2 kernel (which sits at
-// the documented plain-HIP 2-phase ceiling, ~1000 TF): 8 waves per
-// 512-thread block (2x4), per-wave C = 128x64 = 8x4 fragments; one
-// K-tile (BK=64) is computed as 4 "quadrant" phases of 16 MFMAs each,
+// The deep-pipeline variant of gemm.hip's 128^2 kernel (which sits at
+// the documented plain-HIP 2-phase ceiling, ~1000 TF): 16 waves per
+// 1024-thread block (4x4), per-wave C = 64x64 = 4x4 fragments; one
+// K-tile (BK=64) is computed as 4 "quadrant" phases of 8 MFMAs each,
 // with the operand traffic software-pipelined at HALF-TILE (16 KiB)
 // granularity:
 //
-//   phase (t,q):  [q==3: s_waitcnt vmcnt(6)]   <- once per K-tile
-//                 s_barrier                      <- ONE barrier per phase
-//                 issue 1 half-tile glds (schedule below)
-//                 ds_read the next quadrant's A fragments (4x b128)
-//                   [q==3: + ALL of tile t+1's B fragments (8) and its
-//                    quadrant-0 A fragments -> the 12-read phase]
-//                 s_setprio(1); 16 x mfma_f32_16x16x32_bf16; s_setprio(0)
+//   phase (u,q):  [q==0: s_waitcnt vmcnt(4), or vmcnt(0) on the last
+//                  tile]                        <- once per K-tile
+//                 s_barrier                     <- ONE barrier per phase
+//                 issue glds (schedule below; one wave-level glds
+//                   instruction per half-tile at 1024 threads)
+//                 ds_read this quadrant's A fragments (2x b128) feeding
+//                   the SAME phase's MFMAs [q==0: + ALL of tile u's B
+//                   fragments (8); q==2: + quadrant 3 pre-read]
+//                 s_setprio(1); 8 x mfma_f32_16x16x32_bf16; s_setprio(0)
 //
-// Issue schedule (1 half-tile per phase, 2 glds each at 512 threads):
-//   (t,0): A1(t+1)   (t,1): B0(t+2)   (t,2): B1(t+2)   (t,3): A0(t+2)
-// Every half-tile therefore has >=3 phases of flight before its first
-// ds_read, and the single vmcnt(6) per tile (newest 3 half-tiles = 6
-// wave-glds stay IN FLIGHT) guarantees everything older has landed —
-// the pipeline never drains.  Slot-reuse safety: a slot's last reads are
-// >=1 phase before the glds that overwrites it, separated by a barrier.
+// Issue schedule:
+//   (u,1): B0(u+2)   (u,2): B1(u+2)   (u,3): A0(u+2)+A1(u+2)
+// Landing proof: at (u,0) the four newest outstanding glds are tile
+// u+1's {B0,B1,A0,A1}; the 5th-newest is tile u's A1 -> vmcnt(4) proves
+// tile u fully landed while keeping the pipeline 4 deep (never drains;
+// the final tile, with no newer issues, drains with vmcnt(0) once).
+// Slot-reuse safety: every glds overwrites a slot whose last ds_read
+// happened in an EARLIER phase, separated by a barrier — B(u) slots are
+// read only at (u,0) and re-staged at (u,1)/(u,2); A(u) slots are last
+// read at (u,2) (quadrant 3 pre-reads there) and re-staged at (u,3).
 //
 // LDS: per operand 2 buffers x 2 halves x [128][64] shorts = 64 KiB;
-// A+B = 128 KiB -> 1 block/CU, 8 waves (2/SIMD).  Same 16-byte-chunk
-// XOR swizzle as gemm.hip (chunk (row,c) holds logical (row, c^(row&7))).
-// The K-loop is unrolled two tiles per iteration so every register-set
-// index is compile-time (guide rule 20) — hence "8 phases per iteration".
+// A+B = 128 KiB -> 1 block/CU, 16 waves (4/SIMD, so the per-lane VGPR
+// cap is 128).  Measured allocation: 124-126 VGPR, ZERO spills (acc 64 +
+// B 32 + A 2x8 + addressing) — unlike the earlier 8-wave/128-row-per-wave
+// variant which hit the 256 cap with spills.
+// Same 16-byte-chunk XOR swizzle as gemm.hip, carried on the glds SOURCE
+// address and the fragment read.  The K-loop is unrolled two tiles per
+// iteration so every register-set index is compile-time (rule 20).
 //
-// Requirements: K % 128 == 0 (even tile count); M/N edges handled by
-// source clamping + epilogue guards.  Dispatched for M >= 256 (training /
-// batch-scoring shapes); smaller M stays on the 128^2 kernel.
+// Requirements: M % 256 == 0, N % 256 == 0, K % 128 == 0 (dispatched for
+// those shapes only; others use the 128^2 kernel).  Env-gated:
+// BODYWORK_GEMM_8PHASE=1.
 #include <ATen/cuda/CUDAContext.h>
 #include <hip/hip_runtime.h>
 #include <torch/extension.h>
@@ -41,7 +49,7 @@
 #define G8_BM 256
 #define G8_BN 256
 #define G8_BK 64
-#define G8_THREADS 512
+#define G8_THREADS 1024
 #define G8_HT (128 * 64)  // shorts per half-tile slot
 
 typedef __attribute__((ext_vector_type(4))) float g8_f32x4;
@@ -56,36 +64,35 @@ __device__ __forceinline__ void g8_glds16(const void* gsrc, void* lds_dst) {
       (__attribute__((address_space(3))) unsigned int*)lds_dst, 16, 0, 0);
 }
 
-// stage one [128][64]-bf16 half-tile: 1024 16-B chunks, 512 threads -> 2
-// wave-level glds per wave.  Lane-linear LDS; XOR swizzle on the source.
-// The per-lane source BYTE offsets (row*K + swizzled column chunk) are
-// tile-invariant — computed once at kernel entry (off0/off1, 32-bit) so
+// stage one [128][64]-bf16 half-tile: 1024 16-B chunks, 1024 threads ->
+// ONE wave-level glds per wave.  Lane-linear LDS; XOR swizzle on the
+// source.  The per-lane source BYTE offset (row*K + swizzled column
+// chunk) is tile-invariant — computed once at kernel entry (32-bit), so
 // the hot loop's staging is {uniform base + int offset} with no 64-bit
 // per-lane math (this kernel is dispatched only for M,N % 256 == 0, so
 // no row clamping is needed).
 __device__ __forceinline__ void g8_stage_half(short* __restrict__ slot,
                                               const char* __restrict__ base,
-                                              int off0, int off1) {
+                                              int off0) {
   const int wave = threadIdx.x >> 6;
-  g8_glds16(base + off0, (char*)slot + (0 * 8 + wave) * 1024);
-  g8_glds16(base + off1, (char*)slot + (1 * 8 + wave) * 1024);
+  g8_glds16(base + off0, (char*)slot + wave * 1024);
 }
 
 template <int EPI, bool HAS_BIAS, bool OUT_FP32, bool EMIT_MASK>
 __device__ __forceinline__ void g8_epilogue(
-    g8_f32x4 (&acc)[8][4], const float* __restrict__ bias,
+    g8_f32x4 (&acc)[4][4], const float* __restrict__ bias,
     const unsigned char* __restrict__ mask,
     unsigned char* __restrict__ mask_out, void* __restrict__ C, long long M,
     long long N, long long m0, long long n0, int wm, int wn, int fl, int kg) {
 #pragma unroll
-  for (int i = 0; i < 8; ++i) {
+  for (int i = 0; i < 4; ++i) {
     unsigned long long mrow[4];
     if (EPI == G8_EPI_MASK) {
       const long long stripe = n0 + wn * 64;
       const bool full = stripe + 64 <= N;
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        long long row = m0 + wm * 128 + i * 16 + kg * 4 + r;
+        long long row = m0 + wm * 64 + i * 16 + kg * 4 + r;
         if (row >= M) {
           mrow[r] = 0ull;
         } else if (full) {
@@ -110,7 +117,7 @@ __device__ __forceinline__ void g8_epilogue(
           (EPI == G8_EPI_BIAS_RELU && HAS_BIAS && col_ok) ? bias[col] : 0.0f;
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        long long row = m0 + wm * 128 + i * 16 + kg * 4 + r;
+        long long row = m0 + wm * 64 + i * 16 + kg * 4 + r;
         bool row_ok = row < M;
         float v = acc[i][j][r];
         if (EPI == G8_EPI_BIAS_RELU) {
@@ -151,8 +158,8 @@ __global__ void gemm_nt_8phase_kernel(
   const long long n0 = (long long)blockIdx.x * G8_BN;
   const int wave = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
-  const int wm = wave >> 2;  // 0..1
-  const int wn = wave & 3;   // 0..3
+  const int wm = wave >> 2;  // 0..3 (64 C-rows per wave)
+  const int wn = wave & 3;   // 0..3 (64 C-cols per wave)
   const int fl = lane & 15;
   const int kg = lane >> 4;
   const int swz = fl & 7;
@@ -162,19 +169,20 @@ __global__ void gemm_nt_8phase_kernel(
 #define G8_ASLOT(buf, half) (lds + ((buf) * 2 + (half)) * G8_HT)
 #define G8_BSLOT(buf, half) (lds + 4 * G8_HT + ((buf) * 2 + (half)) * G8_HT)
 
-  // wave-local read bases: this wave only ever touches A-half wm and
-  // B-half (wn>>1); within-half row offsets are compile-time per fragment
-  const int b_inhalf = (wn & 1) * 64;  // col base within the B half
+  // wave-local read bases: this wave only touches A-half (wm>>1) and
+  // B-half (wn>>1); within-half row offsets are compile-time
+  const int a_inhalf = (wm & 1) * 64;
+  const int b_inhalf = (wn & 1) * 64;
 
-  g8_f32x4 acc[8][4];
+  g8_f32x4 acc[4][4];
 #pragma unroll
-  for (int i = 0; i < 8; ++i)
+  for (int i = 0; i < 4; ++i)
 #pragma unroll
     for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
-  // fragment read helpers (compile-time q/ks/set indices at call sites)
 #define G8_AREAD(dst, buf, mfrag, ks)                                       \
-  dst = ((const lds_vec*)(G8_ASLOT(buf, wm) + ((mfrag) * 16 + fl) * 64 +    \
+  dst = ((const lds_vec*)(G8_ASLOT(buf, (wm >> 1)) +                        \
+                          (a_inhalf + (mfrag) * 16 + fl) * 64 +             \
                           (((ks) * 4 + kg) ^ swz) * 8))                     \
             ->v
 #define G8_BREAD(dst, buf, nfrag, ks)                                       \
@@ -183,116 +191,111 @@ __global__ void gemm_nt_8phase_kernel(
                           (((ks) * 4 + kg) ^ swz) * 8))                     \
             ->v
 
-  bf16x8_v a_set0[2][2], a_set1[2][2];  // [mi][ks] — two quad sets
-  bf16x8_v bA[4][2], bB[4][2];          // [nfrag][ks] — even/odd tile B
+  bf16x8_v a_q[2];     // current quadrant's A fragments [ks]
+  bf16x8_v a_q3[2];    // quadrant 3, pre-read one phase early (slot-free proof)
+  bf16x8_v b_t[4][2];  // current tile's B fragments [nfrag][ks]
 
-  // per-lane staging byte offsets (tile-invariant; see g8_stage_half)
-  int stg_off[2];
+  // per-lane staging byte offset (tile-invariant; see g8_stage_half)
+  int stg_off;
   {
-    const int lane_ = threadIdx.x & 63;
-    const int wave_ = threadIdx.x >> 6;
-#pragma unroll
-    for (int s = 0; s < 2; ++s) {
-      int ci = (s * 8 + wave_) * 64 + lane_;
-      int row = ci >> 3;
-      int sc = (ci & 7) ^ (row & 7);
-      stg_off[s] = (int)((row * K + sc * 8) * 2);
-    }
+    int ci = (int)threadIdx.x;
+    int row = ci >> 3;
+    int sc = (ci & 7) ^ (row & 7);
+    stg_off = (int)((row * K + sc * 8) * 2);
   }
-  // uniform base pointers: A halves at rows m0 / m0+128, B at n0 / n0+128
   const char* Ah0 = (const char*)(A + m0 * K);
   const char* Ah1 = (const char*)(A + (m0 + 128) * K);
   const char* Bh0 = (const char*)(B + n0 * K);
   const char* Bh1 = (const char*)(B + (n0 + 128) * K);
 #define G8_KOFF(T) ((long long)(T) * (G8_BK * 2))
 
-  // ---- prologue: 7 half-tiles, then land tile 0 -------------------------
-  g8_stage_half(G8_ASLOT(0, 0), Ah0, stg_off[0], stg_off[1]);
-  g8_stage_half(G8_ASLOT(0, 1), Ah1, stg_off[0], stg_off[1]);
-  g8_stage_half(G8_BSLOT(0, 0), Bh0, stg_off[0], stg_off[1]);
-  g8_stage_half(G8_BSLOT(0, 1), Bh1, stg_off[0], stg_off[1]);
+  // ---- prologue: tiles 0 and 1 fully issued (8 half-tiles) --------------
+  // Issue order matches the steady-state wait: at (0,0) `vmcnt(4)` keeps
+  // the NEWEST four (tile 1's B then A) in flight and proves tile 0 landed.
+  g8_stage_half(G8_ASLOT(0, 0), Ah0, stg_off);
+  g8_stage_half(G8_ASLOT(0, 1), Ah1, stg_off);
+  g8_stage_half(G8_BSLOT(0, 0), Bh0, stg_off);
+  g8_stage_half(G8_BSLOT(0, 1), Bh1, stg_off);
   if (nt > 1) {
-    g8_stage_half(G8_BSLOT(1, 0), Bh0 + G8_KOFF(1), stg_off[0], stg_off[1]);
-    g8_stage_half(G8_BSLOT(1, 1), Bh1 + G8_KOFF(1), stg_off[0], stg_off[1]);
-    g8_stage_half(G8_ASLOT(1, 0), Ah0 + G8_KOFF(1), stg_off[0], stg_off[1]);
+    g8_stage_half(G8_BSLOT(1, 0), Bh0 + G8_KOFF(1), stg_off);
+    g8_stage_half(G8_BSLOT(1, 1), Bh1 + G8_KOFF(1), stg_off);
+    g8_stage_half(G8_ASLOT(1, 0), Ah0 + G8_KOFF(1), stg_off);
+    g8_stage_half(G8_ASLOT(1, 1), Ah1 + G8_KOFF(1), stg_off);
   }
-  asm volatile("s_waitcnt vmcnt(6)" ::: "memory");  // tile 0 landed
-  __builtin_amdgcn_s_barrier();
-  // boundary-style reads for tile 0: all B(0) + A(0) quad 0 -> set 0
-#pragma unroll
-  for (int nf = 0; nf < 4; ++nf) {
-    G8_BREAD(bA[nf][0], 0, nf, 0);
-    G8_BREAD(bA[nf][1], 0, nf, 1);
-  }
-  G8_AREAD(a_set0[0][0], 0, 0, 0);
-  G8_AREAD(a_set0[0][1], 0, 0, 1);
-  G8_AREAD(a_set0[1][0], 0, 1, 0);
-  G8_AREAD(a_set0[1][1], 0, 1, 1);
 
-  // one phase: issue -> reads -> MFMA.  T: tile index (runtime), TPAR:
-  // compile-time tile parity (buffers/B-set), Q: compile-time quadrant.
-  // a_rd / a_mm: the A-register sets to read-into / mfma-from.
-#define G8_PHASE(T, TPAR, Q, BCUR, BNXT, A_RD, A_MM)                        \
+  // One phase = barrier + (issue) + LDS reads + 8 MFMAs.  Reads feed the
+  // SAME phase's MFMAs (lgkm waits are in-phase), except quadrant 3 which
+  // is pre-read at phase 2 so that phase 3's A-slot glds issues target a
+  // slot with no reads outstanding.  Issue/wait schedule per tile u:
+  //   (u,0): s_waitcnt vmcnt(4); read B(u) x8 + A q0; MFMA q0
+  //   (u,1): issue B0(u+2);      read A q1;           MFMA q1
+  //   (u,2): issue B1(u+2);      read A q2 AND q3;    MFMA q2
+  //   (u,3): issue A0,A1(u+2);                        MFMA q3
+  // Landing proof: at (u,0) the newest 4 outstanding glds are tile u+1's
+  // {B0,B1,A0,A1}; the 5th-newest is A1(u) -> vmcnt(4) proves all of tile
+  // u landed.  Slot-free proof: every glds targets a slot whose last
+  // ds_read happened in an EARLIER phase (barrier-separated): B(u) slots
+  // read only at (u,0), re-staged at (u,1)/(u,2); A(u) slots last read at
+  // (u,2) (q3 pre-read), re-staged at (u,3).
+#define G8_PHASE(T, TPAR, Q)                                                \
   do {                                                                      \
-    /* vmcnt(4): newest 2 half-tiles (4 wave-glds) stay in flight; the    \
-       3rd-newest is A1(T+1) issued at (T,0), which THIS phase's boundary \
-       reads consume — it must be landed (vmcnt(6) would not pin it) */   \
-    if (Q == 3) asm volatile("s_waitcnt vmcnt(4)" ::: "memory");            \
-    __builtin_amdgcn_s_barrier();                                           \
-    /* issue schedule */                                                    \
     if (Q == 0) {                                                           \
+      /* middle tiles: the 4 newest outstanding glds are tile T+1's     */ \
+      /* halves, so vmcnt(4) proves tile T landed.  On the LAST tile    */ \
+      /* nothing newer was issued (guards off) -> full drain once.      */ \
       if ((T) + 1 < nt)                                                     \
-        g8_stage_half(G8_ASLOT(((TPAR) ^ 1), 1), Ah1 + G8_KOFF((T) + 1),    \
-                      stg_off[0], stg_off[1]);                              \
-    } else if (Q == 1) {                                                    \
+        asm volatile("s_waitcnt vmcnt(4)" ::: "memory");                    \
+      else                                                                  \
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");                    \
+    }                                                                       \
+    __builtin_amdgcn_s_barrier();                                           \
+    if (Q == 1) {                                                           \
       if ((T) + 2 < nt)                                                     \
-        g8_stage_half(G8_BSLOT(TPAR, 0), Bh0 + G8_KOFF((T) + 2),            \
-                      stg_off[0], stg_off[1]);                              \
+        g8_stage_half(G8_BSLOT(TPAR, 0), Bh0 + G8_KOFF((T) + 2), stg_off);  \
     } else if (Q == 2) {                                                    \
       if ((T) + 2 < nt)                                                     \
-        g8_stage_half(G8_BSLOT(TPAR, 1), Bh1 + G8_KOFF((T) + 2),            \
-                      stg_off[0], stg_off[1]);                              \
-    } else {                                                                \
-      if ((T) + 2 < nt)                                                     \
-        g8_stage_half(G8_ASLOT(TPAR, 0), Ah0 + G8_KOFF((T) + 2),            \
-                      stg_off[0], stg_off[1]);                              \
-    }                                                                       \
-    /* reads for the NEXT phase */                                          \
-    if (Q < 3) {                                                            \
-      G8_AREAD(A_RD[0][0], TPAR, 2 * (Q + 1), 0);                           \
-      G8_AREAD(A_RD[0][1], TPAR, 2 * (Q + 1), 1);                           \
-      G8_AREAD(A_RD[1][0], TPAR, 2 * (Q + 1) + 1, 0);                       \
-      G8_AREAD(A_RD[1][1], TPAR, 2 * (Q + 1) + 1, 1);                       \
-    } else if ((T) + 1 < nt) { /* boundary: B(T+1) + A(T+1) quad 0 */       \
-      _Pragma("unroll") for (int nf = 0; nf < 4; ++nf) {                    \
-        G8_BREAD(BNXT[nf][0], ((TPAR) ^ 1), nf, 0);                         \
-        G8_BREAD(BNXT[nf][1], ((TPAR) ^ 1), nf, 1);                         \
+        g8_stage_half(G8_BSLOT(TPAR, 1), Bh1 + G8_KOFF((T) + 2), stg_off);  \
+    } else if (Q == 3) {                                                    \
+      if ((T) + 2 < nt) {                                                   \
+        g8_stage_half(G8_ASLOT(TPAR, 0), Ah0 + G8_KOFF((T) + 2), stg_off);  \
+        g8_stage_half(G8_ASLOT(TPAR, 1), Ah1 + G8_KOFF((T) + 2), stg_off);  \
       }                                                                     \
-      G8_AREAD(A_RD[0][0], ((TPAR) ^ 1), 0, 0);                             \
-      G8_AREAD(A_RD[0][1], ((TPAR) ^ 1), 0, 1);                             \
-      G8_AREAD(A_RD[1][0], ((TPAR) ^ 1), 1, 0);                             \
-      G8_AREAD(A_RD[1][1], ((TPAR) ^ 1), 1, 1);                             \
     }                                                                       \
-    /* 16 MFMAs: quadrant Q (m-frags 2Q, 2Q+1) x 4 n-frags x 2 ks */        \
+    if (Q == 0) {                                                           \
+      _Pragma("unroll") for (int nf = 0; nf < 4; ++nf) {                    \
+        G8_BREAD(b_t[nf][0], TPAR, nf, 0);                                  \
+        G8_BREAD(b_t[nf][1], TPAR, nf, 1);                                  \
+      }                                                                     \
+    }                                                                       \
+    if (Q < 3) {                                                            \
+      G8_AREAD(a_q[0], TPAR, Q, 0);                                         \
+      G8_AREAD(a_q[1], TPAR, Q, 1);                                         \
+    }                                                                       \
+    if (Q == 2) {                                                           \
+      G8_AREAD(a_q3[0], TPAR, 3, 0);                                        \
+      G8_AREAD(a_q3[1], TPAR, 3, 1);                                        \
+    }                                                                       \
     __builtin_amdgcn_s_setprio(1);                                          \
-    _Pragma("unroll") for (int mi = 0; mi < 2; ++mi)                        \
-        _Pragma("unroll") for (int nf = 0; nf < 4; ++nf)                    \
+    _Pragma("unroll") for (int nf = 0; nf < 4; ++nf)                        \
         _Pragma("unroll") for (int ks = 0; ks < 2; ++ks)                    \
-            acc[2 * (Q) + mi][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16( \
-                A_MM[mi][ks], BCUR[nf][ks], acc[2 * (Q) + mi][nf], 0, 0, 0); \
+            acc[Q][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(           \
+                (Q == 3 ? a_q3[ks] : a_q[ks]), b_t[nf][ks], acc[Q][nf], 0,  \
+                0, 0);                                                      \
     __builtin_amdgcn_s_setprio(0);                                          \
   } while (0)
 
-  // main loop: 2 K-tiles (8 phases) per iteration; nt is even (K%128==0)
+  // two tiles per iteration: buffer indices stay compile-time (rule 20)
   for (long long t = 0; t < nt; t += 2) {
-    G8_PHASE(t, 0, 0, bA, bB, a_set1, a_set0);
-    G8_PHASE(t, 0, 1, bA, bB, a_set0, a_set1);
-    G8_PHASE(t, 0, 2, bA, bB, a_set1, a_set0);
-    G8_PHASE(t, 0, 3, bA, bB, a_set0, a_set1);
-    G8_PHASE(t + 1, 1, 0, bB, bA, a_set1, a_set0);
-    G8_PHASE(t + 1, 1, 1, bB, bA, a_set0, a_set1);
-    G8_PHASE(t + 1, 1, 2, bB, bA, a_set1, a_set0);
-    G8_PHASE(t + 1, 1, 3, bB, bA, a_set0, a_set1);
+    G8_PHASE(t, 0, 0);
+    G8_PHASE(t, 0, 1);
+    G8_PHASE(t, 0, 2);
+    G8_PHASE(t, 0, 3);
+    if (t + 1 < nt) {
+      G8_PHASE(t + 1, 1, 0);
+      G8_PHASE(t + 1, 1, 1);
+      G8_PHASE(t + 1, 1, 2);
+      G8_PHASE(t + 1, 1, 3);
+    }
   }
 #undef G8_PHASE
 #undef G8_AREAD
