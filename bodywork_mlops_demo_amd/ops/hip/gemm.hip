@@ -457,12 +457,16 @@ static void launch_gemm(bool tn, int epi, bool has_bias, bool out_fp32,
   const bf16_t* bp = (const bf16_t*)b.data_ptr();
   void* cp = c.data_ptr();
 
-  // deep-pipeline 256^2 8-phase kernel (gemm8.hip) for exact-tile shapes
+  // deep-pipeline 16-wave 256^2 8-phase kernel (gemm8.hip) for
+  // exact-tile shapes — DEFAULT ON: A/B-measured on MI355X at
+  // 1249-1292 TF vs this file's 128^2 kernel's 1011-1084 on the hot
+  // shapes (profiles/r01_gemm_variant_study.md; +25% on the MLP
+  // training cycle).  BODYWORK_GEMM_8PHASE=0 opts out.
   if (!tn && K % 128 == 0 && K >= 256 && K <= (1 << 22) && M % 256 == 0 &&
       N % 256 == 0 && M >= 256) {
     static int use8 = [] {
       const char* e = getenv("BODYWORK_GEMM_8PHASE");
-      return (e && e[0] == '1') ? 1 : 0;
+      return (e && e[0] == '0') ? 0 : 1;
     }();
     if (use8) {
       extern void launch_gemm8(int, bool, bool, bool, const void*,
