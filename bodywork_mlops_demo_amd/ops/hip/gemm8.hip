@@ -38,8 +38,10 @@
 // iteration so every register-set index is compile-time (rule 20).
 //
 // Requirements: M % 256 == 0, N % 256 == 0, K % 128 == 0 (dispatched for
-// those shapes only; others use the 128^2 kernel).  Env-gated:
-// BODYWORK_GEMM_8PHASE=1.
+// those shapes only; others use the 128^2 kernel).  DEFAULT path for
+// qualifying shapes — A/B-measured 1249-1292 TF vs the 128^2 kernel's
+// 1011-1084 (same box, profiles/r01_gemm_variant_study.md);
+// BODYWORK_GEMM_8PHASE=0 opts out.
 #include <ATen/cuda/CUDAContext.h>
 #include <hip/hip_runtime.h>
 #include <torch/extension.h>
