@@ -250,7 +250,6 @@ __global__ void gemm_nt_8phase_kernel(
       else                                                                  \
         asm volatile("s_waitcnt vmcnt(0)" ::: "memory");                    \
     }                                                                       \
-    __builtin_amdgcn_s_barrier();                                           \
     if (Q == 1) {                                                           \
       if ((T) + 2 < nt)                                                     \
         g8_stage_half(G8_BSLOT(TPAR, 0), Bh0 + G8_KOFF((T) + 2), stg_off);  \
@@ -277,6 +276,11 @@ __global__ void gemm_nt_8phase_kernel(
       G8_AREAD(a_q3[0], TPAR, 3, 0);                                        \
       G8_AREAD(a_q3[1], TPAR, 3, 1);                                        \
     }                                                                       \
+    /* reads/issues stay ABOVE the barrier (pinned by the clobber) so   */ \
+    /* ds_read latency hides behind barrier arrival + prior MFMAs; the  */ \
+    /* compiler's counted lgkm waits gate the consuming MFMAs below.    */ \
+    asm volatile("" ::: "memory");                                          \
+    __builtin_amdgcn_s_barrier();                                           \
     __builtin_amdgcn_s_setprio(1);                                          \
     _Pragma("unroll") for (int nf = 0; nf < 4; ++nf)                        \
         _Pragma("unroll") for (int ks = 0; ks < 2; ++ks)                    \
