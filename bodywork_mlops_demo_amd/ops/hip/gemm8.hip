@@ -250,6 +250,25 @@ __global__ void gemm_nt_8phase_kernel(
       else                                                                  \
         asm volatile("s_waitcnt vmcnt(0)" ::: "memory");                    \
     }                                                                       \
+    /* FIRST-TOUCH reads of tile T's slots (phase 0) must stay AFTER    */ \
+    /* the barrier: vmcnt is per-wave, and each slot is staged by ALL   */ \
+    /* 16 waves' glds, so only wait+barrier proves the whole half-tile  */ \
+    /* landed.  Later phases re-read already-validated tile-T data, so  */ \
+    /* their ds_reads sit ABOVE the barrier (pinned by the clobber) and */ \
+    /* their latency hides behind barrier arrival + the prior quadrant's*/ \
+    /* MFMAs.  Overwriting glds issues sit BELOW the barrier so a slot's*/ \
+    /* last reads and its overwrite are always barrier-separated.       */ \
+    if (Q == 1) {                                                           \
+      G8_AREAD(a_q[0], TPAR, 1, 0);                                         \
+      G8_AREAD(a_q[1], TPAR, 1, 1);                                         \
+    } else if (Q == 2) {                                                    \
+      G8_AREAD(a_q[0], TPAR, 2, 0);                                         \
+      G8_AREAD(a_q[1], TPAR, 2, 1);                                         \
+      G8_AREAD(a_q3[0], TPAR, 3, 0);                                        \
+      G8_AREAD(a_q3[1], TPAR, 3, 1);                                        \
+    }                                                                       \
+    asm volatile("" ::: "memory");                                          \
+    __builtin_amdgcn_s_barrier();                                           \
     if (Q == 1) {                                                           \
       if ((T) + 2 < nt)                                                     \
         g8_stage_half(G8_BSLOT(TPAR, 0), Bh0 + G8_KOFF((T) + 2), stg_off);  \
@@ -261,26 +280,14 @@ __global__ void gemm_nt_8phase_kernel(
         g8_stage_half(G8_ASLOT(TPAR, 0), Ah0 + G8_KOFF((T) + 2), stg_off);  \
         g8_stage_half(G8_ASLOT(TPAR, 1), Ah1 + G8_KOFF((T) + 2), stg_off);  \
       }                                                                     \
-    }                                                                       \
-    if (Q == 0) {                                                           \
+    } else {                                                                \
       _Pragma("unroll") for (int nf = 0; nf < 4; ++nf) {                    \
         G8_BREAD(b_t[nf][0], TPAR, nf, 0);                                  \
         G8_BREAD(b_t[nf][1], TPAR, nf, 1);                                  \
       }                                                                     \
+      G8_AREAD(a_q[0], TPAR, 0, 0);                                         \
+      G8_AREAD(a_q[1], TPAR, 0, 1);                                         \
     }                                                                       \
-    if (Q < 3) {                                                            \
-      G8_AREAD(a_q[0], TPAR, Q, 0);                                         \
-      G8_AREAD(a_q[1], TPAR, Q, 1);                                         \
-    }                                                                       \
-    if (Q == 2) {                                                           \
-      G8_AREAD(a_q3[0], TPAR, 3, 0);                                        \
-      G8_AREAD(a_q3[1], TPAR, 3, 1);                                        \
-    }                                                                       \
-    /* reads/issues stay ABOVE the barrier (pinned by the clobber) so   */ \
-    /* ds_read latency hides behind barrier arrival + prior MFMAs; the  */ \
-    /* compiler's counted lgkm waits gate the consuming MFMAs below.    */ \
-    asm volatile("" ::: "memory");                                          \
-    __builtin_amdgcn_s_barrier();                                           \
     __builtin_amdgcn_s_setprio(1);                                          \
     _Pragma("unroll") for (int nf = 0; nf < 4; ++nf)                        \
         _Pragma("unroll") for (int ks = 0; ks < 2; ++ks)                    \
