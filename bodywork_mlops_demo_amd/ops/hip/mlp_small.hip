@@ -171,7 +171,29 @@ __global__ void coldot_kernel(const bf16_t* __restrict__ m,
   const long long r0 = blockIdx.y * rows_per_chunk;
   const long long r1 = min(r0 + rows_per_chunk, n);
   float dot[8] = {0}, sum[8] = {0};
-  for (long long r = r0; r < r1; ++r) {
+  // 4-row unroll: independent 16-B loads in flight (single-row loop
+  // measured latency-bound at ~2.8 TB/s)
+  long long r = r0;
+  for (; r + 3 < r1; r += 4) {
+    bf16x8 raw[4];
+    float vv[4];
+#pragma unroll
+    for (int u = 0; u < 4; ++u) {
+      raw[u] = load_bf16x8(m + (r + u) * (long long)H + col0);
+      vv[u] = v[r + u];
+    }
+#pragma unroll
+    for (int u = 0; u < 4; ++u) {
+      float mv[8];
+      bf16x8_to_f32(raw[u], mv);
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        dot[e] = fmaf(mv[e], vv[u], dot[e]);
+        if (ALSO_COLSUM) sum[e] += mv[e];
+      }
+    }
+  }
+  for (; r < r1; ++r) {
     float vv = v[r];
     float mv[8];
     bf16x8_to_f32(load_bf16x8(m + r * (long long)H + col0), mv);
@@ -200,7 +222,10 @@ at::Tensor coldot_bf16_hip(const at::Tensor& m, const at::Tensor& v,
                        m.options().dtype(at::kFloat));
   auto stream = at::cuda::getCurrentCUDAStream();
   int gx = (H / 8 + 255) / 256;
-  int gy = (int)std::min<long long>((n + 511) / 512, 256);
+  // >=4 waves/SIMD: 256 y-chunks gave only 1 workgroup/CU (4 waves) and
+  // measured ~2-3.7 TB/s; 128-row chunks fill the chip (atomic fan-in
+  // per column stays <=1024, negligible)
+  int gy = (int)std::min<long long>((n + 127) / 128, 1024);
   float* dw = out.data_ptr<float>();
   float* cs = also_colsum ? dw + H : nullptr;
   if (also_colsum)
@@ -223,7 +248,21 @@ __global__ void colsum_kernel(const bf16_t* __restrict__ m,
   const long long r0 = blockIdx.y * rows_per_chunk;
   const long long r1 = min(r0 + rows_per_chunk, n);
   float sum[8] = {0};
-  for (long long r = r0; r < r1; ++r) {
+  long long r = r0;
+  for (; r + 3 < r1; r += 4) {
+    bf16x8 raw[4];
+#pragma unroll
+    for (int u = 0; u < 4; ++u)
+      raw[u] = load_bf16x8(m + (r + u) * (long long)H + col0);
+#pragma unroll
+    for (int u = 0; u < 4; ++u) {
+      float mv[8];
+      bf16x8_to_f32(raw[u], mv);
+#pragma unroll
+      for (int e = 0; e < 8; ++e) sum[e] += mv[e];
+    }
+  }
+  for (; r < r1; ++r) {
     float mv[8];
     bf16x8_to_f32(load_bf16x8(m + r * (long long)H + col0), mv);
 #pragma unroll
@@ -242,7 +281,7 @@ at::Tensor colsum_bf16_hip(const at::Tensor& m) {
   auto out = at::zeros({H}, m.options().dtype(at::kFloat));
   auto stream = at::cuda::getCurrentCUDAStream();
   int gx = (H / 8 + 255) / 256;
-  int gy = (int)std::min<long long>((n + 511) / 512, 256);
+  int gy = (int)std::min<long long>((n + 127) / 128, 1024);
   hipLaunchKernelGGL(colsum_kernel, dim3(gx, gy), dim3(256), 0, stream,
                      (const bf16_t*)m.data_ptr(), out.data_ptr<float>(), n, H);
   return out;
