@@ -211,9 +211,9 @@ __global__ void gemm_nt_8phase_kernel(
   const char* Bh1 = (const char*)(B + (n0 + 128) * K);
 #define G8_KOFF(T) ((long long)(T) * (G8_BK * 2))
 
-  // ---- prologue: tiles 0 and 1 fully issued (8 half-tiles) --------------
-  // Issue order matches the steady-state wait: at (0,0) `vmcnt(4)` keeps
-  // the NEWEST four (tile 1's B then A) in flight and proves tile 0 landed.
+  // ---- prologue: A(0), B(0), B(1) (6 half-tiles; A(1) is issued by the
+  // loop at phase (0,A)).  At the first wait `vmcnt(2)` keeps the newest
+  // two (tile 1's B halves) in flight and proves tile 0 landed.
   g8_stage_half(G8_ASLOT(0, 0), Ah0, stg_off);
   g8_stage_half(G8_ASLOT(0, 1), Ah1, stg_off);
   g8_stage_half(G8_BSLOT(0, 0), Bh0, stg_off);
@@ -221,8 +221,6 @@ __global__ void gemm_nt_8phase_kernel(
   if (nt > 1) {
     g8_stage_half(G8_BSLOT(1, 0), Bh0 + G8_KOFF(1), stg_off);
     g8_stage_half(G8_BSLOT(1, 1), Bh1 + G8_KOFF(1), stg_off);
-    g8_stage_half(G8_ASLOT(1, 0), Ah0 + G8_KOFF(1), stg_off);
-    g8_stage_half(G8_ASLOT(1, 1), Ah1 + G8_KOFF(1), stg_off);
   }
 
   // One phase = barrier + (issue) + LDS reads + 8 MFMAs.  Reads feed the
@@ -239,78 +237,84 @@ __global__ void gemm_nt_8phase_kernel(
   // ds_read happened in an EARLIER phase (barrier-separated): B(u) slots
   // read only at (u,0), re-staged at (u,1)/(u,2); A(u) slots last read at
   // (u,2) (q3 pre-read), re-staged at (u,3).
-#define G8_PHASE(T, TPAR, Q)                                                \
+// Two 32-MFMA phases per K-tile (v4): half the barriers of the
+// 4-phase schedule.  Phase A: wait+barrier, issue A(T+1), first-touch
+// reads of B(T) x8 + A q0,q1, MFMA q0+q1 (16).  Phase B: barrier,
+// issue B(T+2), read A q2,q3, MFMA q2+q3 (16).
+// Landing: at wait(T,A) the newest 2 outstanding glds are B(T+1)'s
+// halves (issued at (T-1,B)); the 3rd/4th-newest are A(T)'s halves
+// (issued at (T-1,A)) -> vmcnt(2) proves tile T landed; the last tile
+// drains with vmcnt(0).  Slot safety is completion-airtight: A(T+1)
+// overwrites A(T-1), whose q2/q3 reads were CONSUMED by MFMAs before
+// any wave passed barrier(T,A); B(T+2) overwrites B(T), whose reads
+// were consumed by MFMA q0/q1 before barrier(T,B).
+#define G8_PHASE_A(T, TPAR)                                                 \
   do {                                                                      \
-    if (Q == 0) {                                                           \
-      /* middle tiles: the 4 newest outstanding glds are tile T+1's     */ \
-      /* halves, so vmcnt(4) proves tile T landed.  On the LAST tile    */ \
-      /* nothing newer was issued (guards off) -> full drain once.      */ \
-      if ((T) + 1 < nt)                                                     \
-        asm volatile("s_waitcnt vmcnt(4)" ::: "memory");                    \
-      else                                                                  \
-        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");                    \
-    }                                                                       \
-    /* FIRST-TOUCH reads of tile T's slots (phase 0) must stay AFTER    */ \
-    /* the barrier: vmcnt is per-wave, and each slot is staged by ALL   */ \
-    /* 16 waves' glds, so only wait+barrier proves the whole half-tile  */ \
-    /* landed.  Later phases re-read already-validated tile-T data, so  */ \
-    /* their ds_reads sit ABOVE the barrier (pinned by the clobber) and */ \
-    /* their latency hides behind barrier arrival + the prior quadrant's*/ \
-    /* MFMAs.  Overwriting glds issues sit BELOW the barrier so a slot's*/ \
-    /* last reads and its overwrite are always barrier-separated.       */ \
-    if (Q == 1) {                                                           \
-      G8_AREAD(a_q[0], TPAR, 1, 0);                                         \
-      G8_AREAD(a_q[1], TPAR, 1, 1);                                         \
-    } else if (Q == 2) {                                                    \
-      G8_AREAD(a_q[0], TPAR, 2, 0);                                         \
-      G8_AREAD(a_q[1], TPAR, 2, 1);                                         \
-      G8_AREAD(a_q3[0], TPAR, 3, 0);                                        \
-      G8_AREAD(a_q3[1], TPAR, 3, 1);                                        \
-    }                                                                       \
-    asm volatile("" ::: "memory");                                          \
+    if ((T) + 1 < nt)                                                       \
+      asm volatile("s_waitcnt vmcnt(2)" ::: "memory");                      \
+    else                                                                    \
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");                      \
     __builtin_amdgcn_s_barrier();                                           \
-    if (Q == 1) {                                                           \
-      if ((T) + 2 < nt)                                                     \
-        g8_stage_half(G8_BSLOT(TPAR, 0), Bh0 + G8_KOFF((T) + 2), stg_off);  \
-    } else if (Q == 2) {                                                    \
-      if ((T) + 2 < nt)                                                     \
-        g8_stage_half(G8_BSLOT(TPAR, 1), Bh1 + G8_KOFF((T) + 2), stg_off);  \
-    } else if (Q == 3) {                                                    \
-      if ((T) + 2 < nt) {                                                   \
-        g8_stage_half(G8_ASLOT(TPAR, 0), Ah0 + G8_KOFF((T) + 2), stg_off);  \
-        g8_stage_half(G8_ASLOT(TPAR, 1), Ah1 + G8_KOFF((T) + 2), stg_off);  \
-      }                                                                     \
-    } else {                                                                \
-      _Pragma("unroll") for (int nf = 0; nf < 4; ++nf) {                    \
-        G8_BREAD(b_t[nf][0], TPAR, nf, 0);                                  \
-        G8_BREAD(b_t[nf][1], TPAR, nf, 1);                                  \
-      }                                                                     \
-      G8_AREAD(a_q[0], TPAR, 0, 0);                                         \
-      G8_AREAD(a_q[1], TPAR, 0, 1);                                         \
+    if ((T) + 1 < nt) {                                                     \
+      g8_stage_half(G8_ASLOT(1 - (TPAR), 0), Ah0 + G8_KOFF((T) + 1),        \
+                    stg_off);                                               \
+      g8_stage_half(G8_ASLOT(1 - (TPAR), 1), Ah1 + G8_KOFF((T) + 1),        \
+                    stg_off);                                               \
     }                                                                       \
+    _Pragma("unroll") for (int nf = 0; nf < 4; ++nf) {                      \
+      G8_BREAD(b_t[nf][0], TPAR, nf, 0);                                    \
+      G8_BREAD(b_t[nf][1], TPAR, nf, 1);                                    \
+    }                                                                       \
+    G8_AREAD(a_q[0], TPAR, 0, 0);                                           \
+    G8_AREAD(a_q[1], TPAR, 0, 1);                                           \
+    G8_AREAD(a_q3[0], TPAR, 1, 0);                                          \
+    G8_AREAD(a_q3[1], TPAR, 1, 1);                                          \
     __builtin_amdgcn_s_setprio(1);                                          \
     _Pragma("unroll") for (int nf = 0; nf < 4; ++nf)                        \
         _Pragma("unroll") for (int ks = 0; ks < 2; ++ks)                    \
-            acc[Q][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(           \
-                (Q == 3 ? a_q3[ks] : a_q[ks]), b_t[nf][ks], acc[Q][nf], 0,  \
-                0, 0);                                                      \
+            acc[0][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(           \
+                a_q[ks], b_t[nf][ks], acc[0][nf], 0, 0, 0);                 \
+    _Pragma("unroll") for (int nf = 0; nf < 4; ++nf)                        \
+        _Pragma("unroll") for (int ks = 0; ks < 2; ++ks)                    \
+            acc[1][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(           \
+                a_q3[ks], b_t[nf][ks], acc[1][nf], 0, 0, 0);                \
+    __builtin_amdgcn_s_setprio(0);                                          \
+  } while (0)
+
+#define G8_PHASE_B(T, TPAR)                                                 \
+  do {                                                                      \
+    __builtin_amdgcn_s_barrier();                                           \
+    if ((T) + 2 < nt) {                                                     \
+      g8_stage_half(G8_BSLOT(TPAR, 0), Bh0 + G8_KOFF((T) + 2), stg_off);    \
+      g8_stage_half(G8_BSLOT(TPAR, 1), Bh1 + G8_KOFF((T) + 2), stg_off);    \
+    }                                                                       \
+    G8_AREAD(a_q[0], TPAR, 2, 0);                                           \
+    G8_AREAD(a_q[1], TPAR, 2, 1);                                           \
+    G8_AREAD(a_q3[0], TPAR, 3, 0);                                          \
+    G8_AREAD(a_q3[1], TPAR, 3, 1);                                          \
+    __builtin_amdgcn_s_setprio(1);                                          \
+    _Pragma("unroll") for (int nf = 0; nf < 4; ++nf)                        \
+        _Pragma("unroll") for (int ks = 0; ks < 2; ++ks)                    \
+            acc[2][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(           \
+                a_q[ks], b_t[nf][ks], acc[2][nf], 0, 0, 0);                 \
+    _Pragma("unroll") for (int nf = 0; nf < 4; ++nf)                        \
+        _Pragma("unroll") for (int ks = 0; ks < 2; ++ks)                    \
+            acc[3][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(           \
+                a_q3[ks], b_t[nf][ks], acc[3][nf], 0, 0, 0);                \
     __builtin_amdgcn_s_setprio(0);                                          \
   } while (0)
 
   // two tiles per iteration: buffer indices stay compile-time (rule 20)
   for (long long t = 0; t < nt; t += 2) {
-    G8_PHASE(t, 0, 0);
-    G8_PHASE(t, 0, 1);
-    G8_PHASE(t, 0, 2);
-    G8_PHASE(t, 0, 3);
+    G8_PHASE_A(t, 0);
+    G8_PHASE_B(t, 0);
     if (t + 1 < nt) {
-      G8_PHASE(t + 1, 1, 0);
-      G8_PHASE(t + 1, 1, 1);
-      G8_PHASE(t + 1, 1, 2);
-      G8_PHASE(t + 1, 1, 3);
+      G8_PHASE_A(t + 1, 1);
+      G8_PHASE_B(t + 1, 1);
     }
   }
-#undef G8_PHASE
+#undef G8_PHASE_A
+#undef G8_PHASE_B
 #undef G8_AREAD
 #undef G8_BREAD
 #undef G8_ASLOT
