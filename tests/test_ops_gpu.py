@@ -241,6 +241,30 @@ class TestGemmGPU:
         ref = ops.linear_bf16(x, w, bias=b, relu=True)
         assert torch.equal(out, ref)
 
+    @pytest.mark.parametrize("m,n,k", [
+        (512, 256, 384),      # 8-phase dispatch (M,N%256, K%128)
+        (256, 512, 640),
+    ])
+    def test_linear_epilogues_8phase_shapes(self, m, n, k):
+        """The deep-pipeline 256^2 kernel (gemm8.hip) is the default for
+        M,N%256==0 / K%128==0 — exercise its bias+relu, mask-apply and
+        mask-emit epilogues against oracles on qualifying shapes."""
+        from bodywork_mlops_demo_amd.ops import reference
+
+        x = (torch.randn(m, k, device=DEV) * 0.5).bfloat16()
+        w = (torch.randn(n, k, device=DEV) * 0.5).bfloat16()
+        b = torch.randn(n, device=DEV).bfloat16()
+        ref = x.float() @ w.float().t()
+        got = ops.linear_bf16(x, w, bias=b, relu=True)
+        assert self._relerr(got.cpu(), torch.relu(ref + b.float()).cpu()) < 2e-2
+        out, mbits = ops.linear_relu_mask_bf16(x, w, b)
+        assert torch.equal(out, got)
+        assert torch.equal(mbits.cpu(),
+                           reference.pack_relu_mask(out.cpu().float() > 0))
+        applied = ops.linear_bf16(x, w, mask=mbits, out_fp32=True)
+        want = torch.where(out.float() > 0, ref, torch.zeros_like(ref))
+        assert self._relerr(applied.cpu(), want.cpu()) < 2e-2
+
     @pytest.mark.parametrize("r,m,n", [
         (256, 128, 128),
         (1000, 200, 130),
