@@ -26,6 +26,9 @@ import math
 import torch
 
 from bodywork_mlops_demo_amd import ops
+from bodywork_mlops_demo_amd.utils.logging import configure_logger
+
+log = configure_logger(__name__)
 
 
 class GPUMLPRegressor:
@@ -149,15 +152,34 @@ class GPUMLPRegressor:
 
             world = dist.get_world_size(process_group)
 
-        # single-GPU path: the ENTIRE Adam step (philox batch sampling,
-        # gather, forward, backward, fused Adam + shadow refresh) is one
-        # captured hipGraph, replayed per step — host dispatch of the ~30
-        # kernels was costing as much as the kernels themselves
-        if (process_group is None and self.device.type == "cuda"
-                and steps >= 4 and ops.hip_available()):
-            self._fit_captured(X.to(self.device), y.to(self.device), steps,
-                               min(batch_size, n), lr, seed)
-            return self
+        # graph path: the ENTIRE Adam step (philox batch sampling,
+        # gather, forward, backward, [DP: RCCL all-reduce], fused Adam +
+        # shadow refresh) is one captured hipGraph, replayed per step —
+        # host dispatch of the ~30 kernels was costing as much as the
+        # kernels themselves.  With a process_group the all-reduce is
+        # captured too (RCCL collectives are hipGraph-capturable); if
+        # capture fails on this build we fall back to the eager DP loop.
+        graph_ok = (self.device.type == "cuda" and steps >= 4
+                    and ops.hip_available()
+                    and not getattr(self, "_dp_graph_unsupported", False))
+        if graph_ok and process_group is not None:
+            import torch.distributed as dist
+
+            graph_ok = dist.get_backend(process_group) == "nccl"
+        if graph_ok:
+            try:
+                self._fit_captured(X.to(self.device), y.to(self.device),
+                                   steps, min(batch_size, n), lr, seed,
+                                   process_group)
+                return self
+            except RuntimeError:
+                if process_group is None:
+                    raise
+                log.warning("hipGraph capture of the DP step failed; "
+                            "falling back to the eager per-step loop")
+                self._dp_graph_unsupported = True
+                if getattr(self, "_train_static", None) is not None:
+                    self._train_static["graph"] = None
 
         for _ in range(steps):
             idx = torch.randint(0, n, (min(batch_size, n),), generator=g)
@@ -209,7 +231,8 @@ class GPUMLPRegressor:
         self.w3_bf.copy_(self.w3.bfloat16())
         return self
 
-    def _fit_captured(self, X, y, steps: int, bs: int, lr: float, seed: int):
+    def _fit_captured(self, X, y, steps: int, bs: int, lr: float, seed: int,
+                      process_group=None):
         """hipGraph-captured Adam steps (see fit()).  Bias correction is a
         device [2]-tensor the graph reads; the philox batch counter and
         data size live on-device; the day's data is copied into persistent
@@ -217,8 +240,14 @@ class GPUMLPRegressor:
         the drift loop (capture cost paid once, replay ~zero host cost)."""
         n = X.shape[0]
         beta1, beta2 = 0.9, 0.999
+        world = 1
+        if process_group is not None:
+            import torch.distributed as dist
+
+            world = dist.get_world_size(process_group)
         st = getattr(self, "_train_static", None)
-        if st is None or st["cap"] < n or st["bs"] != bs or st["lr"] != lr:
+        if (st is None or st["cap"] < n or st["bs"] != bs or st["lr"] != lr
+                or st.get("world", 1) != world):
             cap = max(n, int(st["cap"]) if st else 0)
             st = {
                 "cap": cap, "bs": bs, "lr": lr,
@@ -227,7 +256,7 @@ class GPUMLPRegressor:
                 "n_dev": torch.zeros(1, dtype=torch.int64, device=self.device),
                 "ctr": torch.zeros(1, dtype=torch.int64, device=self.device),
                 "bc": torch.ones(2, device=self.device),
-                "graph": None,
+                "graph": None, "world": world,
             }
             self._train_static = st
         st["X"][:n].copy_(X)
@@ -241,6 +270,19 @@ class GPUMLPRegressor:
             xb = st["X"].index_select(0, idx)
             yb = st["y"].index_select(0, idx)
             grads = self._step_grads(xb, yb)
+            if process_group is not None:
+                import torch.distributed as dist
+
+                # one flat fp32 bucket -> one RCCL launch per step, same
+                # as the eager DP loop; allocations made during capture
+                # come from the graph pool and are stable across replays
+                flat = torch.cat([gr.reshape(-1) for gr in grads])
+                dist.all_reduce(flat, group=process_group)
+                flat /= world
+                off = 0
+                for i, gr in enumerate(grads):
+                    grads[i] = flat[off:off + gr.numel()].view_as(gr)
+                    off += gr.numel()
             shadows = [self.w1_bf, self.b1_bf, self.W2wt_bf, self.b2_bf,
                        self.w3_bf, None]
             for p, gr, (m, v), sh in zip(self.parameters(), grads,
