@@ -74,3 +74,33 @@ def test_cycle_with_store_gpu(tmp_path):
     assert len(store.list_keys(contract.MODELS_PREFIX)) == 2
     artefact, _ = store.get_latest_model()
     assert type(artefact).__name__ == "LinearRegression"
+
+
+def test_mlp_dp_graph_captured_allreduce():
+    """world=1 RCCL group on one GPU: the captured training step must
+    record the in-graph all-reduce (models/mlp.py fit DP path) without
+    tripping the eager fallback."""
+    import os
+
+    import torch.distributed as dist
+
+    from bodywork_mlops_demo_amd.models.mlp import GPUMLPRegressor
+
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    os.environ.setdefault("MASTER_PORT", "29411")
+    dist.init_process_group("nccl", rank=0, world_size=1)
+    try:
+        pg = dist.group.WORLD
+        m = GPUMLPRegressor(hidden=1024, device=DEV)
+        X = torch.rand(16384, device=DEV) * 100
+        y = 1.0 + 0.5 * X + torch.randn_like(X) * 10
+        mse0 = float(((m.predict(X) - y) ** 2).mean())
+        m.fit(X, y, steps=40, batch_size=8192, lr=3e-3, process_group=pg)
+        assert not getattr(m, "_dp_graph_unsupported", False), \
+            "capture of the RCCL all-reduce fell back to the eager loop"
+        assert m._train_static["graph"] is not None
+        assert m._train_static["world"] == 1
+        mse = float(((m.predict(X) - y) ** 2).mean())
+        assert mse < 0.5 * mse0, (mse, mse0)
+    finally:
+        dist.destroy_process_group()
