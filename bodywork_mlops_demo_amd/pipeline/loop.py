@@ -67,6 +67,13 @@ def run_loop(
     if store is not None:
         try:
             y_np, X_np, _ = store.get_all_datasets()
+            if world_size > 1:
+                # cycles generate per-rank shards; resuming from the
+                # store must re-shard the merged history the same way
+                # (train.py's rank::world convention) or every rank
+                # would train on duplicated data
+                y_np = y_np[rank::world_size].copy()
+                X_np = X_np[rank::world_size].copy()
             state.append_day(
                 torch.from_numpy(y_np).to(device),
                 torch.from_numpy(X_np).to(device),
@@ -88,6 +95,19 @@ def run_loop(
         results.append(r)
         offline_mape = (r["offline"] or {}).get("MAPE", float("inf"))
         drifted = r["online"]["MAPE"] > drift_threshold * offline_mape
+        if process_group is not None:
+            # collective decision: every rank must take the same
+            # train/skip branch next cycle (a rank-local decision would
+            # deadlock the training all-reduce); any drifted rank
+            # triggers a fleet-wide retrain
+            import torch.distributed as dist
+
+            flag = torch.tensor(
+                [1.0 if drifted else 0.0],
+                device=device if torch.cuda.is_available() else "cpu",
+            )
+            dist.all_reduce(flag, op=dist.ReduceOp.MAX, group=process_group)
+            drifted = bool(flag.item() > 0)
         t = r["timings"]
         log.info(
             f"cycle {day + 1}/{days} ({state.date}): "

@@ -92,3 +92,40 @@ def test_dp_mlp_ranks_stay_in_sync():
     # identical parameters on both ranks after all-reduced updates
     for a, b in zip(results[0], results[1]):
         assert a == pytest.approx(b, rel=1e-5, abs=1e-5)
+
+
+def _drift_loop_worker(rank, world, port, out):
+    os.environ.update(
+        RANK=str(rank), WORLD_SIZE=str(world),
+        MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port), LOCAL_RANK=str(rank),
+    )
+    import torch.distributed as dist
+
+    from bodywork_mlops_demo_amd.parallel import init_distributed
+    from bodywork_mlops_demo_amd.pipeline.loop import run_loop
+
+    init_distributed(backend="gloo")
+    results = run_loop(
+        None, days=6, n_rows=600, model_type="linear", device="cpu",
+        start_date="2026-04-01", process_group=dist.group.WORLD,
+        rank=rank, world_size=world,
+        retrain_policy="drift", drift_threshold=2.5,
+    )
+    out[rank] = [r["timings"]["train_s"] > 0 for r in results]
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(180)
+def test_dp_drift_policy_decision_is_collective():
+    """retrain_policy='drift' in a DP world: every rank must take the
+    SAME train/skip branch each cycle (a rank-local decision would
+    deadlock the training all-reduce — pipeline/loop.py all-reduces the
+    drift flag with MAX)."""
+    ctx = mp.get_context("spawn")
+    with ctx.Manager() as mgr:
+        out = mgr.dict()
+        mp.spawn(_drift_loop_worker, args=(2, 29617, out), nprocs=2, join=True)
+        results = dict(out)
+    assert results[0] == results[1], "ranks diverged on retrain decisions"
+    assert results[0][0]  # first day always trains
