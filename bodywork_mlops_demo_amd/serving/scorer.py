@@ -63,9 +63,24 @@ class BatchedScorer:
             for _ in range(2):
                 self.model.predict(x_static[: min(b, 1024)])
         torch.cuda.current_stream().wait_stream(s)
-        graph = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(graph):
-            y_static = self.model.predict(x_static)
+        # thread_local capture mode: the serving process has background
+        # threads (uvicorn workers, torch allocator/event threads) whose
+        # harmless HIP queries abort a *global*-mode capture with
+        # "operation not permitted when stream is capturing" — observed
+        # intermittently with 2 replicas + the runner watchdog.  One
+        # retry covers transient races during capture setup itself.
+        y_static = None
+        for attempt in (0, 1):
+            graph = torch.cuda.CUDAGraph()
+            try:
+                with torch.cuda.graph(graph,
+                                      capture_error_mode="thread_local"):
+                    y_static = self.model.predict(x_static)
+                break
+            except RuntimeError:
+                if attempt == 1:
+                    raise
+                torch.cuda.synchronize()
         self._graphs[b] = (graph, x_static, y_static)
         log.info(f"captured scoring hipGraph for batch bucket {b}")
 
