@@ -15,6 +15,8 @@ from __future__ import annotations
 
 import torch
 
+from bodywork_mlops_demo_amd.utils.device import canonical_device
+
 from bodywork_mlops_demo_amd import ops
 
 
@@ -24,7 +26,7 @@ class GPULinearRegressor:
     def __init__(self, intercept: float = 0.0, coef: float = 0.0, device="cpu"):
         self.intercept_ = float(intercept)
         self.coef_ = float(coef)
-        self.device = torch.device(device)
+        self.device = canonical_device(device)
         self._ab: torch.Tensor | None = None  # device-resident [a, b]
 
     def _ab_tensor(self) -> torch.Tensor:
@@ -73,7 +75,7 @@ class GPULinearRegressor:
         return ops.linear_score(X, self.intercept_, self.coef_)
 
     def to(self, device):
-        self.device = torch.device(device)
+        self.device = canonical_device(device)
         self._ab = None
         return self
 

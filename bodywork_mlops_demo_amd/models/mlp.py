@@ -25,6 +25,8 @@ import math
 
 import torch
 
+from bodywork_mlops_demo_amd.utils.device import canonical_device
+
 from bodywork_mlops_demo_amd import ops
 from bodywork_mlops_demo_amd.utils.logging import configure_logger
 
@@ -42,7 +44,7 @@ class GPUMLPRegressor:
 
     def __init__(self, hidden: int = 4096, device="cpu", seed: int = 7):
         self.hidden = hidden
-        self.device = torch.device(device)
+        self.device = canonical_device(device)
         g = torch.Generator(device="cpu").manual_seed(seed)
         h = hidden
         # He init, fp32 master weights
@@ -74,7 +76,7 @@ class GPUMLPRegressor:
         self.w3_bf = self.w3.bfloat16()
 
     def to(self, device):
-        self.device = torch.device(device)
+        self.device = canonical_device(device)
         self._to_device()
         return self
 
@@ -379,7 +381,7 @@ class GPUMLPRegressor:
         h = m.coefs_[1].shape[0]
         self = cls.__new__(cls)
         self.hidden = h
-        self.device = torch.device(device)
+        self.device = canonical_device(device)
         wr = torch.from_numpy(m.coefs_[0].reshape(-1).copy()).float()
         br = torch.from_numpy(m.intercepts_[0].copy()).float()
         # invert the raw-x fold: wr = w1/sigma, br = b1 - wr*mu

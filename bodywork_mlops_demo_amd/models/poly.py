@@ -23,6 +23,8 @@ from __future__ import annotations
 import numpy as np
 import torch
 
+from bodywork_mlops_demo_amd.utils.device import canonical_device
+
 from bodywork_mlops_demo_amd import ops
 
 
@@ -35,7 +37,7 @@ class GPUPolyRegressor:
             raise ValueError("degree must be in [1, 5]")
         self.degree = degree
         self.l2 = float(l2)
-        self.device = torch.device(device)
+        self.device = canonical_device(device)
         self.coef_t_: list[float] = [0.0] * (degree + 1)  # normalised basis
         self._coef_dev: torch.Tensor | None = None
 
@@ -69,7 +71,7 @@ class GPUPolyRegressor:
         return ops.poly_score(X, self.coef_t_, self.X_MU, self.X_SIGMA)
 
     def to(self, device):
-        self.device = torch.device(device)
+        self.device = canonical_device(device)
         self._coef_dev = None
         return self
 

@@ -17,6 +17,8 @@ from __future__ import annotations
 import numpy as np
 import torch
 
+from bodywork_mlops_demo_amd.utils.device import canonical_device
+
 from bodywork_mlops_demo_amd.utils.logging import configure_logger
 
 log = configure_logger(__name__)
@@ -28,7 +30,7 @@ class BatchedScorer:
     def __init__(self, model, device: str | torch.device = "cpu",
                  use_graphs: bool = True):
         self.model = model.to(device)
-        self.device = torch.device(device)
+        self.device = canonical_device(device)
         self.use_graphs = use_graphs and self.device.type == "cuda"
         self._graphs: dict[int, tuple] = {}
         # models with large per-row activations advertise a chunk bound
@@ -106,7 +108,12 @@ class BatchedScorer:
             try:
                 self._capture(b)
             except Exception as e:  # capture unavailable — direct launches
-                log.warning(f"hipGraph capture failed ({e}); direct launch")
+                import traceback
+
+                log.warning(f"hipGraph capture failed for bucket {b} "
+                            f"(model={type(self.model).__name__}, "
+                            f"graphs={sorted(self._graphs)}): {e}; "
+                            f"direct launch\n{traceback.format_exc()}")
                 self.use_graphs = False
                 return self.model.predict(X)
         graph, x_static, y_static = self._graphs[b]

@@ -104,3 +104,26 @@ def test_mlp_dp_graph_captured_allreduce():
         assert mse < 0.5 * mse0, (mse, mse0)
     finally:
         dist.destroy_process_group()
+
+
+def test_scorer_captures_with_indexless_device():
+    """Regression: device string "cuda" (no index) must behave like
+    "cuda:0" — the coefficient cache used to miss on every predict
+    (torch.device("cuda") != torch.device("cuda:0")), re-allocating
+    inside graph capture and aborting it with 'operation not permitted
+    when stream is capturing' (seen in runner-launched serving replicas,
+    which pass no --device)."""
+    from bodywork_mlops_demo_amd.models import GPULinearRegressor
+    from bodywork_mlops_demo_amd.serving.scorer import BatchedScorer
+
+    m = GPULinearRegressor(1.0, 0.5, device="cuda")
+    assert m._ab_tensor() is m._ab_tensor(), "coefficient tensor not cached"
+    scorer = BatchedScorer(m, "cuda")
+    X = torch.rand(1500, device=DEV) * 100
+    y = scorer.score_tensor(X)
+    assert scorer.use_graphs and scorer._graphs, \
+        "scoring hipGraph was not captured for an index-less device"
+    want = 1.0 + 0.5 * X
+    assert torch.allclose(y, want, rtol=1e-5)
+    # replay path (second call) must agree
+    assert torch.allclose(scorer.score_tensor(X), want, rtol=1e-5)
