@@ -26,6 +26,23 @@ timeout 700 python bench.py --model mlp --rows 1000000 --steps 3 --warmup 2 \
   --mlp-steps 20 > "$OUT/bench_mlp_1m.json" 2>/dev/null
 tail -1 "$OUT/bench_mlp_1m.json" | tee -a "$OUT/summary.txt"
 
+log "config 5b: MLP-4096, 10M rows"
+timeout 700 python bench.py --model mlp --rows 10000000 --steps 2 --warmup 1 \
+  > "$OUT/bench_mlp_10m.json" 2>/dev/null
+tail -1 "$OUT/bench_mlp_10m.json" | tee -a "$OUT/summary.txt"
+
+log "full subprocess DAG (runner, 2 serving replicas)"
+python - <<'PYEOF'
+from datetime import date
+from bodywork_mlops_demo_amd.store import LocalStore
+from bodywork_mlops_demo_amd.stages import datagen
+datagen.run(LocalStore("/tmp/cb_dagstore"), n=1440, date=date(2026, 2, 1),
+            device="cuda:0")
+PYEOF
+timeout 400 python bodywork_mlops_demo_amd/pipeline/runner.py pipeline.yaml \
+  --store /tmp/cb_dagstore > "$OUT/dag.log" 2>&1
+grep -E "captured scoring|pipeline run" "$OUT/dag.log" | tail -2 | tee -a "$OUT/summary.txt"
+
 log "config 4: 30-day drift loop (reference scale)"
 timeout 500 python -m bodywork_mlops_demo_amd loop --days 30 --rows 1440 \
   --store "$OUT/loopstore" --device cuda:0 \
