@@ -127,3 +127,24 @@ def test_scorer_captures_with_indexless_device():
     assert torch.allclose(y, want, rtol=1e-5)
     # replay path (second call) must agree
     assert torch.allclose(scorer.score_tensor(X), want, rtol=1e-5)
+
+
+def test_scorer_hot_redeploy_through_captured_graphs():
+    """update_model() must retarget every captured graph WITHOUT
+    recapture: the kernels read coefficients through device tensors the
+    swap copies into (serving hot-redeploy across daily retrains)."""
+    from bodywork_mlops_demo_amd.models import GPULinearRegressor
+    from bodywork_mlops_demo_amd.serving.scorer import BatchedScorer
+
+    scorer = BatchedScorer(GPULinearRegressor(1.0, 0.5, device=DEV), DEV)
+    X = torch.rand(2000, device=DEV) * 100
+    y1 = scorer.score_tensor(X)
+    n_graphs = len(scorer._graphs)
+    assert n_graphs >= 1
+    assert torch.allclose(y1, 1.0 + 0.5 * X, rtol=1e-5)
+
+    assert scorer.update_model(GPULinearRegressor(-3.0, 2.0, device=DEV))
+    assert len(scorer._graphs) == n_graphs, "hot swap must not recapture"
+    y2 = scorer.score_tensor(X)
+    assert torch.allclose(y2, -3.0 + 2.0 * X, rtol=1e-5), \
+        "captured graph still scoring with the old weights"
