@@ -324,6 +324,273 @@ __global__ void gemm_nt_8phase_kernel(
       acc, bias, mask, mask_out, C, M, N, m0, n0, wm, wn, fl, kg);
 }
 
+// ---- 8-wave / 512-thread variant (env BODYWORK_GEMM_WAVES=8) --------------
+//
+// Same 256^2 tile and LDS layout, but 8 waves in a 2(M)x4(N) grid with
+// 128x64 output per wave: acc[8][4] = 128 VGPR at the 2-waves/SIMD /
+// 256-VGPR budget (the guide's verified geometry).  B is read once per
+// tile (8 b128), A in 2-mfrag pairs per phase (4 b128); peak register
+// liveness ~= acc 128 + B 32 + A 2x16 + addressing ~ 220.  Half-tile
+// staging needs TWO wave-glds at 512 threads (chunks tid and tid+512).
+//
+// Phase schedule per tile u (one barrier per 16-MFMA phase):
+//   (u,0): wait vmcnt(4) [vmcnt(0) on last tile]; barrier;
+//          issue A0,A1(u+1) [4 glds]; read B(u) x8 + A mf01; MFMA mf01
+//   (u,1): read A mf23 (pre-barrier); barrier; issue B0(u+2); MFMA mf23
+//   (u,2): read A mf45+mf67 (pre-barrier); barrier; issue B1(u+2); MFMA mf45
+//   (u,3): barrier; MFMA mf67 (pre-read registers; no reads, no issues)
+// Landing: at wait(u,0) the 4 newest outstanding glds are B(u+1)'s
+// (2 per half x 2 halves, issued at (u-1,1)/(u-1,2)); the next-newest
+// are A(u)'s 4 (issued at (u-1,0)) -> vmcnt(4) proves tile u landed.
+// Slot safety mirrors the 16-wave kernel: first-touch reads post-
+// barrier; A(u+1) (issued post-barrier(u,0)) overwrites A(u-1), whose
+// mf45/mf67 reads were consumed by MFMAs before barrier(u,0);
+// B(u+2) overwrites B(u), read at (u,0) and consumed before (u,1)'s
+// barrier -> every overwrite is completion-separated by a barrier.
+
+#define G8W_THREADS 512
+
+__device__ __forceinline__ void g8w_stage_half(short* __restrict__ slot,
+                                               const char* __restrict__ base,
+                                               int off0, int off1) {
+  const int wave = threadIdx.x >> 6;
+  g8_glds16(base + off0, (char*)slot + wave * 1024);
+  g8_glds16(base + off1, (char*)slot + 8192 + wave * 1024);
+}
+
+template <int EPI, bool HAS_BIAS, bool OUT_FP32, bool EMIT_MASK>
+__device__ __forceinline__ void g8w_epilogue(
+    g8_f32x4 (&acc)[8][4], const float* __restrict__ bias,
+    const unsigned char* __restrict__ mask,
+    unsigned char* __restrict__ mask_out, void* __restrict__ C, long long M,
+    long long N, long long m0, long long n0, int wm2, int wn4, int fl,
+    int kg) {
+#pragma unroll
+  for (int i = 0; i < 8; ++i) {
+    unsigned long long mrow[4];
+    if (EPI == G8_EPI_MASK) {
+      const long long stripe = n0 + wn4 * 64;
+      const bool full = stripe + 64 <= N;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        long long row = m0 + wm2 * 128 + i * 16 + kg * 4 + r;
+        if (row >= M) {
+          mrow[r] = 0ull;
+        } else if (full) {
+          mrow[r] = *(const unsigned long long*)(mask + row * (N >> 3) +
+                                                 (stripe >> 3));
+        } else {
+          unsigned long long v = 0;
+          for (int b8 = 0; b8 < 8; ++b8)
+            if (stripe + b8 * 8 < N)
+              v |= (unsigned long long)
+                       mask[row * (N >> 3) + ((stripe >> 3) + b8)]
+                   << (8 * b8);
+          mrow[r] = v;
+        }
+      }
+    }
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      long long col = n0 + wn4 * 64 + j * 16 + fl;
+      bool col_ok = col < N;
+      float bval =
+          (EPI == G8_EPI_BIAS_RELU && HAS_BIAS && col_ok) ? bias[col] : 0.0f;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        long long row = m0 + wm2 * 128 + i * 16 + kg * 4 + r;
+        bool row_ok = row < M;
+        float v = acc[i][j][r];
+        if (EPI == G8_EPI_BIAS_RELU) {
+          v += bval;
+          v = fmaxf(v, 0.0f);
+        } else if (EPI == G8_EPI_MASK) {
+          v = (mrow[r] >> (j * 16 + fl)) & 1 ? v : 0.0f;
+        }
+        if (EMIT_MASK) {
+          unsigned long long b = __ballot(v > 0.0f);
+          if (fl == 0 && row_ok && col < N) {
+            unsigned short bits = (unsigned short)((b >> (kg * 16)) & 0xFFFF);
+            *(unsigned short*)(mask_out + row * (N >> 3) + (col >> 3)) = bits;
+          }
+        }
+        if (row_ok && col_ok) {
+          if (OUT_FP32)
+            ((float*)C)[row * N + col] = v;
+          else
+            ((bf16_t*)C)[row * N + col] = f32_to_bf16(v);
+        }
+      }
+    }
+  }
+}
+
+template <int EPI, bool HAS_BIAS, bool OUT_FP32, bool EMIT_MASK = false>
+__launch_bounds__(G8W_THREADS)
+__global__ void gemm_nt_8phase_w8_kernel(
+    const bf16_t* __restrict__ A, const bf16_t* __restrict__ B,
+    const float* __restrict__ bias, const unsigned char* __restrict__ mask,
+    unsigned char* __restrict__ mask_out, void* __restrict__ C, long long M,
+    long long N, long long K) {
+  __shared__ short lds[8 * G8_HT];
+  const long long m0 = (long long)blockIdx.y * G8_BM;
+  const long long n0 = (long long)blockIdx.x * G8_BN;
+  const int wave = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int wm2 = wave >> 2;  // 0..1 (128 C-rows per wave)
+  const int wn4 = wave & 3;   // 0..3 (64 C-cols per wave)
+  const int fl = lane & 15;
+  const int kg = lane >> 4;
+  const int swz = fl & 7;
+  const long long nt = K / G8_BK;
+
+#define G8W_ASLOT(buf) (lds + ((buf) * 2 + wm2) * G8_HT)
+#define G8W_ASLOTH(buf, half) (lds + ((buf) * 2 + (half)) * G8_HT)
+#define G8W_BSLOT(buf) (lds + 4 * G8_HT + ((buf) * 2 + (wn4 >> 1)) * G8_HT)
+#define G8W_BSLOTH(buf, half) (lds + 4 * G8_HT + ((buf) * 2 + (half)) * G8_HT)
+  const int b_inhalf = (wn4 & 1) * 64;
+
+  g8_f32x4 acc[8][4];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  // wave reads its OWN A half (wm2): in-half row = mfrag*16 + fl
+#define G8W_AREAD(dst, buf, mfrag, ks)                                      \
+  dst = ((const lds_vec*)(G8W_ASLOT(buf) + ((mfrag) * 16 + fl) * 64 +       \
+                          (((ks) * 4 + kg) ^ swz) * 8))                     \
+            ->v
+#define G8W_BREAD(dst, buf, nfrag, ks)                                      \
+  dst = ((const lds_vec*)(G8W_BSLOT(buf) +                                  \
+                          (b_inhalf + (nfrag) * 16 + fl) * 64 +             \
+                          (((ks) * 4 + kg) ^ swz) * 8))                     \
+            ->v
+
+  bf16x8_v a_lo[4];    // current pair's A fragments [mfrag&1][ks]
+  bf16x8_v a_hi[4];    // pre-read pair (mf67) for phase 3
+  bf16x8_v b_t[4][2];  // the tile's B fragments [nfrag][ks]
+
+  // two tile-invariant per-lane staging byte offsets (chunks tid, tid+512)
+  int stg0, stg1;
+  {
+    int c0 = (int)threadIdx.x, c1 = c0 + 512;
+    int r0 = c0 >> 3, r1 = c1 >> 3;
+    stg0 = (int)((r0 * K + ((c0 & 7) ^ (r0 & 7)) * 8) * 2);
+    stg1 = (int)((r1 * K + ((c1 & 7) ^ (r1 & 7)) * 8) * 2);
+  }
+  const char* Ah0 = (const char*)(A + m0 * K);
+  const char* Ah1 = (const char*)(A + (m0 + 128) * K);
+  const char* Bh0 = (const char*)(B + n0 * K);
+  const char* Bh1 = (const char*)(B + (n0 + 128) * K);
+
+  // prologue: A(0), B(0), B(1) — A(1) is issued by the loop at (0,0)
+  g8w_stage_half(G8W_ASLOTH(0, 0), Ah0, stg0, stg1);
+  g8w_stage_half(G8W_ASLOTH(0, 1), Ah1, stg0, stg1);
+  g8w_stage_half(G8W_BSLOTH(0, 0), Bh0, stg0, stg1);
+  g8w_stage_half(G8W_BSLOTH(0, 1), Bh1, stg0, stg1);
+  if (nt > 1) {
+    g8w_stage_half(G8W_BSLOTH(1, 0), Bh0 + G8_KOFF(1), stg0, stg1);
+    g8w_stage_half(G8W_BSLOTH(1, 1), Bh1 + G8_KOFF(1), stg0, stg1);
+  }
+
+#define G8W_MFMA_PAIR(mf0, ks_arr_lo)                                       \
+  _Pragma("unroll") for (int nf = 0; nf < 4; ++nf)                          \
+      _Pragma("unroll") for (int ks = 0; ks < 2; ++ks)                      \
+          acc[(mf0)][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(         \
+              ks_arr_lo[ks], b_t[nf][ks], acc[(mf0)][nf], 0, 0, 0)
+
+#define G8W_PHASE(T, TPAR, Q)                                               \
+  do {                                                                      \
+    if (Q == 1) {                                                           \
+      G8W_AREAD(a_lo[0], TPAR, 2, 0);                                       \
+      G8W_AREAD(a_lo[1], TPAR, 2, 1);                                       \
+      G8W_AREAD(a_lo[2], TPAR, 3, 0);                                       \
+      G8W_AREAD(a_lo[3], TPAR, 3, 1);                                       \
+    } else if (Q == 2) {                                                    \
+      G8W_AREAD(a_lo[0], TPAR, 4, 0);                                       \
+      G8W_AREAD(a_lo[1], TPAR, 4, 1);                                       \
+      G8W_AREAD(a_lo[2], TPAR, 5, 0);                                       \
+      G8W_AREAD(a_lo[3], TPAR, 5, 1);                                       \
+      G8W_AREAD(a_hi[0], TPAR, 6, 0);                                       \
+      G8W_AREAD(a_hi[1], TPAR, 6, 1);                                       \
+      G8W_AREAD(a_hi[2], TPAR, 7, 0);                                       \
+      G8W_AREAD(a_hi[3], TPAR, 7, 1);                                       \
+    }                                                                       \
+    if (Q == 0) {                                                           \
+      if ((T) + 1 < nt)                                                     \
+        asm volatile("s_waitcnt vmcnt(4)" ::: "memory");                    \
+      else                                                                  \
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");                    \
+    }                                                                       \
+    asm volatile("" ::: "memory");                                          \
+    __builtin_amdgcn_s_barrier();                                           \
+    if (Q == 0) {                                                           \
+      if ((T) + 1 < nt) {                                                   \
+        g8w_stage_half(G8W_ASLOTH(1 - (TPAR), 0), Ah0 + G8_KOFF((T) + 1),   \
+                       stg0, stg1);                                         \
+        g8w_stage_half(G8W_ASLOTH(1 - (TPAR), 1), Ah1 + G8_KOFF((T) + 1),   \
+                       stg0, stg1);                                         \
+      }                                                                     \
+      _Pragma("unroll") for (int nf = 0; nf < 4; ++nf) {                    \
+        G8W_BREAD(b_t[nf][0], TPAR, nf, 0);                                 \
+        G8W_BREAD(b_t[nf][1], TPAR, nf, 1);                                 \
+      }                                                                     \
+      G8W_AREAD(a_lo[0], TPAR, 0, 0);                                       \
+      G8W_AREAD(a_lo[1], TPAR, 0, 1);                                       \
+      G8W_AREAD(a_lo[2], TPAR, 1, 0);                                       \
+      G8W_AREAD(a_lo[3], TPAR, 1, 1);                                       \
+    } else if (Q == 1) {                                                    \
+      if ((T) + 2 < nt)                                                     \
+        g8w_stage_half(G8W_BSLOTH(TPAR, 0), Bh0 + G8_KOFF((T) + 2), stg0,   \
+                       stg1);                                               \
+    } else if (Q == 2) {                                                    \
+      if ((T) + 2 < nt)                                                     \
+        g8w_stage_half(G8W_BSLOTH(TPAR, 1), Bh1 + G8_KOFF((T) + 2), stg0,   \
+                       stg1);                                               \
+    }                                                                       \
+    __builtin_amdgcn_s_setprio(1);                                          \
+    if (Q == 0) {                                                           \
+      G8W_MFMA_PAIR(0, (a_lo + 0));                                         \
+      G8W_MFMA_PAIR(1, (a_lo + 2));                                         \
+    } else if (Q == 1) {                                                    \
+      G8W_MFMA_PAIR(2, (a_lo + 0));                                         \
+      G8W_MFMA_PAIR(3, (a_lo + 2));                                         \
+    } else if (Q == 2) {                                                    \
+      G8W_MFMA_PAIR(4, (a_lo + 0));                                         \
+      G8W_MFMA_PAIR(5, (a_lo + 2));                                         \
+    } else {                                                                \
+      G8W_MFMA_PAIR(6, (a_hi + 0));                                         \
+      G8W_MFMA_PAIR(7, (a_hi + 2));                                         \
+    }                                                                       \
+    __builtin_amdgcn_s_setprio(0);                                          \
+  } while (0)
+
+  for (long long t = 0; t < nt; t += 2) {
+    G8W_PHASE(t, 0, 0);
+    G8W_PHASE(t, 0, 1);
+    G8W_PHASE(t, 0, 2);
+    G8W_PHASE(t, 0, 3);
+    if (t + 1 < nt) {
+      G8W_PHASE(t + 1, 1, 0);
+      G8W_PHASE(t + 1, 1, 1);
+      G8W_PHASE(t + 1, 1, 2);
+      G8W_PHASE(t + 1, 1, 3);
+    }
+  }
+#undef G8W_PHASE
+#undef G8W_MFMA_PAIR
+#undef G8W_AREAD
+#undef G8W_BREAD
+#undef G8W_ASLOT
+#undef G8W_ASLOTH
+#undef G8W_BSLOT
+#undef G8W_BSLOTH
+
+  g8w_epilogue<EPI, HAS_BIAS, OUT_FP32, EMIT_MASK>(
+      acc, bias, mask, mask_out, C, M, N, m0, n0, wm2, wn4, fl, kg);
+}
+
 // ---- launcher (called from gemm.hip's dispatch) ---------------------------
 void launch_gemm8(int epi, bool has_bias, bool out_fp32, bool emit_mask,
                   const void* ap, const void* bp, const float* bias,
@@ -334,10 +601,21 @@ void launch_gemm8(int epi, bool has_bias, bool out_fp32, bool emit_mask,
   auto stream = at::cuda::getCurrentCUDAStream();
   const bf16_t* a = (const bf16_t*)ap;
   const bf16_t* b = (const bf16_t*)bp;
+  static int w8 = [] {  // A/B: 8-wave/512-thread variant (2 waves/SIMD)
+    const char* e = getenv("BODYWORK_GEMM_WAVES");
+    return (e && e[0] == '8') ? 1 : 0;
+  }();
 #define L8(EPI_, HB_, OF_, EM_)                                             \
-  hipLaunchKernelGGL((gemm_nt_8phase_kernel<EPI_, HB_, OF_, EM_>), grid,    \
-                     dim3(G8_THREADS), 0, stream, a, b, bias, mask,         \
-                     mask_out, cp, M, N, K)
+  do {                                                                      \
+    if (w8)                                                                 \
+      hipLaunchKernelGGL((gemm_nt_8phase_w8_kernel<EPI_, HB_, OF_, EM_>),   \
+                         grid, dim3(G8W_THREADS), 0, stream, a, b, bias,    \
+                         mask, mask_out, cp, M, N, K);                      \
+    else                                                                    \
+      hipLaunchKernelGGL((gemm_nt_8phase_kernel<EPI_, HB_, OF_, EM_>),      \
+                         grid, dim3(G8_THREADS), 0, stream, a, b, bias,     \
+                         mask, mask_out, cp, M, N, K);                      \
+  } while (0)
   if (epi == G8_EPI_BIAS_RELU && emit_mask) {
     if (has_bias) L8(G8_EPI_BIAS_RELU, true, false, true);
     else          L8(G8_EPI_BIAS_RELU, false, false, true);
