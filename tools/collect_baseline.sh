@@ -9,7 +9,8 @@ mkdir -p "$OUT"
 log() { echo "== $*" | tee -a "$OUT/summary.txt"; }
 
 log "GPU test suite"
-python -m pytest tests -m gpu -q 2>&1 | tail -1 | tee -a "$OUT/summary.txt"
+python -m pytest tests -m gpu -q 2>&1 | grep -E "passed|failed" | tail -1 \
+  | tee -a "$OUT/summary.txt"
 
 log "config 2: linear 10M-row cycle (steady state)"
 timeout 400 python bench.py --rows 10000000 --steps 10 --warmup 3 \
