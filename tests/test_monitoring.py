@@ -92,3 +92,35 @@ def test_analytics_cli(tmp_path):
     assert out.returncode == 0, out.stderr[-1500:]
     assert "summary" in out.stdout
     assert (tmp_path / "joined.csv").exists()
+
+
+def test_canonical_device():
+    """torch.device('cuda') != torch.device('cuda:0'); the helper pins
+    the index so device-tensor cache checks can't silently miss (the
+    round-1 serving-capture bug)."""
+    import torch
+
+    from bodywork_mlops_demo_amd.utils.device import canonical_device
+
+    assert canonical_device("cpu") == torch.device("cpu")
+    d = canonical_device("cuda")
+    assert d.type == "cuda" and d.index is not None
+    assert canonical_device("cuda:0") == torch.device("cuda", 0)
+    assert canonical_device(torch.device("cuda")) == d
+
+
+def test_scorer_bucket_selection():
+    from bodywork_mlops_demo_amd.models import GPULinearRegressor
+    from bodywork_mlops_demo_amd.serving.scorer import BatchedScorer
+
+    s = BatchedScorer(GPULinearRegressor(1.0, 0.5), "cpu")
+    assert s._bucket(1) == 1
+    assert s._bucket(2) == 16
+    assert s._bucket(4096) == 4096
+    assert s._bucket(4097) == 65536
+    # models advertising a chunk bound cap the bucket list
+    class Capped(GPULinearRegressor):
+        PREDICT_CHUNK = 1 << 20
+
+    c = BatchedScorer(Capped(1.0, 0.5), "cpu")
+    assert c.buckets[-1] == 1 << 20
