@@ -207,3 +207,53 @@ def test_drift_policy_skips_retrains(tmp_store):
     assert n_models == sum(trained)
     # every day still produced online metrics
     assert len(tmp_store.list_keys(contract.TEST_METRICS_PREFIX)) == 8
+
+
+@pytest.mark.timeout(420)
+def test_runner_repeat_reference_dag_order(tmp_path, monkeypatch):
+    """With the reference's train-FIRST DAG order (bodywork.yaml:5) each
+    run trains on data through *yesterday* (today's dataset is generated
+    after training, reference stage_3 date-stamping) and the model is
+    keyed by its newest training data — so with a bootstrap dataset
+    dated the day BEFORE the first run, N runs yield N distinct models,
+    each lagging its run date by one day, exactly like the reference's
+    daily cron."""
+    from datetime import date as date_t
+
+    import bodywork_mlops_demo_amd.pipeline.runner as runner_mod
+    from bodywork_mlops_demo_amd.stages import datagen as datagen_mod
+    from bodywork_mlops_demo_amd.store import LocalStore
+    from bodywork_mlops_demo_amd.utils.clock import CLOCK
+
+    store_dir = str(tmp_path / "store")
+    datagen_mod.run(LocalStore(store_dir), n=200,
+                    date=date_t(2026, 6, 30), device="cpu")  # day BEFORE
+    cfg_path = tmp_path / "p.yaml"
+    cfg_path.write_text("""
+version: "1.0"
+project:
+  name: cron-ref-order
+  DAG: stage-1-train-model >> stage-3-generate-next-dataset
+stages:
+  stage-1-train-model:
+    executable_module_path: bodywork_mlops_demo_amd/stages/train.py
+    batch: {max_completion_time_seconds: 120, retries: 1}
+  stage-3-generate-next-dataset:
+    executable_module_path: bodywork_mlops_demo_amd/stages/datagen.py
+    args: ["--n", "200"]
+    batch: {max_completion_time_seconds: 120, retries: 1}
+""")
+    monkeypatch.setenv("BODYWORK_AMD_DATE", "2026-07-01")
+    CLOCK.set("2026-07-01")
+    with pytest.raises(SystemExit) as e:
+        runner_mod.main([str(cfg_path), "--store", store_dir,
+                         "--repeat", "3"])
+    assert e.value.code == 0
+    store = LocalStore(store_dir)
+    models = store.list_keys(contract.MODELS_PREFIX)
+    assert len(models) == 3, models  # one distinct model per run
+    for d in ("2026-06-30", "2026-07-01", "2026-07-02"):
+        assert any(d in k for k in models), (d, models)
+    # datasets: bootstrap + one per run day
+    ds = store.list_keys(contract.DATASETS_PREFIX)
+    assert len(ds) == 4, ds
