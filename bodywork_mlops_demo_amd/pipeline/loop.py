@@ -104,7 +104,8 @@ def run_loop(
 
             flag = torch.tensor(
                 [1.0 if drifted else 0.0],
-                device=device if torch.cuda.is_available() else "cpu",
+                device=device
+                if dist.get_backend(process_group) == "nccl" else "cpu",
             )
             dist.all_reduce(flag, op=dist.ReduceOp.MAX, group=process_group)
             drifted = bool(flag.item() > 0)
