@@ -124,3 +124,44 @@ def test_scorer_bucket_selection():
 
     c = BatchedScorer(Capped(1.0, 0.5), "cpu")
     assert c.buckets[-1] == 1 << 20
+
+
+def test_request_tracer_histogram_and_reentrancy():
+    """snapshot() holds the tracer lock and calls percentile() (which
+    re-acquires it) — the RLock regression; plus histogram math."""
+    import threading
+
+    from bodywork_mlops_demo_amd.monitoring.tracing import RequestTracer
+
+    t = RequestTracer(name="test_tracer")
+    for lat in (1e-4, 1e-4, 1e-3, 1e-2):
+        t.observe(lat, rows=10)
+    snap = t.snapshot()  # must not deadlock
+    assert snap["requests"] == 4
+    assert snap["rows_scored"] == 40
+    assert snap["p50_s"] <= snap["p99_s"]
+    assert 5e-5 < snap["mean_latency_s"] < 5e-2
+
+    # concurrent observers + snapshotters don't deadlock or corrupt
+    def worker():
+        for _ in range(200):
+            t.observe(2e-4, rows=1)
+            t.snapshot()
+
+    threads = [threading.Thread(target=worker) for _ in range(4)]
+    for th in threads:
+        th.start()
+    for th in threads:
+        th.join()
+    assert t.snapshot()["requests"] == 4 + 4 * 200
+
+
+def test_loadtest_url_resolution():
+    """Replica discovery: explicit list, comma-separated env value (the
+    runner's stand-in for the k8s ClusterIP), or a single URL."""
+    from bodywork_mlops_demo_amd.stages.loadtest import _urls
+
+    assert _urls("http://a:5000/score/v1") == ["http://a:5000/score/v1"]
+    assert _urls("http://a:5000/x, http://b:5001/x") == [
+        "http://a:5000/x", "http://b:5001/x"]
+    assert _urls(["u1", "u2"]) == ["u1", "u2"]
