@@ -257,3 +257,45 @@ stages:
     # datasets: bootstrap + one per run day
     ds = store.list_keys(contract.DATASETS_PREFIX)
     assert len(ds) == 4, ds
+
+
+def test_cyclestate_buffer_growth_and_window():
+    """Geometric capacity growth (the round-1 allocator-fragmentation
+    fix) and the history_days sliding window must preserve row order."""
+    import torch
+
+    from datetime import date as date_t
+
+    s = CycleState("cpu", date_t(2026, 1, 1))
+    for day in range(5):
+        n = 10 * (day + 1)
+        s.append_day(torch.full((n,), float(day)),
+                     torch.full((n,), float(day) + 0.5))
+    assert s.y.numel() == 10 + 20 + 30 + 40 + 50
+    # geometric growth: capacity at least the data size, amortised doubling
+    assert s._buf_y.numel() >= s.y.numel()
+    # values preserved in day order
+    assert float(s.y[0]) == 0.0 and float(s.y[-1]) == 4.0
+    assert float(s.X[0]) == 0.5 and float(s.X[-1]) == 4.5
+
+    w = CycleState("cpu", date_t(2026, 1, 1), history_days=2)
+    for day in range(6):
+        w.append_day(torch.full((8,), float(day)), torch.full((8,), 0.0))
+        assert w.y.numel() == min(8 * (day + 1), 16)
+    # window holds exactly the last two days, oldest first
+    assert float(w.y[0]) == 4.0 and float(w.y[-1]) == 5.0
+
+
+def test_cyclestate_persist_rank_rotation():
+    from datetime import date as date_t
+
+    ranks = []
+    for world in (1, 4):
+        s = CycleState("cpu", date_t(2026, 1, 1), rank=0, world_size=world)
+        got = []
+        for _ in range(6):
+            got.append(s.persist_rank)
+            s.cycle_count += 1
+        ranks.append(got)
+    assert ranks[0] == [0] * 6                 # world 1: always rank 0
+    assert ranks[1] == [0, 1, 2, 3, 0, 1]      # world 4: rotates per cycle
