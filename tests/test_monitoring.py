@@ -165,3 +165,33 @@ def test_loadtest_url_resolution():
     assert _urls("http://a:5000/x, http://b:5001/x") == [
         "http://a:5000/x", "http://b:5001/x"]
     assert _urls(["u1", "u2"]) == ["u1", "u2"]
+
+
+def test_notebooks_parse_and_imports_resolve():
+    """The 5 notebook mirrors (reference C9) must stay in sync with the
+    library API: every code cell parses and every package import
+    resolves."""
+    import ast
+    import importlib
+    import json
+    import pathlib
+
+    nbs = sorted(pathlib.Path("notebooks").glob("*.ipynb"))
+    assert len(nbs) == 5, [n.name for n in nbs]
+    for nb in nbs:
+        cells = json.loads(nb.read_text())["cells"]
+        code = "\n\n".join("".join(c["source"]) for c in cells
+                           if c["cell_type"] == "code")
+        code = "\n".join(l for l in code.splitlines()
+                         if not l.strip().startswith(("%", "!")))
+        tree = ast.parse(code)  # raises on drift
+        for n in ast.walk(tree):
+            mods = []
+            if isinstance(n, ast.ImportFrom) and n.module and \
+                    n.module.startswith("bodywork"):
+                mods = [n.module]
+            elif isinstance(n, ast.Import):
+                mods = [a.name for a in n.names
+                        if a.name.startswith("bodywork")]
+            for m in mods:
+                importlib.import_module(m)
