@@ -299,3 +299,23 @@ def test_cyclestate_persist_rank_rotation():
         ranks.append(got)
     assert ranks[0] == [0] * 6                 # world 1: always rank 0
     assert ranks[1] == [0, 1, 2, 3, 0, 1]      # world 4: rotates per cycle
+
+
+def test_check_requirements_validation():
+    """Per-stage pinned envs (reference bodywork.yaml:10-16) are
+    validated against the single runtime environment: missing packages
+    and exact-pin mismatches are reported, satisfied pins are not."""
+    from bodywork_mlops_demo_amd.config.schema import StageSpec
+
+    spec = StageSpec(
+        name="s", executable_module_path="x.py",
+        requirements=["numpy", "numpy==0.0.1", "definitely-not-a-pkg==1.0",
+                      "pytest>=1.0"],
+    )
+    problems = PipelineRunner.check_requirements(spec)
+    assert any("definitely-not-a-pkg" in p and "not installed" in p
+               for p in problems)
+    assert any("numpy==0.0.1" in p and "installed" in p for p in problems)
+    # plain name present + >= pin: no complaints
+    assert not any(p.startswith("numpy:") for p in problems)
+    assert not any("pytest" in p for p in problems)
