@@ -205,18 +205,19 @@ class GPUMLPRegressor:
     def reinit_(self, seed: int = 7) -> "GPUMLPRegressor":
         """Re-randomise weights and reset optimiser state IN PLACE, so a
         cached training graph (whose nodes hold these tensor pointers)
-        stays valid for the next day's fresh fit."""
-        import math as _math
+        stays valid for the next day's fresh fit.
 
-        g = torch.Generator(device=self.device) if self.device.type == "cuda" \
-            else torch.Generator()
-        g.manual_seed(seed)
+        Draws come from the same CPU generator stream, in the same order
+        and pattern, as ``__init__`` — so ``reinit_(s)`` on a warm cache
+        and a cold ``GPUMLPRegressor(seed=s)`` start from bit-identical
+        weights, keeping results reproducible across resume boundaries."""
+        g = torch.Generator(device="cpu").manual_seed(seed)
         h = self.hidden
-        self.w1.normal_(0.0, _math.sqrt(2.0), generator=g)
+        self.w1.copy_(torch.randn(h, generator=g) * math.sqrt(2.0))
         self.b1.zero_()
-        self.W2.normal_(0.0, _math.sqrt(2.0 / h), generator=g)
+        self.W2.copy_(torch.randn(h, h, generator=g) * math.sqrt(2.0 / h))
         self.b2.zero_()
-        self.w3.normal_(0.0, _math.sqrt(2.0 / h), generator=g)
+        self.w3.copy_(torch.randn(h, generator=g) * math.sqrt(2.0 / h))
         self.b3.zero_()
         if self._opt_state is not None:
             for m, v in self._opt_state:

@@ -46,10 +46,23 @@ class ServiceHandle:
     ports: list[int]
     cmds: list[list[str]] = field(default_factory=list)
     envs: list[dict] = field(default_factory=list)
-    respawns: int = 0
+    # per-replica respawn counters: one crash-looping replica must not
+    # exhaust its siblings' recovery budget
+    respawns: list[int] = field(default_factory=list)
+    proxy: object | None = None  # FrontProxy when enabled
+
+    def __post_init__(self):
+        if not self.respawns:
+            self.respawns = [0] * len(self.procs)
 
     @property
     def urls(self) -> list[str]:
+        if self.proxy is not None:
+            return [f"http://127.0.0.1:{self.proxy.port}/score/v1"]
+        return [f"http://127.0.0.1:{p}/score/v1" for p in self.ports]
+
+    @property
+    def replica_urls(self) -> list[str]:
         return [f"http://127.0.0.1:{p}/score/v1" for p in self.ports]
 
 
@@ -73,6 +86,7 @@ class PipelineRunner:
         secrets_file: str | None = None,
         base_port: int = 5000,
         n_gpus: int | None = None,
+        front_proxy: bool = True,
     ):
         self.config = (
             config if isinstance(config, PipelineConfig) else load_config(config)
@@ -82,6 +96,7 @@ class PipelineRunner:
             "BODYWORK_AMD_STORE", "./artefact-store"
         )
         self.base_port = base_port
+        self.front_proxy = front_proxy
         self.services: dict[str, ServiceHandle] = {}
         self._secrets = self._load_secrets(secrets_file)
         if n_gpus is None:
@@ -109,8 +124,15 @@ class PipelineRunner:
         # stage 2 via k8s cluster DNS in the reference (stage_4:28)
         urls = [u for h in self.services.values() for u in h.urls]
         if urls:
+            # SERVICE_URL is the ONE stable endpoint (the front proxy when
+            # enabled — the k8s ClusterIP equivalent of stage_4:28);
+            # SERVICE_URLS exposes the replica set for clients that drive
+            # the shared-nothing GPU fan-out directly
             env["BODYWORK_AMD_SERVICE_URL"] = urls[0]
-            env["BODYWORK_AMD_SERVICE_URLS"] = ",".join(urls)
+            replica_urls = [
+                u for h in self.services.values() for u in h.replica_urls
+            ]
+            env["BODYWORK_AMD_SERVICE_URLS"] = ",".join(replica_urls)
         for var, group in spec.secrets.items():
             if var in os.environ:
                 continue
@@ -198,14 +220,60 @@ class PipelineRunner:
         )
         return False
 
+    def _redeploy_service(self, handle: ServiceHandle, timeout: float) -> bool:
+        """Hot-redeploy an already-running service: ask every live replica
+        to reload the latest model artefact (``POST /reload/v1``).  Returns
+        False if any replica is dead or refuses, in which case the caller
+        falls back to stop + restart."""
+        import requests
+
+        if any(p.poll() is not None for p in handle.procs):
+            return False
+        for port in handle.ports:
+            try:
+                r = requests.post(
+                    f"http://127.0.0.1:{port}/reload/v1", timeout=timeout
+                )
+                if not (r.ok and r.json().get("status") == "ok"):
+                    return False
+            except Exception:
+                return False
+        log.info(f"service {handle.stage}: hot-reloaded model on "
+                 f"{len(handle.ports)} replica(s)")
+        return True
+
     def _start_service_stage(self, spec: StageSpec, report: RunReport) -> bool:
         assert spec.service is not None
         svc = spec.service
+        # A handle may already exist (--repeat with --keep-services).
+        # Starting new replicas on the same ports would fail to bind while
+        # the health probe still passes against the OLD processes — the new
+        # procs die, the old ones leak and keep serving the STALE model.
+        # Instead: hot-reload the live replicas (picks up the newly trained
+        # artefact), or stop them before respawning.
+        existing = self.services.get(spec.name)
+        if existing is not None:
+            if self._redeploy_service(existing,
+                                      svc.max_startup_time_seconds):
+                return True
+            log.warning(f"service {spec.name}: live redeploy failed; "
+                        "restarting replicas")
+            self._stop_service(existing)
+            del self.services[spec.name]
         env_base = self._stage_env(spec)
         procs, ports = [], []
         cmds, envs = [], []
+        # with the front proxy the declared service.port is the consumer-
+        # facing endpoint (ClusterIP parity, bodywork.yaml:41) and the
+        # replicas bind behind it on port+1..port+replicas
+        use_proxy = self.front_proxy and svc.replicas > 1
         for r in range(svc.replicas):
-            port = self.base_port + r if svc.replicas > 1 else svc.port
+            if use_proxy:
+                port = svc.port + 1 + r
+            elif svc.replicas > 1:
+                port = self.base_port + r
+            else:
+                port = svc.port
             env = dict(env_base)
             env["PORT"] = str(port)
             if self.n_gpus > 0:
@@ -225,6 +293,10 @@ class PipelineRunner:
                 f"service {spec.name} failed startup probe", "error"
             )
             return False
+        if use_proxy:
+            from bodywork_mlops_demo_amd.pipeline.proxy import FrontProxy
+
+            handle.proxy = FrontProxy(ports, port=svc.port).start()
         self.services[spec.name] = handle
         return True
 
@@ -264,7 +336,9 @@ class PipelineRunner:
             for i, proc in enumerate(handle.procs):
                 if proc.poll() is None:
                     continue
-                if handle.respawns >= max_respawns_per_replica * len(handle.procs):
+                # budget is PER REPLICA: a crash-looping replica exhausts
+                # only its own counter, never its siblings' recovery
+                if handle.respawns[i] >= max_respawns_per_replica:
                     log.error(
                         f"service {handle.stage} replica {i} dead "
                         f"(rc={proc.returncode}) and respawn budget spent"
@@ -279,7 +353,7 @@ class PipelineRunner:
                 )
                 handle.procs[i] = subprocess.Popen(handle.cmds[i],
                                                    env=handle.envs[i])
-                handle.respawns += 1
+                handle.respawns[i] += 1
                 respawned += 1
         return respawned
 
@@ -295,6 +369,9 @@ class PipelineRunner:
 
     @staticmethod
     def _stop_service(handle: ServiceHandle) -> None:
+        if handle.proxy is not None:
+            handle.proxy.stop()
+            handle.proxy = None
         for p in handle.procs:
             if p.poll() is None:
                 p.terminate()

@@ -110,6 +110,32 @@ def create_app(store: ArtefactStore, device: str = "cpu",
         data, content_type = prom
         return Response(content=data, media_type=content_type)
 
+    @app.post("/reload/v1")
+    async def reload_model() -> Response:
+        """Hot-redeploy: re-read the latest model artefact from the store
+        and swap it into the resident scorer.  Weights are copied into the
+        captured graphs' tensors when shapes match (no recapture); an
+        incompatible model gets a fresh scorer.  The pipeline runner calls
+        this on every replica after each retrain when the service is kept
+        running across pipeline repeats (k8s rolling-redeploy parity)."""
+        try:
+            artefact, model_date = store.get_latest_model()
+            model = regressor_from_artifact(artefact, device=device)
+        except Exception as e:  # no/unreadable artefact
+            return JSONResponse({"status": "error", "error": str(e)},
+                                status_code=500)
+        scorer = state.get("scorer")
+        if scorer is None or not scorer.update_model(model):
+            state["scorer"] = BatchedScorer(model, device,
+                                            use_graphs=use_graphs)
+        state["model_info"] = str(model)
+        state["model_date"] = str(model_date)
+        log.info(f"reloaded model={state['model_info']} trained on "
+                 f"{model_date}")
+        return JSONResponse({"status": "ok",
+                             "model_date": state["model_date"],
+                             "model_info": state["model_info"]})
+
     @app.get("/healthz")
     async def healthz() -> Response:
         ok = "scorer" in state

@@ -86,15 +86,20 @@ def run(
         )
     elif model_type == "mlp":
         # reuse the cached model object so its captured training graph
-        # survives across daily retrains (weights re-randomised in place)
+        # survives across daily retrains (weights re-randomised in place).
+        # The same date-derived seed feeds both the warm (reinit_) and
+        # cold (constructor) paths, and both draw the same CPU generator
+        # stream — the day's initial weights are identical regardless of
+        # cache warmth or process restarts.
+        init_seed = 7 + data_date.toordinal()
         model = None
         if model_cache is not None:
             cached = model_cache.get("mlp")
             if (isinstance(cached, GPUMLPRegressor)
                     and str(cached.device) == str(torch.device(device))):
-                model = cached.reinit_(seed=7 + data_date.toordinal())
+                model = cached.reinit_(seed=init_seed)
         if model is None:
-            model = GPUMLPRegressor(device=device)
+            model = GPUMLPRegressor(device=device, seed=init_seed)
             if model_cache is not None:
                 model_cache["mlp"] = model
         model.fit(

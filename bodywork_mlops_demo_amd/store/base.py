@@ -123,6 +123,17 @@ class ArtefactStore(ABC):
         pairs = self.all_by_date(contract.DATASETS_PREFIX)
         if not pairs:
             raise FileNotFoundError("no datasets in store")
+        # one artefact per date: a day persisted in both formats (e.g. a
+        # csv run resumed with persist_fmt='npy') must not be loaded twice
+        # — prefer the binary record, which is the hot-path canonical form
+        by_date: dict = {}
+        for key, d in pairs:
+            cur = by_date.get(d)
+            if cur is None or (cur.endswith(".csv")
+                               and not key.endswith(".csv")):
+                by_date[d] = key
+        pairs = sorted(((k, d) for d, k in by_date.items()),
+                       key=lambda p: p[1])
         ys, Xs = [], []
         for key, _ in pairs:
             y, X = self.get_dataset(key)

@@ -21,7 +21,10 @@ class LocalStore(ArtefactStore):
 
     def _path(self, key: str) -> str:
         path = os.path.normpath(os.path.join(self.root, key))
-        if not path.startswith(self.root):
+        # separator-aware root check: a plain prefix test lets a sibling
+        # dir sharing the root as a string prefix through
+        # (root='/x/store', key '../store2/f' -> '/x/store2/f')
+        if path != self.root and not path.startswith(self.root + os.sep):
             raise ValueError(f"key escapes store root: {key!r}")
         return path
 
