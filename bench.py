@@ -45,7 +45,22 @@ def main() -> None:
     p.add_argument("--no-graphs", action="store_true")
     p.add_argument("--store", default=None,
                    help="artefact store dir (default: fresh tmpdir)")
+    p.add_argument("--serving", default="inprocess",
+                   choices=["inprocess", "http"],
+                   help="http = deploy a real uvicorn replica per rank and "
+                        "drive stage 4 over the wire (mean_response_time "
+                        "keeps its reference meaning, stage_4:105,111); "
+                        "inprocess = resident BatchedScorer, no HTTP in "
+                        "the timed region")
+    p.add_argument("--http-mode", default="binary",
+                   choices=["binary", "batch", "serial"],
+                   help="wire format for --serving http")
+    p.add_argument("--history", default="1",
+                   help="training window in days, or 'all' for the "
+                        "reference's read-all-accumulated-data semantics "
+                        "(stage_1:59-71; history grows by --rows each step)")
     args = p.parse_args()
+    history_days = None if args.history == "all" else int(args.history)
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
@@ -96,7 +111,7 @@ def main() -> None:
     store = LocalStore(store_dir)
 
     state = CycleState(device, date(2026, 1, 1), rank=rank, world_size=world,
-                       history_days=1)
+                       history_days=history_days)
     scorer_cache: dict = {}
 
     def one_cycle():
@@ -106,6 +121,7 @@ def main() -> None:
             mlp_steps=args.mlp_steps, mlp_batch_size=args.mlp_batch,
             use_graphs=not args.no_graphs and use_cuda,
             scorer_cache=scorer_cache,
+            serving=args.serving, http_mode=args.http_mode,
         )
 
     def barrier_sync():
@@ -140,12 +156,23 @@ def main() -> None:
         torch.distributed.all_reduce(rbuf, group=pg)
         rows_scored = int(rbuf.item())
 
+    replica = scorer_cache.get("http_replica")
+    if replica is not None:
+        replica.stop()
+
     if rank == 0:
         ms_per_step = elapsed / args.steps * 1000.0
         value = rows_scored / elapsed
+        serving_desc = (
+            f"1 uvicorn replica/GPU, {args.http_mode} wire over HTTP"
+            if args.serving == "http"
+            else "1 in-process replica/GPU, hipGraph-captured batches"
+        )
         print(json.dumps({
             "metric": "rows/sec scored (stage_2 path, full "
-                      "train-serve-drift-test cycle)",
+                      "train-serve-drift-test cycle"
+                      + (", over-the-wire HTTP serving)"
+                         if args.serving == "http" else ")"),
             "value": value,
             "unit": "rows/s",
             "n_gpus": args.gpus,
@@ -157,14 +184,16 @@ def main() -> None:
             "vs_baseline": None,
             "dtype": "bf16" if args.model == "mlp" else "fp32",
             "data": "synthetic (on-GPU philox drift generator, "
-                    f"{args.rows} rows/GPU/day, random-init weights)",
+                    f"{args.rows} rows/GPU/day, random-init weights"
+                    + (", growing read-all training history"
+                       if history_days is None else "") + ")",
             "config": {
                 "model": {"linear": "linear-ols",
                           "mlp": "mlp-4096x2"}.get(args.model, args.model),
                 "rows_per_gpu_per_day": args.rows,
-                "history_days": 1,
+                "history_days": args.history,
                 "parallelism": f"dp{args.gpus}",
-                "serving": "1 replica/GPU, hipGraph-captured batches",
+                "serving": serving_desc,
                 "train_phase": ("adam bf16 MFMA GEMM, "
                                 f"{args.mlp_steps}x{args.mlp_batch}"
                                 if args.model == "mlp"

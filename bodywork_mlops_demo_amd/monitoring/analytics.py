@@ -31,7 +31,7 @@ def download_metrics(store: ArtefactStore, prefix: str) -> pd.DataFrame:
         return pd.DataFrame()
     df = pd.DataFrame(rows)
     for col in df.columns:
-        if col != "date":
+        if col not in ("date", "response_time_kind"):
             df[col] = pd.to_numeric(df[col], errors="coerce")
     df["date"] = pd.to_datetime(df["date"])
     return df.sort_values("date").reset_index(drop=True)

@@ -13,6 +13,7 @@ This is what ``bench.py`` times: BASELINE.json's "rows/sec scored
 """
 from __future__ import annotations
 
+import os
 from datetime import date as date_t, timedelta
 from time import perf_counter
 
@@ -26,6 +27,70 @@ from bodywork_mlops_demo_amd.store import ArtefactStore
 from bodywork_mlops_demo_amd.utils.logging import configure_logger
 
 log = configure_logger(__name__)
+
+
+class HttpServingReplica:
+    """A real uvicorn serving replica owned by a bench/loop rank, so the
+    timed cycle crosses the HTTP boundary (the reference's only
+    self-metric, ``mean_response_time``, is over-the-wire —
+    ``stage_4_test_model_scoring_service.py:105,111``).  The process is
+    started once and hot-reloaded (``POST /reload/v1``) on every
+    redeploy, mirroring what the pipeline runner does for kept services.
+    """
+
+    def __init__(self, store_uri: str, port: int, device_index: int | None,
+                 use_graphs: bool = True):
+        import subprocess
+        import sys
+
+        self.port = port
+        self.url = f"http://127.0.0.1:{port}/score/v1"
+        env = dict(os.environ)
+        if device_index is not None:
+            env["HIP_VISIBLE_DEVICES"] = str(device_index)
+            env["CUDA_VISIBLE_DEVICES"] = str(device_index)
+        cmd = [sys.executable, "-m", "bodywork_mlops_demo_amd.stages.serve",
+               "--store", store_uri, "--host", "127.0.0.1",
+               "--port", str(port)]
+        if not use_graphs:
+            cmd.append("--no-graphs")
+        self.proc = subprocess.Popen(cmd, env=env)
+
+    def await_healthy(self, timeout: float = 240.0) -> None:
+        import requests
+
+        deadline = perf_counter() + timeout
+        while perf_counter() < deadline:
+            if self.proc.poll() is not None:
+                raise RuntimeError(
+                    f"serving replica exited rc={self.proc.returncode}")
+            try:
+                r = requests.get(
+                    f"http://127.0.0.1:{self.port}/healthz", timeout=2)
+                if r.ok and r.json().get("status") == "ok":
+                    return
+            except Exception:
+                pass
+            import time as _time
+
+            _time.sleep(0.2)
+        raise RuntimeError("serving replica failed startup probe")
+
+    def reload(self) -> None:
+        import requests
+
+        r = requests.post(f"http://127.0.0.1:{self.port}/reload/v1",
+                          timeout=120)
+        r.raise_for_status()
+
+    def stop(self) -> None:
+        if self.proc.poll() is None:
+            self.proc.terminate()
+            try:
+                self.proc.wait(timeout=10)
+            except Exception:
+                self.proc.kill()
+                self.proc.wait(timeout=10)
 
 
 class CycleState:
@@ -181,6 +246,9 @@ def run_cycle(
     use_graphs: bool = True,
     scorer_cache: dict | None = None,
     skip_train: bool = False,
+    serving: str = "inprocess",
+    http_mode: str = "binary",
+    http_port: int = 5600,
 ) -> dict:
     """Run one full cycle; returns per-phase timings + metrics.
 
@@ -190,11 +258,20 @@ def run_cycle(
     ``skip_train`` keeps the currently deployed model (drift-policy
     loops retrain only when the live metrics degrade); it requires a
     scorer already resident in ``scorer_cache``.
+
+    ``serving="http"`` deploys a REAL uvicorn replica per rank (pinned to
+    the rank's GPU) and stage 4 drives the wire (``http_mode``: binary |
+    batch | serial), so ``mean_response_time`` keeps its reference
+    meaning; ``"inprocess"`` scores through the resident BatchedScorer
+    (no HTTP inside the timed region — the faster hermetic path, whose
+    per-row time is persisted as kind ``amortised-gpu-batch``).
     """
     device = state.device
     dev_cuda = device.startswith("cuda")
     timings: dict[str, float] = {}
-    if skip_train and not (scorer_cache and scorer_cache.get("scorer")):
+    deployed = scorer_cache and (scorer_cache.get("scorer")
+                                 or scorer_cache.get("http_replica"))
+    if skip_train and not deployed:
         skip_train = False  # nothing deployed yet -> must train
 
     def sync():
@@ -220,6 +297,11 @@ def run_cycle(
         timings["train_s"] = 0.0
         timings["deploy_s"] = 0.0
         metrics = scorer_cache.get("offline_metrics", {})
+        if serving == "http":
+            replica = scorer_cache["http_replica"]
+            return _finish_cycle(state, store, n_rows, persist_fmt, None,
+                                 timings, metrics, sync,
+                                 http=(replica.url, http_mode))
         scorer = scorer_cache["scorer"]
         return _finish_cycle(state, store, n_rows, persist_fmt, scorer,
                              timings, metrics, sync)
@@ -242,6 +324,37 @@ def run_cycle(
     # -- stage 2: deploy — joblib artefact round-trip, model into HBM,
     #    hipGraphs captured (reference stage_2:108-119 semantics) ----------
     t0 = perf_counter()
+    if serving == "http":
+        # real serving replica on this rank's GPU; model redeploys by
+        # hot-reloading the artefact rank 0 just persisted
+        if store is None:
+            raise ValueError("serving='http' requires a real artefact store")
+        if process_group is not None:  # rank 0 persisted; others wait
+            import torch.distributed as dist
+
+            dist.barrier(group=process_group)
+        replica = (scorer_cache or {}).get("http_replica")
+        if replica is None:
+            dev_idx = (int(device.split(":")[1]) if ":" in device else 0) \
+                if dev_cuda else None
+            replica = HttpServingReplica(
+                store.uri, http_port + state.rank, dev_idx,
+                use_graphs=use_graphs and dev_cuda)
+            try:
+                replica.await_healthy()
+            except Exception:
+                replica.stop()
+                raise
+            if scorer_cache is not None:
+                scorer_cache["http_replica"] = replica
+        else:
+            replica.reload()
+        if scorer_cache is not None:
+            scorer_cache["offline_metrics"] = metrics
+        timings["deploy_s"] = perf_counter() - t0
+        return _finish_cycle(state, store, n_rows, persist_fmt, None,
+                             timings, metrics, sync,
+                             http=(replica.url, http_mode))
     if store is not None:
         if process_group is not None:  # rank 0 persisted; others wait
             import torch.distributed as dist
@@ -282,8 +395,11 @@ def run_cycle(
 
 
 def _finish_cycle(state, store, n_rows, persist_fmt, scorer, timings,
-                  metrics, sync):
-    """Stages 3+4 and clock advance (shared by train and skip-train paths)."""
+                  metrics, sync, http: tuple[str, str] | None = None):
+    """Stages 3+4 and clock advance (shared by train and skip-train paths).
+
+    ``http=(url, mode)`` routes stage 4 over the wire to a live replica
+    instead of the in-process scorer."""
     device = state.device
 
     # -- stage 3: generate day t+1 ------------------------------------------
@@ -301,13 +417,22 @@ def _finish_cycle(state, store, n_rows, persist_fmt, scorer, timings,
 
     # -- stage 4: test the deployed model on unseen t+1 data ----------------
     t0 = perf_counter()
-    test_metrics = stage4.run(
-        store if store is not None else _NullStore(),
-        device=device,
-        scorer=scorer,
-        data=(y_next, X_next, next_date),
-        persist=store is not None and state.rank == 0,
-    )
+    if http is not None:
+        url, mode = http
+        test_metrics = stage4.run(
+            store if store is not None else _NullStore(),
+            url=url, mode=mode, device=device,
+            data=(y_next, X_next, next_date),
+            persist=store is not None and state.rank == 0,
+        )
+    else:
+        test_metrics = stage4.run(
+            store if store is not None else _NullStore(),
+            device=device,
+            scorer=scorer,
+            data=(y_next, X_next, next_date),
+            persist=store is not None and state.rank == 0,
+        )
     sync()
     timings["test_s"] = perf_counter() - t0
     timings["rows_scored"] = int(y_next.shape[0])

@@ -81,5 +81,11 @@ class LocalStore(ArtefactStore):
     def delete(self, key: str) -> None:
         os.unlink(self._path(key))
 
+    @property
+    def uri(self) -> str:
+        """URI that re-opens this store (``open_store(uri)``) — e.g. for
+        handing to a serving-replica subprocess."""
+        return self.root
+
     def __repr__(self) -> str:
         return f"LocalStore({self.root!r})"

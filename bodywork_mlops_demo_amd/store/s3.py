@@ -23,6 +23,10 @@ class S3Store(ArtefactStore):
         self.bucket = bucket
         self._s3 = boto3.client("s3")
 
+    @property
+    def uri(self) -> str:
+        return f"s3://{self.bucket}"
+
     def list_keys(self, prefix: str) -> list[str]:
         keys: list[str] = []
         token = None

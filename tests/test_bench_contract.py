@@ -44,6 +44,47 @@ def test_bench_single_rank_cpu(tmp_path):
 
 
 @pytest.mark.timeout(600)
+def test_bench_http_serving_cpu(tmp_path):
+    """--serving http: stage 2 is a real uvicorn replica and stage 4
+    drives the wire, so mean_response_time keeps its reference meaning
+    (stage_4:105,111)."""
+    proc = subprocess.run(
+        [sys.executable, "bench.py", "--rows", "5000", "--steps", "1",
+         "--warmup", "1", "--serving", "http",
+         "--store", str(tmp_path / "store")],
+        cwd=REPO, capture_output=True, text=True, timeout=540,
+    )
+    assert proc.returncode == 0, proc.stderr[-2000:]
+    d = _parse_json_line(proc.stdout)
+    assert "HTTP" in d["metric"]
+    assert "uvicorn" in d["config"]["serving"]
+    assert d["value"] > 0
+    # persisted test metrics carry the over-the-wire kind
+    import glob
+
+    tm = glob.glob(str(tmp_path / "store" / "test-metrics" / "*.csv"))
+    assert tm
+    text = open(tm[0]).read()
+    assert "response_time_kind" in text and "http-binary" in text
+
+
+@pytest.mark.timeout(600)
+def test_bench_history_all_cpu(tmp_path):
+    """--history all: the reference's read-all-accumulated-data training
+    semantics (stage_1:59-71) — the training set grows every step."""
+    proc = subprocess.run(
+        [sys.executable, "bench.py", "--rows", "5000", "--steps", "2",
+         "--warmup", "1", "--history", "all",
+         "--store", str(tmp_path / "store")],
+        cwd=REPO, capture_output=True, text=True, timeout=540,
+    )
+    assert proc.returncode == 0, proc.stderr[-2000:]
+    d = _parse_json_line(proc.stdout)
+    assert d["config"]["history_days"] == "all"
+    assert "read-all" in d["data"]
+
+
+@pytest.mark.timeout(600)
 def test_bench_torchrun_world2_cpu(tmp_path):
     env = dict(os.environ, MASTER_ADDR="127.0.0.1")
     proc = subprocess.run(
