@@ -69,7 +69,10 @@ def main() -> None:
         args.gpus = world
 
     use_cuda = torch.cuda.is_available()
-    device = f"cuda:{local_rank}" if use_cuda else "cpu"
+    # modulo mapping lets world > device_count oversubscribe (e.g. the
+    # world=2-on-one-GPU RCCL rehearsal for the 8-GPU scale run)
+    dev_idx = local_rank % torch.cuda.device_count() if use_cuda else 0
+    device = f"cuda:{dev_idx}" if use_cuda else "cpu"
 
     if use_cuda:
         # the HIP extension is the compute path — fail loudly if missing
@@ -81,7 +84,7 @@ def main() -> None:
                 "build with: PYTORCH_ROCM_ARCH=gfx950 python setup.py "
                 "build_ext --inplace"
             )
-        torch.cuda.set_device(local_rank)
+        torch.cuda.set_device(dev_idx)
 
     pg = None
     if world > 1:

@@ -37,7 +37,9 @@ def init_distributed(backend: str | None = None, timeout_s: float = 300.0):
     if backend is None:
         backend = "nccl" if torch.cuda.is_available() else "gloo"
     if backend == "nccl":
-        torch.cuda.set_device(local_rank)
+        # modulo mapping: world may oversubscribe the visible GPUs (the
+        # world=2-on-one-MI355X RCCL rehearsal runs two ranks on device 0)
+        torch.cuda.set_device(local_rank % torch.cuda.device_count())
     if not dist.is_initialized():
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
         os.environ.setdefault("MASTER_PORT", "29500")

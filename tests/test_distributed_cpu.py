@@ -129,3 +129,26 @@ def test_dp_drift_policy_decision_is_collective():
         results = dict(out)
     assert results[0] == results[1], "ranks diverged on retrain decisions"
     assert results[0][0]  # first day always trains
+
+
+@pytest.mark.timeout(600)
+def test_dist_parity_tool_world2_gloo(tmp_path):
+    """tools/dist_parity.py (the hardware RCCL rehearsal script) on the
+    gloo/CPU path: trained models must be bit-identical across ranks."""
+    import subprocess
+    import sys
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    env = dict(os.environ, MASTER_ADDR="127.0.0.1")
+    proc = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29641", "tools/dist_parity.py",
+         "--model", "linear,mlp", "--rows", "20000", "--cycles", "2",
+         "--mlp-steps", "4", "--mlp-batch", "4096"],
+        cwd=repo, capture_output=True, text=True, timeout=540, env=env,
+    )
+    out = proc.stdout + proc.stderr
+    assert proc.returncode == 0, out[-3000:]
+    assert "DIST_PARITY OK model=linear world=2 max_diff=0.0" in out
+    assert "DIST_PARITY OK model=mlp world=2 max_diff=0.0" in out
