@@ -35,10 +35,14 @@ def init_distributed(backend: str | None = None, timeout_s: float = 300.0):
     rank = int(os.environ.get("RANK", "0"))
     local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
     if backend is None:
-        backend = "nccl" if torch.cuda.is_available() else "gloo"
+        # RCCL refuses two ranks on one device ("Duplicate GPU detected",
+        # ncclInvalidUsage — profiles/r02_world2_rccl.md), so NCCL/RCCL is
+        # only auto-selected when every rank can own a distinct GPU; an
+        # oversubscribed world (the 1-GPU world=2 rehearsal) falls back to
+        # gloo transport with the compute still on the GPU.
+        n_gpu = torch.cuda.device_count() if torch.cuda.is_available() else 0
+        backend = "nccl" if 0 < world <= n_gpu else "gloo"
     if backend == "nccl":
-        # modulo mapping: world may oversubscribe the visible GPUs (the
-        # world=2-on-one-MI355X RCCL rehearsal runs two ranks on device 0)
         torch.cuda.set_device(local_rank % torch.cuda.device_count())
     if not dist.is_initialized():
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")

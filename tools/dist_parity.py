@@ -115,11 +115,16 @@ def main() -> None:
     p.add_argument("--rows", type=int, default=200_000)
     p.add_argument("--mlp-steps", type=int, default=10)
     p.add_argument("--mlp-batch", type=int, default=16384)
+    p.add_argument("--backend", default=None,
+                   help="force nccl|gloo (default: auto — nccl when every "
+                        "rank can own a distinct GPU, else gloo transport "
+                        "with GPU-resident compute; RCCL refuses 2 ranks "
+                        "on 1 device)")
     args = p.parse_args()
 
     from bodywork_mlops_demo_amd.parallel import init_distributed
 
-    rank, world, local_rank = init_distributed()
+    rank, world, local_rank = init_distributed(backend=args.backend)
     if world < 2:
         raise SystemExit("run under torchrun with --nproc-per-node >= 2")
     use_cuda = torch.cuda.is_available()
