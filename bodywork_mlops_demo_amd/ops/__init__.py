@@ -387,6 +387,60 @@ def gemm_tn_bf16(
     return reference.gemm_tn_bf16_cpu(a, b, out_fp32)
 
 
+def e4m3_exponent(amax: float) -> int:
+    """Per-tensor shared exponent for e4m3: smallest e with amax/2^e <= 448
+    (no saturation), clamped to the E8M0 range."""
+    import math
+
+    if not math.isfinite(amax) or amax <= 0.0:
+        return 0
+    e = math.ceil(math.log2(amax / 448.0))
+    while amax / (2.0 ** e) > 448.0:  # guard float fuzz at the boundary
+        e += 1
+    return max(-127, min(127, e))
+
+
+def quantize_e4m3(x: torch.Tensor, e: int) -> torch.Tensor:
+    """x -> OCP e4m3 byte codes with value = 2^e * stored (RNE, saturating).
+
+    GPU path: packed v_cvt_pk_fp8_f32 (ops/hip/gemm_mx8.hip); CPU oracle:
+    nearest-value-on-grid (ops/reference.py).
+    """
+    if x.device.type == "cuda":
+        core = _core(x.device)
+        return core.quantize_e4m3(x.contiguous(), int(e))
+    return reference.quantize_e4m3_cpu(x, int(e))
+
+
+def gemm_mx8_nt(
+    a8: torch.Tensor,
+    ea: int,
+    b8: torch.Tensor,
+    eb: int,
+    bias: torch.Tensor | None = None,
+    relu: bool = False,
+    out_fp32: bool = False,
+) -> torch.Tensor:
+    """C[M,N] = 2^(ea+eb) * (a8[M,K] . b8[N,K]^T) over e4m3 operands —
+    the 2x-rate MX path.
+
+    Uses the gfx950 block-scaled K=128 MFMA
+    (__builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4) with the
+    per-tensor shared exponents passed as uniform E8M0 scale operands, so
+    dequantisation is exact (power-of-two) and free (HW-fused).  The
+    non-scaled fp8 MFMAs run at the BF16 rate on CDNA4 — this instruction
+    is the only 2x fp8 path.  Requires M%256==0, N%256==0, K%128==0.
+    Fused bias+relu epilogue available (the MLP serving forward).
+    CPU oracle: decode + fp32 matmul.
+    """
+    if a8.device.type == "cuda":
+        core = _core(a8.device)
+        return core.gemm_mx8_nt(
+            a8.contiguous(), int(ea), b8.contiguous(), int(eb),
+            bias.contiguous() if bias is not None else None, relu, out_fp32)
+    return reference.gemm_mx8_nt_cpu(a8, ea, b8, eb, bias, relu, out_fp32)
+
+
 def transpose_bf16(src: torch.Tensor) -> torch.Tensor:
     """[C,R] bf16 = [R,C] bf16 transposed (64x64 LDS tiles, 16-B I/O)."""
     if src.device.type == "cuda":

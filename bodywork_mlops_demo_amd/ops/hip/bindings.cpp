@@ -40,6 +40,12 @@ std::tuple<at::Tensor, at::Tensor> linear_relu_mask_bf16_hip(
     const c10::optional<at::Tensor>& bias);
 at::Tensor gemm_tn_bf16_hip(const at::Tensor& a, const at::Tensor& b,
                             bool out_fp32);
+// gemm_mx8.hip — MX-fp8 (e4m3) K=128 scaled-MFMA path
+at::Tensor quantize_e4m3_hip(const at::Tensor& x, int64_t e);
+at::Tensor gemm_mx8_nt_hip(const at::Tensor& a8, int64_t ea,
+                           const at::Tensor& b8, int64_t eb,
+                           const c10::optional<at::Tensor>& bias, bool relu,
+                           bool out_fp32);
 // optim.hip
 void adam_step_hip(at::Tensor p, const at::Tensor& g, at::Tensor m,
                    at::Tensor v, const c10::optional<at::Tensor>& p_bf16,
@@ -71,6 +77,9 @@ TORCH_LIBRARY(bodywork_hip, m) {
   m.def("linear_relu_mask_bf16(Tensor x, Tensor w, Tensor? bias) -> "
         "(Tensor, Tensor)");
   m.def("gemm_tn_bf16(Tensor a, Tensor b, bool out_fp32) -> Tensor");
+  m.def("quantize_e4m3(Tensor x, int e) -> Tensor");
+  m.def("gemm_mx8_nt(Tensor a8, int ea, Tensor b8, int eb, Tensor? bias, "
+        "bool relu, bool out_fp32) -> Tensor");
   m.def("adam_step(Tensor(a!) p, Tensor g, Tensor(b!) m, Tensor(c!) v, "
         "Tensor(d!)? p_bf16, float lr, float beta1, float beta2, float eps, "
         "int t, Tensor? bc) -> ()");
@@ -96,6 +105,8 @@ TORCH_LIBRARY_IMPL(bodywork_hip, CUDA, m) {
   m.impl("linear_bf16", linear_bf16_hip);
   m.impl("linear_relu_mask_bf16", linear_relu_mask_bf16_hip);
   m.impl("gemm_tn_bf16", gemm_tn_bf16_hip);
+  m.impl("quantize_e4m3", quantize_e4m3_hip);
+  m.impl("gemm_mx8_nt", gemm_mx8_nt_hip);
   m.impl("adam_step", adam_step_hip);
   m.impl("batch_indices", batch_indices_hip);
   m.impl("transpose_to_bf16", transpose_to_bf16_hip);

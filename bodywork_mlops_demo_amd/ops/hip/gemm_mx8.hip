@@ -1,0 +1,361 @@
+// MX-FP8 (e4m3) 256x256-tile 8-phase NT GEMM for gfx950 — the 2x-rate
+// low-precision path.
+//
+// Why this shape: the non-scaled fp8 MFMAs (mfma_f32_16x16x32_fp8_fp8)
+// run at the BF16 rate on CDNA4 — the ONLY 2x-rate fp8 instruction is
+// the block-scaled __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4
+// (K=128, per-32-element E8M0 scales, HW-fused dequant).  This kernel
+// feeds it PER-TENSOR power-of-two scales: every lane passes the same
+// E8M0 byte (the tensor's shared exponent), so the hardware dequant does
+// the rescale for free and there is ZERO scale memory traffic in the
+// K-loop — which lets the kernel keep gemm8.hip's proven 16-wave 8-phase
+// structure completely unchanged in its memory system:
+//
+//   BK = 128 fp8 bytes/row  ==  gemm8's BK=64 bf16 bytes/row (128 B)
+//   -> identical half-tile size (16 KiB), identical glds schedule,
+//      identical vmcnt(2)/vmcnt(0) landing proofs, identical XOR swizzle
+//      (8 x 16-B chunks per row), identical barrier structure.
+//
+// What changes: one K-tile is ONE K=128 MFMA per output fragment (vs
+// four K=32 bf16 MFMAs), each lane's A/B fragment is 32 consecutive k
+// bytes = chunks {2*kg, 2*kg+1} (two ds_read_b128), and the MFMA count
+// per phase halves while each MFMA is 4x the work -> the schedule gets
+// MORE compute per staged byte, i.e. deeper latency hiding than bf16.
+//
+// Numerics: A and B are e4m3 with per-tensor shared exponents ea, eb
+// (value = 2^e * stored).  Scale operands sa = ea+127, sb = eb+127
+// (E8M0); C = 2^(ea+eb) * (Aq . Bq) accumulated in fp32 by the MFMA.
+// Power-of-two scaling keeps dequantisation exact.
+//
+// The reference computes its scoring forward in fp64 sklearn
+// (stage_2_serve_model.py:78); this kernel is the opt-in low-precision
+// serving/training path with MAPE parity gates in the tests.
+#include <ATen/cuda/CUDAContext.h>
+#include <hip/hip_runtime.h>
+#include <torch/extension.h>
+
+#include "bf16_utils.h"
+
+#define MX_BM 256
+#define MX_BN 256
+#define MX_BK 128           // fp8 elements per K-tile (128 bytes/row)
+#define MX_THREADS 1024
+#define MX_HTB (128 * 128)  // bytes per half-tile slot
+
+typedef __attribute__((ext_vector_type(8))) int mx_i32x8;
+typedef __attribute__((ext_vector_type(4))) int mx_i32x4;
+typedef __attribute__((ext_vector_type(4))) float mx_f32x4;
+
+#define MX_EPI_NONE 0
+#define MX_EPI_BIAS_RELU 1
+
+__device__ __forceinline__ void mx_glds16(const void* gsrc, void* lds_dst) {
+  __builtin_amdgcn_global_load_lds(
+      (const __attribute__((address_space(1))) unsigned int*)gsrc,
+      (__attribute__((address_space(3))) unsigned int*)lds_dst, 16, 0, 0);
+}
+
+// stage one [128 rows][128 B] half-tile: 1024 16-B chunks, 1024 threads
+// -> ONE wave-level glds per wave (identical to gemm8's g8_stage_half).
+__device__ __forceinline__ void mx_stage_half(char* __restrict__ slot,
+                                              const char* __restrict__ base,
+                                              int off0) {
+  const int wave = threadIdx.x >> 6;
+  mx_glds16(base + off0, slot + wave * 1024);
+}
+
+template <int EPI, bool HAS_BIAS, bool OUT_FP32>
+__device__ __forceinline__ void mx_epilogue(
+    mx_f32x4 (&acc)[4][4], const float* __restrict__ bias,
+    void* __restrict__ C, long long M, long long N, long long m0,
+    long long n0, int wm, int wn, int fl, int kg) {
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      long long col = n0 + wn * 64 + j * 16 + fl;
+      bool col_ok = col < N;
+      float bval =
+          (EPI == MX_EPI_BIAS_RELU && HAS_BIAS && col_ok) ? bias[col] : 0.0f;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        long long row = m0 + wm * 64 + i * 16 + kg * 4 + r;
+        float v = acc[i][j][r];
+        if (EPI == MX_EPI_BIAS_RELU) {
+          v += bval;
+          v = fmaxf(v, 0.0f);
+        }
+        if (row < M && col_ok) {
+          if (OUT_FP32)
+            ((float*)C)[row * N + col] = v;
+          else
+            ((bf16_t*)C)[row * N + col] = f32_to_bf16(v);
+        }
+      }
+    }
+  }
+}
+
+template <int EPI, bool HAS_BIAS, bool OUT_FP32>
+__launch_bounds__(MX_THREADS)
+__global__ void gemm_mx8_nt_kernel(
+    const unsigned char* __restrict__ A,  // [M,K] e4m3, row-major
+    const unsigned char* __restrict__ B,  // [N,K] e4m3, row-major
+    const float* __restrict__ bias, void* __restrict__ C, long long M,
+    long long N, long long K, int sa, int sb) {  // sa/sb: E8M0 bytes
+  __shared__ char lds[8 * MX_HTB];  // ONE __shared__ object (guide trap (a))
+  const long long m0 = (long long)blockIdx.y * MX_BM;
+  const long long n0 = (long long)blockIdx.x * MX_BN;
+  const int wave = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int wm = wave >> 2;  // 0..3 (64 C-rows per wave)
+  const int wn = wave & 3;   // 0..3 (64 C-cols per wave)
+  const int fl = lane & 15;
+  const int kg = lane >> 4;
+  const int swz = fl & 7;
+  const long long nt = K / MX_BK;
+
+#define MX_ASLOT(buf, half) (lds + ((buf) * 2 + (half)) * MX_HTB)
+#define MX_BSLOT(buf, half) (lds + 4 * MX_HTB + ((buf) * 2 + (half)) * MX_HTB)
+
+  const int a_inhalf = (wm & 1) * 64;
+  const int b_inhalf = (wn & 1) * 64;
+
+  mx_f32x4 acc[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  // one fragment = 32 consecutive k bytes = LDS chunks {2kg, 2kg+1}
+  // (XOR-swizzled), read as two b128s into an i32x8 MFMA operand
+#define MX_READ8(dst, rowbase, rowoff)                                      \
+  do {                                                                      \
+    const char* _rb = (rowbase) + (rowoff) * 128;                           \
+    mx_i32x4 _lo = *(const mx_i32x4*)(_rb + (((2 * kg) ^ swz) * 16));       \
+    mx_i32x4 _hi = *(const mx_i32x4*)(_rb + (((2 * kg + 1) ^ swz) * 16));   \
+    dst[0] = _lo[0]; dst[1] = _lo[1]; dst[2] = _lo[2]; dst[3] = _lo[3];     \
+    dst[4] = _hi[0]; dst[5] = _hi[1]; dst[6] = _hi[2]; dst[7] = _hi[3];     \
+  } while (0)
+
+#define MX_AREAD(dst, buf, mfrag)                                           \
+  MX_READ8(dst, MX_ASLOT(buf, (wm >> 1)), a_inhalf + (mfrag) * 16 + fl)
+#define MX_BREAD(dst, buf, nfrag)                                           \
+  MX_READ8(dst, MX_BSLOT(buf, (wn >> 1)), b_inhalf + (nfrag) * 16 + fl)
+
+  mx_i32x8 a_q[2];     // this phase's two A fragments (mfrag pair)
+  mx_i32x8 b_t[4];     // current tile's B fragments [nfrag]
+
+  // per-lane staging byte offset (tile-invariant): chunk ci covers
+  // row = ci>>3, 16-B column chunk (ci&7)^XOR(row) of the 128-B row
+  int stg_off;
+  {
+    int ci = (int)threadIdx.x;
+    int row = ci >> 3;
+    int sc = (ci & 7) ^ (row & 7);
+    stg_off = (int)(row * K + sc * 16);
+  }
+  const char* Ah0 = (const char*)(A + m0 * K);
+  const char* Ah1 = (const char*)(A + (m0 + 128) * K);
+  const char* Bh0 = (const char*)(B + n0 * K);
+  const char* Bh1 = (const char*)(B + (n0 + 128) * K);
+#define MX_KOFF(T) ((long long)(T) * MX_BK)
+
+  // prologue: A(0), B(0), B(1) — A(1) is issued by the loop at (0,A)
+  mx_stage_half(MX_ASLOT(0, 0), Ah0, stg_off);
+  mx_stage_half(MX_ASLOT(0, 1), Ah1, stg_off);
+  mx_stage_half(MX_BSLOT(0, 0), Bh0, stg_off);
+  mx_stage_half(MX_BSLOT(0, 1), Bh1, stg_off);
+  if (nt > 1) {
+    mx_stage_half(MX_BSLOT(1, 0), Bh0 + MX_KOFF(1), stg_off);
+    mx_stage_half(MX_BSLOT(1, 1), Bh1 + MX_KOFF(1), stg_off);
+  }
+
+  // Two phases per K-tile, one barrier each — gemm8's v4 schedule with
+  // identical glds issue order, so the same landing proof holds: at
+  // wait(T,A) the newest 2 outstanding glds are B(T+1)'s halves, the
+  // 3rd/4th-newest are A(T)'s -> vmcnt(2) proves tile T landed; the
+  // last tile drains with vmcnt(0).  Slot safety: A(T+1) overwrites
+  // A(T-1) whose mfrag2/3 reads were consumed before barrier(T,A);
+  // B(T+2) overwrites B(T), consumed before barrier(T,B).
+#define MX_MFMA(acc_i, afrag)                                               \
+  _Pragma("unroll") for (int nf = 0; nf < 4; ++nf)                          \
+      acc[acc_i][nf] = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(    \
+          afrag, b_t[nf], acc[acc_i][nf], 0, 0, 0, sa, 0, sb)
+
+#define MX_PHASE_A(T, TPAR)                                                 \
+  do {                                                                      \
+    if ((T) + 1 < nt)                                                       \
+      asm volatile("s_waitcnt vmcnt(2)" ::: "memory");                      \
+    else                                                                    \
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");                      \
+    __builtin_amdgcn_s_barrier();                                           \
+    if ((T) + 1 < nt) {                                                     \
+      mx_stage_half(MX_ASLOT(1 - (TPAR), 0), Ah0 + MX_KOFF((T) + 1),        \
+                    stg_off);                                               \
+      mx_stage_half(MX_ASLOT(1 - (TPAR), 1), Ah1 + MX_KOFF((T) + 1),        \
+                    stg_off);                                               \
+    }                                                                       \
+    _Pragma("unroll") for (int nf = 0; nf < 4; ++nf)                        \
+        MX_BREAD(b_t[nf], TPAR, nf);                                        \
+    MX_AREAD(a_q[0], TPAR, 0);                                              \
+    MX_AREAD(a_q[1], TPAR, 1);                                              \
+    __builtin_amdgcn_s_setprio(1);                                          \
+    MX_MFMA(0, a_q[0]);                                                     \
+    MX_MFMA(1, a_q[1]);                                                     \
+    __builtin_amdgcn_s_setprio(0);                                          \
+  } while (0)
+
+#define MX_PHASE_B(T, TPAR)                                                 \
+  do {                                                                      \
+    __builtin_amdgcn_s_barrier();                                           \
+    if ((T) + 2 < nt) {                                                     \
+      mx_stage_half(MX_BSLOT(TPAR, 0), Bh0 + MX_KOFF((T) + 2), stg_off);    \
+      mx_stage_half(MX_BSLOT(TPAR, 1), Bh1 + MX_KOFF((T) + 2), stg_off);    \
+    }                                                                       \
+    MX_AREAD(a_q[0], TPAR, 2);                                              \
+    MX_AREAD(a_q[1], TPAR, 3);                                              \
+    __builtin_amdgcn_s_setprio(1);                                          \
+    MX_MFMA(2, a_q[0]);                                                     \
+    MX_MFMA(3, a_q[1]);                                                     \
+    __builtin_amdgcn_s_setprio(0);                                          \
+  } while (0)
+
+  // two tiles per iteration: buffer indices stay compile-time
+  for (long long t = 0; t < nt; t += 2) {
+    MX_PHASE_A(t, 0);
+    MX_PHASE_B(t, 0);
+    if (t + 1 < nt) {
+      MX_PHASE_A(t + 1, 1);
+      MX_PHASE_B(t + 1, 1);
+    }
+  }
+#undef MX_PHASE_A
+#undef MX_PHASE_B
+#undef MX_MFMA
+#undef MX_AREAD
+#undef MX_BREAD
+#undef MX_READ8
+#undef MX_ASLOT
+#undef MX_BSLOT
+
+  mx_epilogue<EPI, HAS_BIAS, OUT_FP32>(acc, bias, C, M, N, m0, n0, wm, wn,
+                                       fl, kg);
+}
+
+// ---- per-tensor e4m3 quantisation ----------------------------------------
+//
+// x (bf16 or fp32) -> e4m3 bytes, value = 2^e * stored.  e is chosen
+// host-side so |x|max / 2^e <= 448 (no saturation).  The HW fast-path
+// packed convert (cvt_pk_fp8_f32) rounds to nearest-even.
+template <typename T>
+__global__ void quantize_e4m3_kernel(const T* __restrict__ x,
+                                     unsigned char* __restrict__ out,
+                                     long long n, float inv_scale) {
+  long long i = (long long)(blockIdx.x) * blockDim.x + threadIdx.x;
+  long long stride = (long long)gridDim.x * blockDim.x;
+  for (; i * 2 + 1 < n; i += stride) {
+    float v0, v1;
+    if constexpr (sizeof(T) == 2) {
+      v0 = bf16_to_f32(((const bf16_t*)x)[i * 2]);
+      v1 = bf16_to_f32(((const bf16_t*)x)[i * 2 + 1]);
+    } else {
+      v0 = ((const float*)x)[i * 2];
+      v1 = ((const float*)x)[i * 2 + 1];
+    }
+    // packed RNE convert; lower 16 bits hold the two e4m3 bytes
+    int packed = __builtin_amdgcn_cvt_pk_fp8_f32(v0 * inv_scale,
+                                                 v1 * inv_scale, 0, false);
+    *(unsigned short*)(out + i * 2) = (unsigned short)(packed & 0xFFFF);
+  }
+  // odd tail element (single-element convert via the same packed op)
+  if (i * 2 < n && n % 2 == 1 && i * 2 == n - 1) {
+    float v = sizeof(T) == 2 ? bf16_to_f32(((const bf16_t*)x)[n - 1])
+                             : ((const float*)x)[n - 1];
+    int packed =
+        __builtin_amdgcn_cvt_pk_fp8_f32(v * inv_scale, 0.0f, 0, false);
+    out[n - 1] = (unsigned char)(packed & 0xFF);
+  }
+}
+
+// ---- host wrappers --------------------------------------------------------
+
+at::Tensor quantize_e4m3_hip(const at::Tensor& x, int64_t e) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous(), "quantize_e4m3: cuda contig");
+  TORCH_CHECK(x.scalar_type() == torch::kBFloat16 ||
+                  torch::kFloat == x.scalar_type(),
+              "quantize_e4m3: bf16 or fp32 input");
+  auto out = torch::empty_like(x, x.options().dtype(torch::kUInt8));
+  long long n = x.numel();
+  if (n == 0) return out;
+  float inv_scale = ldexpf(1.0f, (int)-e);
+  auto stream = at::cuda::getCurrentCUDAStream();
+  long long pairs = (n + 1) / 2;
+  int threads = 256;
+  int blocks = (int)std::min<long long>((pairs + threads - 1) / threads,
+                                        8192);
+  if (x.scalar_type() == torch::kBFloat16)
+    hipLaunchKernelGGL(quantize_e4m3_kernel<short>, dim3(blocks),
+                       dim3(threads), 0, stream,
+                       (const short*)x.data_ptr(),
+                       out.data_ptr<unsigned char>(), n, inv_scale);
+  else
+    hipLaunchKernelGGL(quantize_e4m3_kernel<float>, dim3(blocks),
+                       dim3(threads), 0, stream, x.data_ptr<float>(),
+                       out.data_ptr<unsigned char>(), n, inv_scale);
+  return out;
+}
+
+// C[M,N] = 2^(ea+eb) * (A[M,K]e4m3 . B[N,K]e4m3^T), optional fused
+// bias+relu epilogue, fp32 or bf16 output.
+at::Tensor gemm_mx8_nt_hip(const at::Tensor& a8, int64_t ea,
+                           const at::Tensor& b8, int64_t eb,
+                           const c10::optional<at::Tensor>& bias, bool relu,
+                           bool out_fp32) {
+  TORCH_CHECK(a8.is_cuda() && b8.is_cuda(), "gemm_mx8_nt: cuda tensors");
+  TORCH_CHECK(a8.scalar_type() == torch::kUInt8 &&
+                  b8.scalar_type() == torch::kUInt8,
+              "gemm_mx8_nt: e4m3 byte tensors");
+  TORCH_CHECK(a8.dim() == 2 && b8.dim() == 2 && a8.size(1) == b8.size(1),
+              "gemm_mx8_nt: [M,K]x[N,K]");
+  TORCH_CHECK(a8.is_contiguous() && b8.is_contiguous(), "contiguous");
+  long long M = a8.size(0), K = a8.size(1), N = b8.size(0);
+  TORCH_CHECK(M % MX_BM == 0 && N % MX_BN == 0 && K % MX_BK == 0,
+              "gemm_mx8_nt requires M%256==0, N%256==0, K%128==0 (got ", M,
+              "x", N, "x", K, ")");
+  int sa = (int)std::min<long long>(std::max<long long>(ea + 127, 0), 254);
+  int sb = (int)std::min<long long>(std::max<long long>(eb + 127, 0), 254);
+  TORCH_CHECK(sa == ea + 127 && sb == eb + 127,
+              "per-tensor exponent out of E8M0 range");
+  auto opts = a8.options().dtype(out_fp32 ? torch::kFloat : torch::kBFloat16);
+  auto C = torch::empty({M, N}, opts);
+  const float* bias_p = nullptr;
+  bool has_bias = false;
+  if (bias.has_value() && bias->defined()) {
+    TORCH_CHECK(bias->is_cuda() && bias->scalar_type() == torch::kFloat &&
+                    bias->numel() == N,
+                "bias: fp32 [N] cuda");
+    bias_p = bias->data_ptr<float>();
+    has_bias = true;
+  }
+  dim3 grid((unsigned)(N / MX_BN), (unsigned)(M / MX_BM));
+  auto stream = at::cuda::getCurrentCUDAStream();
+  const unsigned char* ap = a8.data_ptr<unsigned char>();
+  const unsigned char* bp = b8.data_ptr<unsigned char>();
+#define LMX(EPI_, HB_, OF_)                                                 \
+  hipLaunchKernelGGL((gemm_mx8_nt_kernel<EPI_, HB_, OF_>), grid,            \
+                     dim3(MX_THREADS), 0, stream, ap, bp, bias_p,           \
+                     C.data_ptr(), M, N, K, sa, sb)
+  if (relu) {
+    if (has_bias) { if (out_fp32) LMX(MX_EPI_BIAS_RELU, true, true);
+                    else          LMX(MX_EPI_BIAS_RELU, true, false); }
+    else          { if (out_fp32) LMX(MX_EPI_BIAS_RELU, false, true);
+                    else          LMX(MX_EPI_BIAS_RELU, false, false); }
+  } else {
+    TORCH_CHECK(!has_bias, "bias requires relu epilogue for now");
+    if (out_fp32) LMX(MX_EPI_NONE, false, true);
+    else          LMX(MX_EPI_NONE, false, false);
+  }
+#undef LMX
+  return C;
+}

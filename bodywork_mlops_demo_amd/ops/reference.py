@@ -286,3 +286,71 @@ def adam_step_cpu(p, g, m, v, p_bf16, lr, t, beta1, beta2, eps):
     p.sub_(lr * mhat / (vhat.sqrt() + eps))
     if p_bf16 is not None:
         p_bf16.copy_(p.bfloat16())
+
+
+# --------------------------------------------------------------------------
+# MX-fp8 (OCP e4m3) oracle — CPU semantics for ops/hip/gemm_mx8.hip
+# --------------------------------------------------------------------------
+
+def _e4m3_table() -> np.ndarray:
+    """All 256 OCP e4m3fn code values (S.EEEE.MMM, bias 7; 0x7F/0xFF NaN)."""
+    codes = np.arange(256, dtype=np.uint8)
+    s = (codes >> 7) & 1
+    e = (codes >> 3) & 0xF
+    m = codes & 7
+    v = np.where(e == 0, (m / 8.0) * 2.0 ** -6,
+                 (1.0 + m / 8.0) * (2.0 ** (e.astype(np.int32) - 7)))
+    v = np.where(s == 1, -v, v)
+    v[(e == 15) & (m == 7)] = np.nan  # e4m3fn NaN (no infinities)
+    return v.astype(np.float64)
+
+
+_E4M3 = _e4m3_table()
+# positive grid for nearest-value quantisation (finite codes 0..0x7E)
+_E4M3_POS = np.sort(_E4M3[:127][np.isfinite(_E4M3[:127])])
+
+
+def quantize_e4m3_cpu(x: torch.Tensor, e: int) -> torch.Tensor:
+    """Round-to-nearest-even x/2^e onto the e4m3 grid; returns byte codes
+    matching the GPU cvt_pk_fp8_f32 path (saturating at +-448)."""
+    v = (x.detach().to(torch.float64).cpu().numpy() / (2.0 ** e))
+    v = np.clip(v, -448.0, 448.0)
+    mag = np.abs(v)
+    idx = np.searchsorted(_E4M3_POS, mag)
+    lo = _E4M3_POS[np.clip(idx - 1, 0, len(_E4M3_POS) - 1)]
+    hi = _E4M3_POS[np.clip(idx, 0, len(_E4M3_POS) - 1)]
+    pick_hi = (mag - lo) > (hi - mag)
+    # ties -> even mantissa; on this grid even-mantissa neighbours are the
+    # ones whose code is even, resolve exact ties toward the lower-ulp
+    # magnitude with even code
+    tie = (mag - lo) == (hi - mag)
+    q = np.where(pick_hi, hi, lo)
+    if tie.any():
+        lo_codes = np.searchsorted(_E4M3_POS, lo)
+        q = np.where(tie & (lo_codes % 2 == 1), hi, q)
+    # encode: find code whose value equals q (positive), then set sign
+    code_pos = np.searchsorted(_E4M3_POS, q).astype(np.uint8)
+    # map grid index back to byte code: positive finite codes are exactly
+    # 0..126 in value order (table is monotone over positive codes)
+    byte = code_pos
+    byte = np.where(np.signbit(v) & (q != 0), byte | 0x80, byte)
+    return torch.from_numpy(byte.astype(np.uint8)).reshape(x.shape)
+
+
+def e4m3_decode_cpu(codes: torch.Tensor, e: int = 0) -> torch.Tensor:
+    vals = _E4M3[codes.detach().cpu().numpy().astype(np.uint8)]
+    return torch.from_numpy((vals * (2.0 ** e)).astype(np.float32)).reshape(
+        codes.shape)
+
+
+def gemm_mx8_nt_cpu(a8: torch.Tensor, ea: int, b8: torch.Tensor, eb: int,
+                    bias: torch.Tensor | None = None, relu: bool = False,
+                    out_fp32: bool = False) -> torch.Tensor:
+    a = e4m3_decode_cpu(a8, ea).to(torch.float32)
+    b = e4m3_decode_cpu(b8, eb).to(torch.float32)
+    c = a @ b.t()
+    if bias is not None:
+        c = c + bias.to(torch.float32)
+    if relu:
+        c = torch.relu(c)
+    return c if out_fp32 else c.to(torch.bfloat16)

@@ -25,6 +25,7 @@ sources = [
     os.path.join(HIP_DIR, "mlp_small.hip"),
     os.path.join(HIP_DIR, "gemm.hip"),
     os.path.join(HIP_DIR, "gemm8.hip"),
+    os.path.join(HIP_DIR, "gemm_mx8.hip"),
     os.path.join(HIP_DIR, "optim.hip"),
 ]
 

@@ -1,0 +1,85 @@
+"""MX-fp8 (e4m3, K=128 scaled-MFMA) GEMM numerics on hardware.
+
+The exact-integer tests verify the 16x16x128 fragment LAYOUT and the
+E8M0 scale-operand semantics bit-exactly (integer operands are e4m3-
+representable and the fp32 accumulation of |v|<=8 over K<=1024 is
+exact); the random tests bound quantisation error vs the plain fp32
+reference of the same op (the framework's numerics-test contract).
+"""
+import pytest
+import torch
+
+from bodywork_mlops_demo_amd import ops
+
+pytestmark = pytest.mark.gpu
+
+DEV = "cuda"
+
+
+def _randint(m, k, seed):
+    g = torch.Generator(device=DEV).manual_seed(seed)
+    return torch.randint(-8, 9, (m, k), generator=g, device=DEV).float()
+
+
+def test_mx8_exact_integer_layout():
+    a = _randint(256, 512, 1)
+    b = _randint(512, 512, 2)  # asymmetric B (guide: catches row/col swap)
+    got = ops.gemm_mx8_nt(ops.quantize_e4m3(a, 0), 0,
+                          ops.quantize_e4m3(b, 0), 0, out_fp32=True)
+    assert torch.equal(got, a @ b.t())
+
+
+def test_mx8_scale_operands_exact():
+    a = _randint(256, 1024, 3)
+    b = _randint(256, 1024, 4)
+    a8 = ops.quantize_e4m3(a * 32.0, 5)     # stored = value / 2^5
+    b8 = ops.quantize_e4m3(b * 0.125, -3)   # stored = value * 2^3
+    got = ops.gemm_mx8_nt(a8, 5, b8, -3, out_fp32=True)
+    assert torch.equal(got, (a * 32.0) @ (b * 0.125).t())
+
+
+def test_mx8_quantizer_matches_cpu_oracle():
+    g = torch.Generator(device=DEV).manual_seed(5)
+    x = torch.randn(8192, generator=g, device=DEV) * 17.0
+    e = ops.e4m3_exponent(x.abs().max().item())
+    gq = ops.quantize_e4m3(x, e).cpu()
+    cq = ops.quantize_e4m3(x.cpu(), e)
+    # RNE agreement everywhere except exact grid-midpoint ties
+    assert (gq == cq).float().mean().item() > 0.999
+    dg = ops.reference.e4m3_decode_cpu(gq, e)
+    dc = ops.reference.e4m3_decode_cpu(cq, e)
+    assert (dg - dc).abs().max().item() <= 2.0 ** (e - 2)
+
+
+def test_mx8_random_accuracy_and_epilogue():
+    g = torch.Generator(device=DEV).manual_seed(6)
+    a = torch.randn(512, 4096, generator=g, device=DEV)
+    b = torch.randn(256, 4096, generator=g, device=DEV)
+    ea = ops.e4m3_exponent(a.abs().max().item())
+    eb = ops.e4m3_exponent(b.abs().max().item())
+    a8 = ops.quantize_e4m3(a, ea)
+    b8 = ops.quantize_e4m3(b, eb)
+    got = ops.gemm_mx8_nt(a8, ea, b8, eb, out_fp32=True)
+    # vs fp32 matmul of the dequantised operands: MFMA accumulation only
+    want_q = (ops.reference.e4m3_decode_cpu(a8.cpu(), ea).to(DEV)
+              @ ops.reference.e4m3_decode_cpu(b8.cpu(), eb).to(DEV).t())
+    rel = ((got - want_q).abs().max()
+           / want_q.abs().max().clamp_min(1e-6)).item()
+    assert rel < 1e-3, rel
+    # vs the plain fp32 reference of the same op: bounded quantisation err
+    want = a @ b.t()
+    mean_rel = ((got - want).abs().mean() / want.abs().mean()).item()
+    assert mean_rel < 0.05, mean_rel
+    # fused bias+relu epilogue
+    bias = torch.randn(256, generator=g, device=DEV)
+    got2 = ops.gemm_mx8_nt(a8, ea, b8, eb, bias=bias, relu=True,
+                           out_fp32=True)
+    want2 = torch.relu(want_q + bias)
+    assert ((got2 - want2).abs().max().item()
+            <= 1e-2 * want2.abs().max().item())
+
+
+def test_mx8_requires_extension_on_gpu():
+    """The MX path must run the HIP kernel on GPU, never a silent
+    fallback."""
+    assert ops.hip_available()

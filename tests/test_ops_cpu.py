@@ -159,3 +159,33 @@ class TestMLPOps:
         assert torch.allclose(dw, h.float().t() @ v)
         assert torch.allclose(cs, h.float().sum(0))
         assert torch.allclose(ops.colsum_bf16(h), cs)
+
+
+def test_e4m3_oracle_roundtrip_and_cpu_gemm():
+    """CPU oracle for the MX-fp8 path: quantise/decode relative error is
+    bounded by the e4m3 grid (2^-3 relative for normals), and the CPU
+    gemm matches a plain fp32 matmul of the decoded operands."""
+    import torch
+
+    from bodywork_mlops_demo_amd import ops
+
+    g = torch.Generator().manual_seed(11)
+    x = torch.randn(4096, generator=g) * 3.0
+    e = ops.e4m3_exponent(x.abs().max().item())
+    codes = ops.quantize_e4m3(x, e)
+    dec = ops.reference.e4m3_decode_cpu(codes, e)
+    err = (dec - x).abs()
+    # RNE on a 3-mantissa-bit grid: err <= 2^-4 * 2^floor(log2|x|) + subnormal floor
+    bound = x.abs() * (2.0 ** -4) + (2.0 ** (e - 7))
+    assert (err <= bound + 1e-7).all()
+
+    a = torch.randint(-8, 9, (8, 64), generator=g).float()
+    b = torch.randint(-8, 9, (16, 64), generator=g).float()
+    got = ops.gemm_mx8_nt(ops.quantize_e4m3(a, 0), 0,
+                          ops.quantize_e4m3(b, 0), 0, out_fp32=True)
+    assert torch.equal(got, a @ b.t())
+
+    # exponent selection never saturates
+    for amax in (0.0, 1e-8, 1.0, 447.9, 448.0, 1e12):
+        ee = ops.e4m3_exponent(amax)
+        assert amax / (2.0 ** ee) <= 448.0
