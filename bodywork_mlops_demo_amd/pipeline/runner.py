@@ -87,6 +87,7 @@ class PipelineRunner:
         base_port: int = 5000,
         n_gpus: int | None = None,
         front_proxy: bool = True,
+        isolate_envs: bool = False,
     ):
         self.config = (
             config if isinstance(config, PipelineConfig) else load_config(config)
@@ -97,6 +98,10 @@ class PipelineRunner:
         )
         self.base_port = base_port
         self.front_proxy = front_proxy
+        # per-stage isolated envs (reference bodywork.yaml:10-16 installs
+        # a different pinned pip list per stage); off = validate-only
+        self.isolate_envs = isolate_envs
+        self._env_mgr = None
         self.services: dict[str, ServiceHandle] = {}
         self._secrets = self._load_secrets(secrets_file)
         if n_gpus is None:
@@ -173,11 +178,20 @@ class PipelineRunner:
         return problems
 
     # -- stage execution ---------------------------------------------------
+    def _stage_python(self, spec: StageSpec) -> str:
+        if not self.isolate_envs:
+            return sys.executable
+        if self._env_mgr is None:
+            from bodywork_mlops_demo_amd.pipeline.envs import StageEnvManager
+
+            self._env_mgr = StageEnvManager()
+        return self._env_mgr.python_for(spec.requirements)
+
     def _module_cmd(self, spec: StageSpec) -> list[str]:
         path = spec.executable_module_path
         mod = path[:-3] if path.endswith(".py") else path
         mod = mod.replace("/", ".").replace(os.sep, ".")
-        return [sys.executable, "-m", mod, *spec.args]
+        return [self._stage_python(spec), "-m", mod, *spec.args]
 
     def _run_batch_stage(self, spec: StageSpec, report: RunReport) -> bool:
         assert spec.batch is not None
@@ -450,6 +464,11 @@ def main(argv=None) -> None:
     p.add_argument("--store", default=None)
     p.add_argument("--secrets", default=None)
     p.add_argument("--keep-services", action="store_true")
+    p.add_argument("--isolate-envs", action="store_true",
+                   help="run each stage in its own cached venv built from "
+                        "the stage's requirements list (reference per-stage "
+                        "pip installs, bodywork.yaml:10-16); default "
+                        "validates requirements against the shared env")
     p.add_argument("--repeat", type=int, default=1,
                    help="run the DAG N times (the reference's daily k8s "
                         "cronjob role, README.md:5); advances the virtual "
@@ -458,7 +477,8 @@ def main(argv=None) -> None:
                    help="seconds to sleep between repeated runs")
     args = p.parse_args(argv)
     runner = PipelineRunner(args.config, store_uri=args.store,
-                            secrets_file=args.secrets)
+                            secrets_file=args.secrets,
+                            isolate_envs=args.isolate_envs)
     ok = True
     for cycle in range(args.repeat):
         report = runner.run(teardown_services=not args.keep_services)

@@ -6,11 +6,14 @@ measures quantisation accuracy on random data; `--bench` prints TFLOP/s
 on the hot shapes next to the bf16 8-phase kernel for the A/B.
 """
 import argparse
+import os
+import sys
 import time
 
 import torch
 
-from bodywork_mlops_demo_amd import ops
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+from bodywork_mlops_demo_amd import ops  # noqa: E402
 
 
 def _quant(x):
