@@ -85,24 +85,37 @@ def test_mx8_requires_extension_on_gpu():
     assert ops.hip_available()
 
 
-def test_mlp_fp8_scoring_matches_bf16():
-    """The MLP's opt-in fp8 scoring forward: same weights, fp8 h2 GEMM vs
-    bf16 h2 GEMM — bounded quantisation error on the final predictions,
-    and non-tile-multiple batches fall back to the bf16 kernel."""
+def test_mlp_fp8_scoring_task_parity():
+    """The MAPE parity gate for fp8 scoring: on a TRAINED model and the
+    task's data distribution, fp8 predictions track bf16 on the y scale
+    and the online MAPE matches.  (An untrained random-weight model is
+    the wrong oracle here: its outputs are near-zero cancellations of
+    large h2 terms, so any layer-wise quantisation error is amplified
+    unboundedly in relative terms — measured bench parity on the real
+    task is MAPE 2.4948 fp8 vs 2.5058 bf16, r02_bench_mlp_fp8.log.)"""
     from bodywork_mlops_demo_amd.models import GPUMLPRegressor
 
-    m_bf = GPUMLPRegressor(hidden=512, device=DEV, seed=123)
-    m_f8 = GPUMLPRegressor(hidden=512, device=DEV, seed=123,
+    y, X = ops.datagen(200_000, 50, 42, device=DEV)
+    m_bf = GPUMLPRegressor(hidden=512, device=DEV, seed=1)
+    m_bf.fit(X, y, steps=30, batch_size=16384)
+    m_f8 = GPUMLPRegressor(hidden=512, device=DEV, seed=1,
                            fp8_scoring=True)
-    g = torch.Generator(device=DEV).manual_seed(9)
-    X = torch.rand(4096, generator=g, device=DEV) * 100
-    y_bf = m_bf.predict(X)
-    y_f8 = m_f8.predict(X)
-    denom = y_bf.abs().mean().clamp_min(1e-3)
-    assert ((y_f8 - y_bf).abs().mean() / denom).item() < 0.02
-    # M % 256 != 0 -> bf16 fallback, still correct
-    y_tail = m_f8.predict(X[:1000])
-    assert torch.allclose(y_tail, y_bf[:1000], rtol=1e-2, atol=1e-2)
+    assert m_f8.copy_weights_from(m_bf)
+
+    Xe = X[:4096]
+    p_bf = m_bf.predict(Xe)
+    p_f8 = m_f8.predict(Xe)
+    # predictions agree on the scale of the labels
+    scale = y.abs().mean()
+    assert ((p_f8 - p_bf).abs().mean() / scale).item() < 0.01
+    # online-MAPE parity (the drift loop's quality gate)
+    mape_bf = ops.score_label_metrics(p_bf, y[:4096])["MAPE"]
+    mape_f8 = ops.score_label_metrics(p_f8, y[:4096])["MAPE"]
+    assert abs(mape_f8 - mape_bf) < 0.05 * mape_bf + 0.01, (mape_bf,
+                                                            mape_f8)
+    # M % 256 != 0 -> bf16 fallback path, must equal bf16 exactly
+    t = m_f8.predict(Xe[:1000])
+    torch.testing.assert_close(t, p_bf[:1000], rtol=1e-5, atol=1e-5)
 
 
 def test_mlp_fp8_scorer_capture_and_hot_redeploy():
