@@ -49,9 +49,13 @@ def test_stage2_scorer_and_stage4_loadtest_inproc(seeded_store):
     metrics = loadtest.run(seeded_store, device="cpu", scorer=scorer)
     assert seeded_store.exists(contract.test_metrics_key(date(2026, 1, 3)))
     rec = seeded_store.get_metrics_csv(contract.test_metrics_key(date(2026, 1, 3)))
+    # stage_4:106-112 schema + the response_time_kind annotation column
+    # (so the in-process amortised figure can't be read as a latency)
     assert list(rec) == [
         "date", "MAPE", "r_squared", "max_residual", "mean_response_time",
-    ]  # stage_4:106-112 schema
+        "response_time_kind",
+    ]
+    assert rec["response_time_kind"] == "amortised-gpu-batch"
     assert 0 <= metrics["r_squared"] <= 1
     assert metrics["mean_response_time"] > 0
 
