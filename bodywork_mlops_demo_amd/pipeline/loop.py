@@ -130,7 +130,8 @@ def main(argv=None) -> None:
     p.add_argument("--days", type=int, default=30)
     p.add_argument("--rows", type=int, default=24 * 60)
     p.add_argument("--model", default="linear",
-                   help="linear | poly[<degree>] | mlp")
+                   help="linear | poly[<degree>] | mlp | mlp-fp8 "
+                        "(mlp with the MX-fp8 scoring forward)")
     p.add_argument("--device", default=None)
     p.add_argument("--start-date", default="2026-01-01")
     p.add_argument("--format", default="csv", choices=["csv", "npy"])
@@ -139,6 +140,11 @@ def main(argv=None) -> None:
                    choices=["always", "drift"])
     p.add_argument("--drift-threshold", type=float, default=1.5)
     args = p.parse_args(argv)
+    if args.model == "mlp-fp8":  # env flag so the deployed scorer opts in
+        import os
+
+        os.environ["BODYWORK_MLP_FP8"] = "1"
+        args.model = "mlp"
     results = run_loop(
         open_store(args.store), days=args.days, n_rows=args.rows,
         model_type=args.model, device=args.device,
