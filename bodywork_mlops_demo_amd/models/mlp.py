@@ -145,13 +145,21 @@ class GPUMLPRegressor:
             h1, m1 = ops.expand1d_bf16(xn, self.w1_bf, self.b1_bf, relu=True,
                                        emit_mask=True)
             h2, m2 = ops.linear_relu_mask_bf16(h1, self.W2w_bf, self.b2_bf)
+        elif self._use_fp8(xn.shape[0]):
+            # fused layer 1: h1 emitted directly as e4m3 (1 HBM byte per
+            # element instead of bf16-write + re-read + fp8-write), then
+            # the 2x-rate K=128 scaled-MFMA h2 GEMM
+            self._ensure_fp8_weights()
+            h1q = ops.expand1d_e4m3(xn, self.w1_bf, self.b1_bf,
+                                    self._e_h1)
+            h2 = ops.gemm_mx8_nt(h1q, self._e_h1, self._w2_q8, self._e_w2,
+                                 bias=self.b2, relu=True, out_fp32=False)
+            h1 = h1q
+            m1 = m2 = None
         else:
             h1 = ops.expand1d_bf16(xn, self.w1_bf, self.b1_bf, relu=True)
-            if self._use_fp8(h1.shape[0]):
-                h2 = self._h2_fp8(h1)
-            else:
-                h2 = ops.linear_bf16(h1, self.W2w_bf, bias=self.b2_bf,
-                                     relu=True)
+            h2 = ops.linear_bf16(h1, self.W2w_bf, bias=self.b2_bf,
+                                 relu=True)
             m1 = m2 = None
         yhat = ops.rowdot_bf16(h2, self.w3_bf, self.b3)
         return yhat, h1, h2, xn, m1, m2
@@ -163,18 +171,11 @@ class GPUMLPRegressor:
         return (self.fp8_scoring and self.device.type == "cuda"
                 and m % 256 == 0 and self.hidden % 256 == 0)
 
-    def _h2_fp8(self, h1: torch.Tensor) -> torch.Tensor:
-        """h2 = relu(h1 @ W2 + b2) through the 2x-rate K=128 scaled MFMA.
-
-        Exponents are STATIC: e_w2 and the weight-derived e_h1 freeze at
-        first quantisation (inside the scorer's warmup forwards, before
-        capture) — after that the whole path is tensor-in/tensor-out with
-        baked int args, so BatchedScorer can hipGraph-capture it (an
-        .item() sync here would abort capture)."""
-        self._ensure_fp8_weights()
-        h1q = ops.quantize_e4m3(h1, self._e_h1)
-        return ops.gemm_mx8_nt(h1q, self._e_h1, self._w2_q8, self._e_w2,
-                               bias=self.b2, relu=True, out_fp32=False)
+    # Exponents on the fp8 path are STATIC: e_w2 and the weight-derived
+    # e_h1 freeze at first quantisation (inside the scorer's warmup
+    # forwards, before capture) — after that the whole path is
+    # tensor-in/tensor-out with baked int args, so BatchedScorer can
+    # hipGraph-capture it (an .item() sync would abort capture).
 
     #: rows per forward chunk: bounds transient activations to
     #: 2 * chunk * H bf16 (= 16 GiB at H=4096) however large the batch

@@ -412,6 +412,26 @@ def quantize_e4m3(x: torch.Tensor, e: int) -> torch.Tensor:
     return reference.quantize_e4m3_cpu(x, int(e))
 
 
+def expand1d_e4m3(
+    x: torch.Tensor,
+    w: torch.Tensor,
+    b: torch.Tensor | None,
+    e: int,
+) -> torch.Tensor:
+    """h = relu(x*w + b) emitted DIRECTLY as e4m3 bytes (value = 2^e *
+    stored) — the MLP fp8 scoring forward's fused layer 1.  Unfused the
+    path costs 5 HBM bytes per element (bf16 write + re-read + fp8
+    write); fused it costs 1.  CPU oracle: quantize_e4m3_cpu of the
+    bf16-rounded activation."""
+    if x.device.type == "cuda":
+        core = _core(x.device)
+        return core.expand1d_e4m3(
+            x.contiguous(), w.contiguous(),
+            b.contiguous() if b is not None else None, int(e))
+    h = reference.expand1d_cpu(x, w, b, relu=True).float()
+    return reference.quantize_e4m3_cpu(h, int(e))
+
+
 def gemm_mx8_nt(
     a8: torch.Tensor,
     ea: int,

@@ -42,6 +42,8 @@ at::Tensor gemm_tn_bf16_hip(const at::Tensor& a, const at::Tensor& b,
                             bool out_fp32);
 // gemm_mx8.hip — MX-fp8 (e4m3) K=128 scaled-MFMA path
 at::Tensor quantize_e4m3_hip(const at::Tensor& x, int64_t e);
+at::Tensor expand1d_e4m3_hip(const at::Tensor& x, const at::Tensor& w,
+                             const c10::optional<at::Tensor>& b, int64_t e);
 at::Tensor gemm_mx8_nt_hip(const at::Tensor& a8, int64_t ea,
                            const at::Tensor& b8, int64_t eb,
                            const c10::optional<at::Tensor>& bias, bool relu,
@@ -78,6 +80,7 @@ TORCH_LIBRARY(bodywork_hip, m) {
         "(Tensor, Tensor)");
   m.def("gemm_tn_bf16(Tensor a, Tensor b, bool out_fp32) -> Tensor");
   m.def("quantize_e4m3(Tensor x, int e) -> Tensor");
+  m.def("expand1d_e4m3(Tensor x, Tensor w, Tensor? b, int e) -> Tensor");
   m.def("gemm_mx8_nt(Tensor a8, int ea, Tensor b8, int eb, Tensor? bias, "
         "bool relu, bool out_fp32) -> Tensor");
   m.def("adam_step(Tensor(a!) p, Tensor g, Tensor(b!) m, Tensor(c!) v, "
@@ -106,6 +109,7 @@ TORCH_LIBRARY_IMPL(bodywork_hip, CUDA, m) {
   m.impl("linear_relu_mask_bf16", linear_relu_mask_bf16_hip);
   m.impl("gemm_tn_bf16", gemm_tn_bf16_hip);
   m.impl("quantize_e4m3", quantize_e4m3_hip);
+  m.impl("expand1d_e4m3", expand1d_e4m3_hip);
   m.impl("gemm_mx8_nt", gemm_mx8_nt_hip);
   m.impl("adam_step", adam_step_hip);
   m.impl("batch_indices", batch_indices_hip);

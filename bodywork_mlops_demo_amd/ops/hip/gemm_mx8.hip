@@ -128,14 +128,18 @@ __global__ void gemm_mx8_nt_kernel(
     for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
   // one fragment = 32 consecutive k bytes = LDS chunks {2kg, 2kg+1}
-  // (XOR-swizzled), read as two b128s into an i32x8 MFMA operand
+  // (XOR-swizzled), read as two b128s straight into the halves of the
+  // i32x8 MFMA operand (a per-element insertelement assembly costs ~5
+  // v_mov per fragment — 304 v_movs per unrolled loop, measured in the
+  // .s — so the loads are written through a union to land in the
+  // operand's own register quadruples)
 #define MX_READ8(dst, rowbase, rowoff)                                      \
   do {                                                                      \
     const char* _rb = (rowbase) + (rowoff) * 128;                           \
-    mx_i32x4 _lo = *(const mx_i32x4*)(_rb + (((2 * kg) ^ swz) * 16));       \
-    mx_i32x4 _hi = *(const mx_i32x4*)(_rb + (((2 * kg + 1) ^ swz) * 16));   \
-    dst[0] = _lo[0]; dst[1] = _lo[1]; dst[2] = _lo[2]; dst[3] = _lo[3];     \
-    dst[4] = _hi[0]; dst[5] = _hi[1]; dst[6] = _hi[2]; dst[7] = _hi[3];     \
+    ((mx_i32x4*)&(dst))[0] =                                                \
+        *(const mx_i32x4*)(_rb + (((2 * kg) ^ swz) * 16));                  \
+    ((mx_i32x4*)&(dst))[1] =                                                \
+        *(const mx_i32x4*)(_rb + (((2 * kg + 1) ^ swz) * 16));              \
   } while (0)
 
 #define MX_AREAD(dst, buf, mfrag)                                           \
@@ -278,7 +282,75 @@ __global__ void quantize_e4m3_kernel(const T* __restrict__ x,
   }
 }
 
+// ---- fused rank-1 expand -> e4m3 ------------------------------------------
+//
+// The MLP fp8 scoring forward's layer 1: h1 = relu(x*w + b) emitted
+// DIRECTLY as e4m3 bytes (value = 2^e * stored).  Unfused, the path
+// writes h1 as bf16 (2 B/elem), re-reads it (2 B) and writes the fp8
+// copy (1 B) — 5 bytes of HBM per element; this kernel writes 1.
+template <bool HAS_BIAS>
+__global__ void expand1d_e4m3_kernel(const float* __restrict__ x,
+                                     const bf16_t* __restrict__ w,
+                                     const bf16_t* __restrict__ b,
+                                     unsigned char* __restrict__ out,
+                                     long long n, int h8 /* H/8 */,
+                                     float inv_scale) {
+  const long long total = (long long)n * h8;
+  const long long stride = (long long)gridDim.x * blockDim.x;
+  for (long long idx = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+       idx < total; idx += stride) {
+    const long long row = idx / h8;
+    const int c8 = (int)(idx % h8);
+    float xv = x[row];
+    float wv[8], bv[8];
+    bf16x8_to_f32(load_bf16x8(w + c8 * 8), wv);
+    if (HAS_BIAS) bf16x8_to_f32(load_bf16x8(b + c8 * 8), bv);
+    unsigned long long packed = 0;
+#pragma unroll
+    for (int p = 0; p < 4; ++p) {
+      float a0 = HAS_BIAS ? fmaf(xv, wv[2 * p], bv[2 * p]) : xv * wv[2 * p];
+      float a1 = HAS_BIAS ? fmaf(xv, wv[2 * p + 1], bv[2 * p + 1])
+                          : xv * wv[2 * p + 1];
+      a0 = fmaxf(a0, 0.0f) * inv_scale;  // relu fused
+      a1 = fmaxf(a1, 0.0f) * inv_scale;
+      unsigned int pk = (unsigned int)
+          __builtin_amdgcn_cvt_pk_fp8_f32(a0, a1, 0, false);
+      packed |= (unsigned long long)(pk & 0xFFFFu) << (16 * p);
+    }
+    *(unsigned long long*)(out + row * (long long)h8 * 8 + c8 * 8) = packed;
+  }
+}
+
 // ---- host wrappers --------------------------------------------------------
+
+at::Tensor expand1d_e4m3_hip(const at::Tensor& x, const at::Tensor& w,
+                             const c10::optional<at::Tensor>& b,
+                             int64_t e) {
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kFloat);
+  TORCH_CHECK(w.scalar_type() == at::kBFloat16);
+  const long long n = x.numel();
+  const long long H = w.numel();
+  TORCH_CHECK(H % 8 == 0, "expand1d_e4m3: H must be a multiple of 8");
+  auto out = at::empty({n, H}, w.options().dtype(at::kByte));
+  const bool has_bias = b.has_value();
+  auto stream = at::cuda::getCurrentCUDAStream();
+  long long total = n * (H / 8);
+  int grid = (int)std::min<long long>((total + 255) / 256, 2048);
+  float inv_scale = ldexpf(1.0f, (int)-e);
+  const bf16_t* wp = (const bf16_t*)w.data_ptr();
+  const bf16_t* bp = has_bias ? (const bf16_t*)b->data_ptr() : nullptr;
+  if (has_bias)
+    hipLaunchKernelGGL((expand1d_e4m3_kernel<true>), dim3(grid), dim3(256),
+                       0, stream, x.data_ptr<float>(), wp, bp,
+                       out.data_ptr<unsigned char>(), n, (int)(H / 8),
+                       inv_scale);
+  else
+    hipLaunchKernelGGL((expand1d_e4m3_kernel<false>), dim3(grid), dim3(256),
+                       0, stream, x.data_ptr<float>(), wp, bp,
+                       out.data_ptr<unsigned char>(), n, (int)(H / 8),
+                       inv_scale);
+  return out;
+}
 
 at::Tensor quantize_e4m3_hip(const at::Tensor& x, int64_t e) {
   TORCH_CHECK(x.is_cuda() && x.is_contiguous(), "quantize_e4m3: cuda contig");

@@ -144,3 +144,26 @@ def test_mlp_fp8_scorer_capture_and_hot_redeploy():
     # fp8 requantisation (same tensors, same baked exponents)
     torch.testing.assert_close(got2, want2, rtol=5e-2, atol=5e-2)
     assert (got2 - got).abs().max().item() > 1e-3  # actually changed
+
+
+def test_expand1d_e4m3_fused_matches_two_pass():
+    """Fused layer-1 expand->e4m3 == quantize(expand1d) within one ulp
+    of the e4m3 grid (the fused kernel skips the intermediate bf16
+    rounding, so values may land one grid point apart at bf16-rounding
+    boundaries)."""
+    g = torch.Generator(device=DEV).manual_seed(12)
+    x = ((torch.rand(512, generator=g, device=DEV) * 100) - 50.0) / 28.9
+    w = torch.randn(512, generator=g, device=DEV).bfloat16()
+    b = torch.randn(512, generator=g, device=DEV).bfloat16()
+    e = 3
+    fused = ops.expand1d_e4m3(x, w, b, e)
+    h1 = ops.expand1d_bf16(x, w, b, relu=True)
+    twopass = ops.quantize_e4m3(h1, e)
+    df = ops.reference.e4m3_decode_cpu(fused.cpu(), e)
+    dt = ops.reference.e4m3_decode_cpu(twopass.cpu(), e)
+    # identical up to bf16-rounding-boundary grid neighbours
+    rel = (df - dt).abs() / dt.abs().clamp_min(2.0 ** (e - 6))
+    assert rel.max().item() < 0.15, rel.max().item()
+    assert (df - dt).abs().mean().item() < 2.0 ** (e - 7)
+    # relu fused: no negatives
+    assert (df >= 0).all()
