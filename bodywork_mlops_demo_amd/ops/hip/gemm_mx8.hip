@@ -112,7 +112,14 @@ __global__ void gemm_mx8_nt_kernel(
   const int wn = wave & 3;   // 0..3 (64 C-cols per wave)
   const int fl = lane & 15;
   const int kg = lane >> 4;
-  const int swz = fl & 7;
+  // XOR swizzle at 32-BYTE granularity — the fp8 fragment size.  The
+  // bf16 kernel swizzles 16-B chunks (= its fragment size) and measures
+  // ZERO bank conflicts; reading the fp8 fragment as two stride-2
+  // 16-B chunks under that swizzle measured 4 conflict cycles per
+  // ds_read_b128 (SQ_LDS_BANK_CONFLICT 6.7e8 over 1.7e8 reads,
+  // profiles/r02_mx8_pmc.md), so the swizzle unit here matches the
+  // 32-B fragment: chunk32 c of row r lives at (c ^ (r & 3)) * 32.
+  const int swz = fl & 3;
   const long long nt = K / MX_BK;
 
 #define MX_ASLOT(buf, half) (lds + ((buf) * 2 + (half)) * MX_HTB)
@@ -135,11 +142,9 @@ __global__ void gemm_mx8_nt_kernel(
   // operand's own register quadruples)
 #define MX_READ8(dst, rowbase, rowoff)                                      \
   do {                                                                      \
-    const char* _rb = (rowbase) + (rowoff) * 128;                           \
-    ((mx_i32x4*)&(dst))[0] =                                                \
-        *(const mx_i32x4*)(_rb + (((2 * kg) ^ swz) * 16));                  \
-    ((mx_i32x4*)&(dst))[1] =                                                \
-        *(const mx_i32x4*)(_rb + (((2 * kg + 1) ^ swz) * 16));              \
+    const char* _rb = (rowbase) + (rowoff) * 128 + ((kg ^ swz) * 32);       \
+    ((mx_i32x4*)&(dst))[0] = *(const mx_i32x4*)(_rb);                       \
+    ((mx_i32x4*)&(dst))[1] = *(const mx_i32x4*)(_rb + 16);                  \
   } while (0)
 
 #define MX_AREAD(dst, buf, mfrag)                                           \
@@ -150,14 +155,17 @@ __global__ void gemm_mx8_nt_kernel(
   mx_i32x8 a_q[2];     // this phase's two A fragments (mfrag pair)
   mx_i32x8 b_t[4];     // current tile's B fragments [nfrag]
 
-  // per-lane staging byte offset (tile-invariant): chunk ci covers
-  // row = ci>>3, 16-B column chunk (ci&7)^XOR(row) of the 128-B row
+  // per-lane staging byte offset (tile-invariant): thread ci stages the
+  // 16-B half `ci&1` of 32-B chunk `(ci&7)>>1` of row `ci>>3`; the
+  // 32-B chunk lands at (chunk ^ (row & 3)) * 32 in LDS, halves stay
+  // adjacent, so the glds SOURCE address carries the swizzle exactly as
+  // in the bf16 kernel
   int stg_off;
   {
     int ci = (int)threadIdx.x;
     int row = ci >> 3;
-    int sc = (ci & 7) ^ (row & 7);
-    stg_off = (int)(row * K + sc * 16);
+    int c32 = ((ci & 7) >> 1) ^ (row & 3);
+    stg_off = (int)(row * K + (c32 * 32 + (ci & 1) * 16));
   }
   const char* Ah0 = (const char*)(A + m0 * K);
   const char* Ah1 = (const char*)(A + (m0 + 128) * K);

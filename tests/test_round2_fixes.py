@@ -298,3 +298,42 @@ def test_runner_front_proxy_single_url_two_replicas(tmp_path):
     finally:
         runner.teardown()
     assert not runner.services
+
+
+@pytest.mark.timeout(300)
+def test_run_cycle_http_serving_and_skip_train(tmp_path):
+    """serving='http' inside run_cycle: the deploy phase starts one real
+    uvicorn replica, reloads it on later cycles, and a skip_train cycle
+    (drift policy) reuses the live replica without retraining."""
+    import torch
+
+    from bodywork_mlops_demo_amd.pipeline.cycle import CycleState, run_cycle
+
+    store = LocalStore(str(tmp_path / "store"))
+    state = CycleState("cpu", date(2026, 9, 1), history_days=1)
+    cache: dict = {}
+    r1 = run_cycle(state, store, 1500, model_type="linear",
+                   serving="http", http_port=5611, scorer_cache=cache,
+                   use_graphs=False)
+    try:
+        assert "http_replica" in cache
+        pid = cache["http_replica"].proc.pid
+        assert r1["online"]["response_time_kind"] == "http-binary"
+        r2 = run_cycle(state, store, 1500, model_type="linear",
+                       serving="http", http_port=5611, scorer_cache=cache,
+                       use_graphs=False)
+        assert cache["http_replica"].proc.pid == pid  # reloaded, not respawned
+        # drift-policy skip: no retrain, same replica keeps serving
+        r3 = run_cycle(state, store, 1500, model_type="linear",
+                       serving="http", http_port=5611, scorer_cache=cache,
+                       use_graphs=False, skip_train=True)
+        assert r3["timings"]["train_s"] == 0.0
+        assert r3["online"]["r_squared"] > 0.5
+    finally:
+        cache["http_replica"].stop()
+    state.drain_io()
+    # three test-metrics artefacts, all marked over-the-wire
+    keys = store.list_keys(contract.TEST_METRICS_PREFIX)
+    assert len(keys) == 3
+    for k in keys:
+        assert store.get_metrics_csv(k)["response_time_kind"] == "http-binary"
