@@ -33,13 +33,16 @@ import torch
 def main() -> None:
     p = argparse.ArgumentParser()
     p.add_argument("--gpus", type=int, default=1)
-    p.add_argument("--steps", type=int, default=4)
+    # default timed region is ~20 cycles so external GPU-utilisation
+    # samplers (the driver polls rocm-smi) can catch the run
+    p.add_argument("--steps", type=int, default=20)
     p.add_argument("--warmup", type=int, default=2)
     p.add_argument("--rows", type=int, default=10_000_000,
                    help="synthetic rows per GPU per day (BASELINE config 2)")
     p.add_argument("--model", default="linear",
-                   help="linear | poly[<degree>] | mlp "
-                        "(mlp = BASELINE config 5, 4096-d MFMA GEMM path)")
+                   help="linear | poly[<degree>] | mlp | mlp-fp8 "
+                        "(mlp = BASELINE config 5, 4096-d MFMA GEMM path; "
+                        "mlp-fp8 adds the 2x-rate MX-fp8 scoring forward)")
     p.add_argument("--mlp-steps", type=int, default=50)
     p.add_argument("--mlp-batch", type=int, default=65536)
     p.add_argument("--no-graphs", action="store_true")

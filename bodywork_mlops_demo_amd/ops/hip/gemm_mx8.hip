@@ -112,13 +112,14 @@ __global__ void gemm_mx8_nt_kernel(
   const int wn = wave & 3;   // 0..3 (64 C-cols per wave)
   const int fl = lane & 15;
   const int kg = lane >> 4;
-  // XOR swizzle at 32-BYTE granularity — the fp8 fragment size.  The
-  // bf16 kernel swizzles 16-B chunks (= its fragment size) and measures
-  // ZERO bank conflicts; reading the fp8 fragment as two stride-2
-  // 16-B chunks under that swizzle measured 4 conflict cycles per
-  // ds_read_b128 (SQ_LDS_BANK_CONFLICT 6.7e8 over 1.7e8 reads,
-  // profiles/r02_mx8_pmc.md), so the swizzle unit here matches the
-  // 32-B fragment: chunk32 c of row r lives at (c ^ (r & 3)) * 32.
+  // XOR swizzle at 32-BYTE granularity — the fp8 fragment size, so a
+  // fragment is two ADJACENT b128 reads (chunk32 c of row r lives at
+  // (c ^ (r & 3)) * 32).  Measured equivalent to the 16-B-chunk
+  // stride-2 variant: SQ_LDS_BANK_CONFLICT reads exactly 4 events per
+  // v_mfma_scale under BOTH layouts (6.711e8 = 4x the MFMA count,
+  // unchanged by the swizzle change) — an artifact of the scaled-MFMA
+  // instruction, not real ds_read serialisation
+  // (profiles/r02_mx8_pmc.md).
   const int swz = fl & 3;
   const long long nt = K / MX_BK;
 

@@ -51,6 +51,28 @@ timeout 500 python -m bodywork_mlops_demo_amd loop --days 30 --rows 1440 \
 python -m bodywork_mlops_demo_amd analytics --store "$OUT/loopstore" \
   2>/dev/null | tail -2 | tee -a "$OUT/summary.txt"
 
+log "r2: linear 10M, GROWING read-all history"
+timeout 400 python bench.py --rows 10000000 --steps 10 --warmup 2 \
+  --history all > "$OUT/bench_linear_histall.json" 2>/dev/null
+tail -1 "$OUT/bench_linear_histall.json" | tee -a "$OUT/summary.txt"
+
+log "r2: linear 10M over the HTTP binary wire (uvicorn replica in-cycle)"
+timeout 500 python bench.py --rows 10000000 --steps 5 --warmup 2 \
+  --serving http > "$OUT/bench_linear_http.json" 2>/dev/null
+tail -1 "$OUT/bench_linear_http.json" | tee -a "$OUT/summary.txt"
+
+log "r2: MLP 125M bf16 vs MX-fp8 scoring A/B"
+timeout 700 python bench.py --model mlp --rows 125000000 --steps 2 \
+  --warmup 1 > "$OUT/bench_mlp_125m_bf16.json" 2>/dev/null
+tail -1 "$OUT/bench_mlp_125m_bf16.json" | tee -a "$OUT/summary.txt"
+timeout 700 python bench.py --model mlp-fp8 --rows 125000000 --steps 2 \
+  --warmup 1 > "$OUT/bench_mlp_125m_fp8.json" 2>/dev/null
+tail -1 "$OUT/bench_mlp_125m_fp8.json" | tee -a "$OUT/summary.txt"
+
+log "r2: MX-fp8 GEMM check + A/B microbench"
+timeout 500 python tools/probe_mx8.py --check --bench 2>/dev/null \
+  | tee -a "$OUT/summary.txt"
+
 log "GEMM microbench"
 python - 2>/dev/null <<'EOF' | tee -a "$OUT/summary.txt"
 import torch, time
