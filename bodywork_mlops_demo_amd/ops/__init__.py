@@ -461,6 +461,33 @@ def gemm_mx8_nt(
     return reference.gemm_mx8_nt_cpu(a8, ea, b8, eb, bias, relu, out_fp32)
 
 
+def gemm_mx8_relu_dot(
+    a8: torch.Tensor,
+    ea: int,
+    b8: torch.Tensor,
+    eb: int,
+    b2: torch.Tensor,
+    w3: torch.Tensor,
+) -> torch.Tensor:
+    """y[M] = Σ_col relu(2^(ea+eb)·(a8·b8ᵀ) + b2) * w3 — the MLP scoring
+    forward's hot GEMM and the rowdot head fused into ONE kernel.
+
+    The [M,N] activation tensor is never written to HBM (saves its bf16
+    write + the rowdot re-read ≈ 16 GB per 1M rows at N=4096): each
+    wave butterfly-reduces its 64-column strip in registers and fl==0
+    lanes accumulate per-row partials with fp32 atomics.  Caller adds
+    the scalar output bias.  CPU oracle: decode + matmul + relu + dot.
+    """
+    if a8.device.type == "cuda":
+        core = _core(a8.device)
+        return core.gemm_mx8_relu_dot(a8.contiguous(), int(ea),
+                                      b8.contiguous(), int(eb),
+                                      b2.contiguous(), w3.contiguous())
+    h2 = reference.gemm_mx8_nt_cpu(a8, ea, b8, eb, bias=b2, relu=True,
+                                   out_fp32=True)
+    return h2 @ w3.float()
+
+
 def transpose_bf16(src: torch.Tensor) -> torch.Tensor:
     """[C,R] bf16 = [R,C] bf16 transposed (64x64 LDS tiles, 16-B I/O)."""
     if src.device.type == "cuda":

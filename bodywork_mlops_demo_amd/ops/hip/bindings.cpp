@@ -48,6 +48,9 @@ at::Tensor gemm_mx8_nt_hip(const at::Tensor& a8, int64_t ea,
                            const at::Tensor& b8, int64_t eb,
                            const c10::optional<at::Tensor>& bias, bool relu,
                            bool out_fp32);
+at::Tensor gemm_mx8_relu_dot_hip(const at::Tensor& a8, int64_t ea,
+                                 const at::Tensor& b8, int64_t eb,
+                                 const at::Tensor& b2, const at::Tensor& w3);
 // optim.hip
 void adam_step_hip(at::Tensor p, const at::Tensor& g, at::Tensor m,
                    at::Tensor v, const c10::optional<at::Tensor>& p_bf16,
@@ -83,6 +86,8 @@ TORCH_LIBRARY(bodywork_hip, m) {
   m.def("expand1d_e4m3(Tensor x, Tensor w, Tensor? b, int e) -> Tensor");
   m.def("gemm_mx8_nt(Tensor a8, int ea, Tensor b8, int eb, Tensor? bias, "
         "bool relu, bool out_fp32) -> Tensor");
+  m.def("gemm_mx8_relu_dot(Tensor a8, int ea, Tensor b8, int eb, "
+        "Tensor b2, Tensor w3) -> Tensor");
   m.def("adam_step(Tensor(a!) p, Tensor g, Tensor(b!) m, Tensor(c!) v, "
         "Tensor(d!)? p_bf16, float lr, float beta1, float beta2, float eps, "
         "int t, Tensor? bc) -> ()");
@@ -111,6 +116,7 @@ TORCH_LIBRARY_IMPL(bodywork_hip, CUDA, m) {
   m.impl("quantize_e4m3", quantize_e4m3_hip);
   m.impl("expand1d_e4m3", expand1d_e4m3_hip);
   m.impl("gemm_mx8_nt", gemm_mx8_nt_hip);
+  m.impl("gemm_mx8_relu_dot", gemm_mx8_relu_dot_hip);
   m.impl("adam_step", adam_step_hip);
   m.impl("batch_indices", batch_indices_hip);
   m.impl("transpose_to_bf16", transpose_to_bf16_hip);

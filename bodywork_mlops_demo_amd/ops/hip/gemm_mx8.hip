@@ -48,6 +48,7 @@ typedef __attribute__((ext_vector_type(4))) float mx_f32x4;
 
 #define MX_EPI_NONE 0
 #define MX_EPI_BIAS_RELU 1
+#define MX_EPI_RELU_DOT 2  // y[row] += sum_col relu(acc+bias)*w3[col]
 
 __device__ __forceinline__ void mx_glds16(const void* gsrc, void* lds_dst) {
   __builtin_amdgcn_global_load_lds(
@@ -67,8 +68,44 @@ __device__ __forceinline__ void mx_stage_half(char* __restrict__ slot,
 template <int EPI, bool HAS_BIAS, bool OUT_FP32>
 __device__ __forceinline__ void mx_epilogue(
     mx_f32x4 (&acc)[4][4], const float* __restrict__ bias,
-    void* __restrict__ C, long long M, long long N, long long m0,
-    long long n0, int wm, int wn, int fl, int kg) {
+    const float* __restrict__ w3, void* __restrict__ C, long long M,
+    long long N, long long m0, long long n0, int wm, int wn, int fl,
+    int kg) {
+  if (EPI == MX_EPI_RELU_DOT) {
+    // fused scoring head: y[row] += sum_col relu(acc + b2[col]) *
+    // w3[col] — the h2 activation tensor is never materialised (saves
+    // its [M,N] bf16 write + the rowdot re-read, ~16 GB of HBM per 1M
+    // rows at N=4096).  Per (i,r) row: 16 lanes (fl) x 4 j-frags hold
+    // the wave's 64-column strip; butterfly-reduce across fl, then the
+    // fl==0 lane of each kg row atomically adds its wave's partial
+    // (4 wn-waves per block and N/256 blocks accumulate per row; the
+    // caller zero-fills y and adds the b3 offset).
+    float w3v[4], b2v[4];
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      long long col = n0 + wn * 64 + j * 16 + fl;
+      w3v[j] = w3[col];
+      b2v[j] = HAS_BIAS ? bias[col] : 0.0f;
+    }
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        float part = 0.0f;
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          part = fmaf(fmaxf(acc[i][j][r] + b2v[j], 0.0f), w3v[j], part);
+#pragma unroll
+        for (int m = 1; m < 16; m <<= 1)
+          part += __shfl_xor(part, m, 64);  // reduce across fl (lane&15)
+        if (fl == 0) {
+          long long row = m0 + wm * 64 + i * 16 + kg * 4 + r;
+          atomicAdd((float*)C + row, part);
+        }
+      }
+    }
+    return;
+  }
 #pragma unroll
   for (int i = 0; i < 4; ++i) {
 #pragma unroll
@@ -101,7 +138,8 @@ __launch_bounds__(MX_THREADS)
 __global__ void gemm_mx8_nt_kernel(
     const unsigned char* __restrict__ A,  // [M,K] e4m3, row-major
     const unsigned char* __restrict__ B,  // [N,K] e4m3, row-major
-    const float* __restrict__ bias, void* __restrict__ C, long long M,
+    const float* __restrict__ bias, const float* __restrict__ w3,
+    void* __restrict__ C, long long M,
     long long N, long long K, int sa, int sb) {  // sa/sb: E8M0 bytes
   __shared__ char lds[8 * MX_HTB];  // ONE __shared__ object (guide trap (a))
   const long long m0 = (long long)blockIdx.y * MX_BM;
@@ -252,8 +290,8 @@ __global__ void gemm_mx8_nt_kernel(
 #undef MX_ASLOT
 #undef MX_BSLOT
 
-  mx_epilogue<EPI, HAS_BIAS, OUT_FP32>(acc, bias, C, M, N, m0, n0, wm, wn,
-                                       fl, kg);
+  mx_epilogue<EPI, HAS_BIAS, OUT_FP32>(acc, bias, w3, C, M, N, m0, n0, wm,
+                                       wn, fl, kg);
 }
 
 // ---- per-tensor e4m3 quantisation ----------------------------------------
@@ -425,7 +463,7 @@ at::Tensor gemm_mx8_nt_hip(const at::Tensor& a8, int64_t ea,
   const unsigned char* bp = b8.data_ptr<unsigned char>();
 #define LMX(EPI_, HB_, OF_)                                                 \
   hipLaunchKernelGGL((gemm_mx8_nt_kernel<EPI_, HB_, OF_>), grid,            \
-                     dim3(MX_THREADS), 0, stream, ap, bp, bias_p,           \
+                     dim3(MX_THREADS), 0, stream, ap, bp, bias_p, nullptr, \
                      C.data_ptr(), M, N, K, sa, sb)
   if (relu) {
     if (has_bias) { if (out_fp32) LMX(MX_EPI_BIAS_RELU, true, true);
@@ -439,4 +477,38 @@ at::Tensor gemm_mx8_nt_hip(const at::Tensor& a8, int64_t ea,
   }
 #undef LMX
   return C;
+}
+
+// y[M] = sum_col relu(a8.b8^T dequant + b2) * w3  — the MLP scoring
+// forward's h2 GEMM and rowdot head in ONE kernel: the [M,N] activation
+// tensor is never written to HBM.  Caller adds the scalar b3.
+at::Tensor gemm_mx8_relu_dot_hip(const at::Tensor& a8, int64_t ea,
+                                 const at::Tensor& b8, int64_t eb,
+                                 const at::Tensor& b2,
+                                 const at::Tensor& w3) {
+  TORCH_CHECK(a8.is_cuda() && b8.is_cuda() && a8.dim() == 2 &&
+                  b8.dim() == 2 && a8.size(1) == b8.size(1),
+              "gemm_mx8_relu_dot: [M,K]x[N,K] cuda");
+  TORCH_CHECK(a8.scalar_type() == torch::kUInt8 &&
+              b8.scalar_type() == torch::kUInt8);
+  TORCH_CHECK(a8.is_contiguous() && b8.is_contiguous());
+  long long M = a8.size(0), K = a8.size(1), N = b8.size(0);
+  TORCH_CHECK(M % MX_BM == 0 && N % MX_BN == 0 && K % MX_BK == 0,
+              "gemm_mx8_relu_dot requires M%256==0, N%256==0, K%128==0");
+  TORCH_CHECK(b2.is_cuda() && b2.scalar_type() == torch::kFloat &&
+                  b2.numel() == N, "b2: fp32 [N]");
+  TORCH_CHECK(w3.is_cuda() && w3.scalar_type() == torch::kFloat &&
+                  w3.numel() == N, "w3: fp32 [N]");
+  int sa = (int)(ea + 127), sb = (int)(eb + 127);
+  TORCH_CHECK(0 <= sa && sa <= 254 && 0 <= sb && sb <= 254,
+              "exponent out of E8M0 range");
+  auto y = torch::zeros({M}, a8.options().dtype(torch::kFloat));
+  dim3 grid((unsigned)(N / MX_BN), (unsigned)(M / MX_BM));
+  auto stream = at::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL((gemm_mx8_nt_kernel<MX_EPI_RELU_DOT, true, true>),
+                     grid, dim3(MX_THREADS), 0, stream,
+                     a8.data_ptr<unsigned char>(),
+                     b8.data_ptr<unsigned char>(), b2.data_ptr<float>(),
+                     w3.data_ptr<float>(), y.data_ptr(), M, N, K, sa, sb);
+  return y;
 }

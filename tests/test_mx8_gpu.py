@@ -167,3 +167,51 @@ def test_expand1d_e4m3_fused_matches_two_pass():
     assert (df - dt).abs().mean().item() < 2.0 ** (e - 7)
     # relu fused: no negatives
     assert (df >= 0).all()
+
+
+def test_gemm_mx8_relu_dot_fused_head():
+    """Fused GEMM + relu + rowdot head == unfused (gemm then torch dot)
+    within fp32 accumulation-order tolerance, on exact-integer operands
+    EXACTLY (integer dot of integer relu outputs is exact in fp32)."""
+    a = _randint(512, 512, 21)
+    b = _randint(512, 512, 22)
+    a8 = ops.quantize_e4m3(a, 0)
+    b8 = ops.quantize_e4m3(b, 0)
+    b2 = _randint(1, 512, 23)[0]
+    w3 = _randint(1, 512, 24)[0]
+    y = ops.gemm_mx8_relu_dot(a8, 0, b8, 0, b2, w3)
+    h2 = torch.relu(a @ b.t() + b2)
+    want = h2 @ w3
+    # integer data: atomic accumulation of integer partials is exact
+    assert torch.equal(y, want), (y - want).abs().max().item()
+
+    # random data: bounded by accumulation-order differences only
+    g = torch.Generator(device=DEV).manual_seed(25)
+    a = torch.randn(256, 4096, generator=g, device=DEV)
+    b = torch.randn(512, 4096, generator=g, device=DEV)
+    ea = ops.e4m3_exponent(a.abs().max().item())
+    eb = ops.e4m3_exponent(b.abs().max().item())
+    a8 = ops.quantize_e4m3(a, ea)
+    b8 = ops.quantize_e4m3(b, eb)
+    b2 = torch.randn(512, generator=g, device=DEV)
+    w3 = torch.randn(512, generator=g, device=DEV)
+    y = ops.gemm_mx8_relu_dot(a8, ea, b8, eb, b2, w3)
+    h2 = ops.gemm_mx8_nt(a8, ea, b8, eb, bias=b2, relu=True, out_fp32=True)
+    want = h2 @ w3
+    torch.testing.assert_close(y, want, rtol=1e-3, atol=1e-2)
+
+
+def test_mlp_fp8_fused_head_predict_matches():
+    """predict through the 2-kernel fp8 forward (fused expand + fused
+    GEMM/dot head) vs the bf16 3-kernel forward on a trained model."""
+    from bodywork_mlops_demo_amd.models import GPUMLPRegressor
+
+    y, X = ops.datagen(100_000, 120, 9, device=DEV)
+    m_bf = GPUMLPRegressor(hidden=512, device=DEV, seed=4)
+    m_bf.fit(X, y, steps=20, batch_size=16384)
+    m_f8 = GPUMLPRegressor(hidden=512, device=DEV, seed=4,
+                           fp8_scoring=True)
+    assert m_f8.copy_weights_from(m_bf)
+    p_bf = m_bf.predict(X[:2048])
+    p_f8 = m_f8.predict(X[:2048])
+    assert ((p_f8 - p_bf).abs().mean() / y.abs().mean()).item() < 0.01

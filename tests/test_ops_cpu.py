@@ -204,3 +204,19 @@ def test_expand1d_e4m3_cpu_oracle():
     want = torch.relu(torch.outer(x, w.float()) + b.float())
     assert q.shape == (2, 4)
     assert (dec - want).abs().max().item() < 0.07  # e4m3 grid error
+
+
+def test_gemm_mx8_relu_dot_cpu_oracle():
+    import torch
+
+    from bodywork_mlops_demo_amd import ops
+
+    g = torch.Generator().manual_seed(31)
+    a = torch.randint(-4, 5, (8, 64), generator=g).float()
+    b = torch.randint(-4, 5, (16, 64), generator=g).float()
+    b2 = torch.randint(-2, 3, (16,), generator=g).float()
+    w3 = torch.randint(-2, 3, (16,), generator=g).float()
+    y = ops.gemm_mx8_relu_dot(ops.quantize_e4m3(a, 0), 0,
+                              ops.quantize_e4m3(b, 0), 0, b2, w3)
+    want = torch.relu(a @ b.t() + b2) @ w3
+    assert torch.allclose(y.float(), want)
