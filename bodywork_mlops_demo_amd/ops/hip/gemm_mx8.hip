@@ -140,10 +140,28 @@ __global__ void gemm_mx8_nt_kernel(
     const unsigned char* __restrict__ B,  // [N,K] e4m3, row-major
     const float* __restrict__ bias, const float* __restrict__ w3,
     void* __restrict__ C, long long M,
-    long long N, long long K, int sa, int sb) {  // sa/sb: E8M0 bytes
+    long long N, long long K, int sa, int sb,
+    int super) {  // sa/sb: E8M0 bytes; super: supertile height in y-blocks
   __shared__ char lds[8 * MX_HTB];  // ONE __shared__ object (guide trap (a))
-  const long long m0 = (long long)blockIdx.y * MX_BM;
-  const long long n0 = (long long)blockIdx.x * MX_BN;
+  // Supertile blockIdx remap (super > 1): consecutive dispatch ids walk
+  // an (x-cols x super-rows) PATCH instead of a full grid-x row, so the
+  // ~256 concurrently-resident workgroups (1 block/CU) touch
+  // super x 1 MB of A + (concurrent/super) x 1 MB of B instead of
+  // re-streaming all of B per grid-y row.  At M=65536, N=K=4096 the
+  // x-major order reads ~8 GB of tiles per launch (= 7.5 TB/s at the
+  // measured rate — HBM-saturated); a square patch cuts that ~4-8x.
+  int bx = blockIdx.x, by = blockIdx.y;
+  if (super > 1) {
+    int linear = blockIdx.x + blockIdx.y * gridDim.x;
+    int band_sz = super * gridDim.x;
+    int band = linear / band_sz;
+    int in_band = linear % band_sz;
+    int gy = min(super, (int)gridDim.y - band * super);
+    bx = in_band / gy;
+    by = band * super + in_band % gy;
+  }
+  const long long m0 = (long long)by * MX_BM;
+  const long long n0 = (long long)bx * MX_BN;
   const int wave = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
   const int wm = wave >> 2;  // 0..3 (64 C-rows per wave)
@@ -370,6 +388,17 @@ __global__ void expand1d_e4m3_kernel(const float* __restrict__ x,
 
 // ---- host wrappers --------------------------------------------------------
 
+// supertile height (y-blocks) for the blockIdx remap; latched once.
+// 16 -> a 16x16-block patch (~32 MB of A+B tiles) per 256 concurrent
+// workgroups.  BODYWORK_MX_SUPER=1 disables, other values A/B.
+static int mx_supertile() {
+  static int v = [] {
+    const char* e = getenv("BODYWORK_MX_SUPER");
+    return e ? atoi(e) : 16;
+  }();
+  return v;
+}
+
 at::Tensor expand1d_e4m3_hip(const at::Tensor& x, const at::Tensor& w,
                              const c10::optional<at::Tensor>& b,
                              int64_t e) {
@@ -464,7 +493,7 @@ at::Tensor gemm_mx8_nt_hip(const at::Tensor& a8, int64_t ea,
 #define LMX(EPI_, HB_, OF_)                                                 \
   hipLaunchKernelGGL((gemm_mx8_nt_kernel<EPI_, HB_, OF_>), grid,            \
                      dim3(MX_THREADS), 0, stream, ap, bp, bias_p, nullptr, \
-                     C.data_ptr(), M, N, K, sa, sb)
+                     C.data_ptr(), M, N, K, sa, sb, mx_supertile())
   if (relu) {
     if (has_bias) { if (out_fp32) LMX(MX_EPI_BIAS_RELU, true, true);
                     else          LMX(MX_EPI_BIAS_RELU, true, false); }
@@ -509,6 +538,7 @@ at::Tensor gemm_mx8_relu_dot_hip(const at::Tensor& a8, int64_t ea,
                      grid, dim3(MX_THREADS), 0, stream,
                      a8.data_ptr<unsigned char>(),
                      b8.data_ptr<unsigned char>(), b2.data_ptr<float>(),
-                     w3.data_ptr<float>(), y.data_ptr(), M, N, K, sa, sb);
+                     w3.data_ptr<float>(), y.data_ptr(), M, N, K, sa, sb,
+                     mx_supertile());
   return y;
 }
