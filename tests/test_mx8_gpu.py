@@ -215,3 +215,20 @@ def test_mlp_fp8_fused_head_predict_matches():
     p_bf = m_bf.predict(X[:2048])
     p_f8 = m_f8.predict(X[:2048])
     assert ((p_f8 - p_bf).abs().mean() / y.abs().mean()).item() < 0.01
+
+
+def test_mx8_supertile_clamp_path():
+    """Supertile blockIdx remap with gridDim.y not divisible by the
+    supertile height (M=768 -> 3 y-blocks, super=16 -> clamped band):
+    results must stay exact — a wrong remap drops or doubles tiles."""
+    a = _randint(768, 512, 41)
+    b = _randint(512, 512, 42)
+    got = ops.gemm_mx8_nt(ops.quantize_e4m3(a, 0), 0,
+                          ops.quantize_e4m3(b, 0), 0, out_fp32=True)
+    assert torch.equal(got, a @ b.t())
+    # and through the fused head
+    b2 = _randint(1, 512, 43)[0]
+    w3 = _randint(1, 512, 44)[0]
+    y = ops.gemm_mx8_relu_dot(ops.quantize_e4m3(a, 0), 0,
+                              ops.quantize_e4m3(b, 0), 0, b2, w3)
+    assert torch.equal(y, torch.relu(a @ b.t() + b2) @ w3)
