@@ -37,6 +37,18 @@ assert err < 2e-2, float(err)
 # fused reductions
 stats = ops.linreg_stats(X, y)
 assert stats[0].item() == X.numel()
+# MX-fp8 path (scaled MFMA + quantiser + fused head): exact under
+# serialized launches too
+ai = torch.randint(-8, 9, (256, 512), device=dev).float()
+bi = torch.randint(-8, 9, (256, 512), device=dev).float()
+a8 = ops.quantize_e4m3(ai, 0)
+b8 = ops.quantize_e4m3(bi, 0)
+got = ops.gemm_mx8_nt(a8, 0, b8, 0, out_fp32=True)
+assert torch.equal(got, ai @ bi.t())
+b2 = torch.randint(-2, 3, (256,), device=dev).float()
+w3 = torch.randint(-2, 3, (256,), device=dev).float()
+yd = ops.gemm_mx8_relu_dot(a8, 0, b8, 0, b2, w3)
+assert torch.equal(yd, torch.relu(ai @ bi.t() + b2) @ w3)
 print("SANITIZED-OK")
 """
 
