@@ -83,3 +83,41 @@ def test_metrics_match_sklearn_on_arbitrary_data(ys, seed):
 
     assert abs(m["MAPE"] - mean_absolute_percentage_error(y, yhat)) < 1e-9
     assert abs(m["max_residual"] - max_error(y, yhat)) < 1e-9
+
+
+@given(st.lists(st.floats(min_value=-1e6, max_value=1e6,
+                          allow_nan=False, allow_infinity=False),
+                min_size=1, max_size=64),
+       st.integers(min_value=-20, max_value=20))
+@settings(max_examples=60, deadline=None)
+def test_e4m3_quantise_error_bound_property(vals, e_off):
+    """For ANY finite inputs and any exponent keeping amax in range,
+    quantise->decode error is bounded by RNE on a 3-mantissa-bit grid:
+    err <= 2^-4 * |x| + subnormal floor; and decode values are always
+    finite with matching sign (or zero)."""
+    x = torch.tensor(vals, dtype=torch.float32)
+    amax = float(x.abs().max())
+    e = ops.e4m3_exponent(amax) + max(0, e_off)  # never under-scale
+    e = min(e, 127)
+    codes = ops.quantize_e4m3(x, e)
+    dec = reference.e4m3_decode_cpu(codes, e)
+    assert torch.isfinite(dec).all()
+    bound = x.abs() * (2.0 ** -4) + (2.0 ** (e - 7)) + 1e-12
+    assert ((dec - x).abs() <= bound).all()
+    nz = dec != 0
+    assert (torch.sign(dec[nz]) == torch.sign(x[nz])).all()
+
+
+@given(st.integers(min_value=0, max_value=2**31),
+       st.integers(min_value=1, max_value=6),
+       st.integers(min_value=1, max_value=6))
+@settings(max_examples=20, deadline=None)
+def test_mx8_cpu_gemm_integer_exactness_property(seed, mi, ni):
+    """Integer-valued operands are e4m3-exact, so the CPU MX gemm oracle
+    must equal the fp32 matmul bit-for-bit at any small shape."""
+    g = torch.Generator().manual_seed(seed)
+    a = torch.randint(-8, 9, (mi * 4, 32), generator=g).float()
+    b = torch.randint(-8, 9, (ni * 4, 32), generator=g).float()
+    got = ops.gemm_mx8_nt(ops.quantize_e4m3(a, 0), 0,
+                          ops.quantize_e4m3(b, 0), 0, out_fp32=True)
+    assert torch.equal(got, a @ b.t())
