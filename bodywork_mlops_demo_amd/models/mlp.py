@@ -158,6 +158,14 @@ class GPUMLPRegressor:
                                          self._e_w2, self.b2, self.w3)
             yhat = yhat + self.b3
             return yhat, h1q, None, xn, None, None
+        elif self._use_fused_head(xn.shape[0]):
+            # bf16 fused head: h2 GEMM + relu + rowdot in one kernel
+            # (the fp8 path's structure at bf16 precision)
+            h1 = ops.expand1d_bf16(xn, self.w1_bf, self.b1_bf, relu=True)
+            yhat = ops.linear_relu_dot_bf16(h1, self.W2w_bf, self.b2,
+                                            self.w3)
+            yhat = yhat + self.b3
+            return yhat, h1, None, xn, None, None
         else:
             h1 = ops.expand1d_bf16(xn, self.w1_bf, self.b1_bf, relu=True)
             h2 = ops.linear_bf16(h1, self.W2w_bf, bias=self.b2_bf,
@@ -171,6 +179,11 @@ class GPUMLPRegressor:
         (the K=128 scaled-MFMA kernel serves M%256==0; other batch sizes
         take the bf16 kernel)."""
         return (self.fp8_scoring and self.device.type == "cuda"
+                and m % 256 == 0 and self.hidden % 256 == 0)
+
+    def _use_fused_head(self, m: int) -> bool:
+        """bf16 fused GEMM+relu+rowdot head (same tile gating)."""
+        return (self.device.type == "cuda"
                 and m % 256 == 0 and self.hidden % 256 == 0)
 
     # Exponents on the fp8 path are STATIC: e_w2 and the weight-derived

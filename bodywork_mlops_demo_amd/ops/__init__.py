@@ -488,6 +488,25 @@ def gemm_mx8_relu_dot(
     return h2 @ w3.float()
 
 
+def linear_relu_dot_bf16(
+    x: torch.Tensor,
+    w: torch.Tensor,
+    b2: torch.Tensor,
+    w3: torch.Tensor,
+) -> torch.Tensor:
+    """y[M] = Σ_col relu(x[M,K] @ w[N,K]ᵀ + b2) * w3 — the bf16 MLP
+    scoring forward's hot GEMM and rowdot head fused into one kernel
+    (the bf16 twin of gemm_mx8_relu_dot: the [M,N] activation never
+    reaches HBM).  Requires M%256==0, N%256==0, K%128==0.
+    CPU oracle: fp32 matmul + relu + dot."""
+    if x.device.type == "cuda":
+        core = _core(x.device)
+        return core.gemm8_relu_dot_bf16(x.contiguous(), w.contiguous(),
+                                        b2.contiguous(), w3.contiguous())
+    h2 = torch.relu(x.float() @ w.float().t() + b2.float())
+    return h2 @ w3.float()
+
+
 def transpose_bf16(src: torch.Tensor) -> torch.Tensor:
     """[C,R] bf16 = [R,C] bf16 transposed (64x64 LDS tiles, 16-B I/O)."""
     if src.device.type == "cuda":

@@ -40,6 +40,10 @@ std::tuple<at::Tensor, at::Tensor> linear_relu_mask_bf16_hip(
     const c10::optional<at::Tensor>& bias);
 at::Tensor gemm_tn_bf16_hip(const at::Tensor& a, const at::Tensor& b,
                             bool out_fp32);
+// gemm8.hip
+at::Tensor gemm8_relu_dot_bf16_hip(const at::Tensor& x, const at::Tensor& w,
+                                   const at::Tensor& b2,
+                                   const at::Tensor& w3);
 // gemm_mx8.hip — MX-fp8 (e4m3) K=128 scaled-MFMA path
 at::Tensor quantize_e4m3_hip(const at::Tensor& x, int64_t e);
 at::Tensor expand1d_e4m3_hip(const at::Tensor& x, const at::Tensor& w,
@@ -88,6 +92,8 @@ TORCH_LIBRARY(bodywork_hip, m) {
         "bool relu, bool out_fp32) -> Tensor");
   m.def("gemm_mx8_relu_dot(Tensor a8, int ea, Tensor b8, int eb, "
         "Tensor b2, Tensor w3) -> Tensor");
+  m.def("gemm8_relu_dot_bf16(Tensor x, Tensor w, Tensor b2, Tensor w3) "
+        "-> Tensor");
   m.def("adam_step(Tensor(a!) p, Tensor g, Tensor(b!) m, Tensor(c!) v, "
         "Tensor(d!)? p_bf16, float lr, float beta1, float beta2, float eps, "
         "int t, Tensor? bc) -> ()");
@@ -117,6 +123,7 @@ TORCH_LIBRARY_IMPL(bodywork_hip, CUDA, m) {
   m.impl("expand1d_e4m3", expand1d_e4m3_hip);
   m.impl("gemm_mx8_nt", gemm_mx8_nt_hip);
   m.impl("gemm_mx8_relu_dot", gemm_mx8_relu_dot_hip);
+  m.impl("gemm8_relu_dot_bf16", gemm8_relu_dot_bf16_hip);
   m.impl("adam_step", adam_step_hip);
   m.impl("batch_indices", batch_indices_hip);
   m.impl("transpose_to_bf16", transpose_to_bf16_hip);

@@ -59,6 +59,11 @@ typedef __attribute__((ext_vector_type(4))) float g8_f32x4;
 #define G8_EPI_NONE 0
 #define G8_EPI_BIAS_RELU 1
 #define G8_EPI_MASK 2
+// fused scoring head: y[row] += sum_col relu(acc + bias[col]) * w3[col]
+// (the [M,N] activation never reaches HBM — mirrors gemm_mx8.hip's
+// MX_EPI_RELU_DOT).  In this mode the `mask` kernel argument carries
+// w3 (const float*) and C is y (fp32 [M], caller-zeroed).
+#define G8_EPI_RELU_DOT 3
 
 __device__ __forceinline__ void g8_glds16(const void* gsrc, void* lds_dst) {
   __builtin_amdgcn_global_load_lds(
@@ -86,6 +91,34 @@ __device__ __forceinline__ void g8_epilogue(
     const unsigned char* __restrict__ mask,
     unsigned char* __restrict__ mask_out, void* __restrict__ C, long long M,
     long long N, long long m0, long long n0, int wm, int wn, int fl, int kg) {
+  if (EPI == G8_EPI_RELU_DOT) {
+    const float* w3 = (const float*)mask;
+    float w3v[4], b2v[4];
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      long long col = n0 + wn * 64 + j * 16 + fl;
+      w3v[j] = w3[col];
+      b2v[j] = HAS_BIAS ? bias[col] : 0.0f;
+    }
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        float part = 0.0f;
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          part = fmaf(fmaxf(acc[i][j][r] + b2v[j], 0.0f), w3v[j], part);
+#pragma unroll
+        for (int m = 1; m < 16; m <<= 1)
+          part += __shfl_xor(part, m, 64);  // reduce across fl (lane&15)
+        if (fl == 0) {
+          long long row = m0 + wm * 64 + i * 16 + kg * 4 + r;
+          atomicAdd((float*)C + row, part);
+        }
+      }
+    }
+    return;
+  }
 #pragma unroll
   for (int i = 0; i < 4; ++i) {
     unsigned long long mrow[4];
@@ -589,6 +622,37 @@ __global__ void gemm_nt_8phase_w8_kernel(
 
   g8w_epilogue<EPI, HAS_BIAS, OUT_FP32, EMIT_MASK>(
       acc, bias, mask, mask_out, C, M, N, m0, n0, wm2, wn4, fl, kg);
+}
+
+// y[M] = sum_col relu(x.w^T + b2) * w3 — the bf16 MLP scoring forward's
+// h2 GEMM and rowdot head in ONE kernel (16-wave variant only; the
+// [M,N] activation tensor never reaches HBM).  Caller adds b3.
+at::Tensor gemm8_relu_dot_bf16_hip(const at::Tensor& x, const at::Tensor& w,
+                                   const at::Tensor& b2,
+                                   const at::Tensor& w3) {
+  TORCH_CHECK(x.is_cuda() && w.is_cuda() && x.dim() == 2 && w.dim() == 2 &&
+                  x.size(1) == w.size(1),
+              "gemm8_relu_dot: [M,K]x[N,K] cuda");
+  TORCH_CHECK(x.scalar_type() == at::kBFloat16 &&
+              w.scalar_type() == at::kBFloat16);
+  TORCH_CHECK(x.is_contiguous() && w.is_contiguous());
+  long long M = x.size(0), K = x.size(1), N = w.size(0);
+  TORCH_CHECK(M % G8_BM == 0 && N % G8_BN == 0 && K % (2 * G8_BK) == 0,
+              "gemm8_relu_dot requires M%256==0, N%256==0, K%128==0");
+  TORCH_CHECK(b2.is_cuda() && b2.scalar_type() == at::kFloat &&
+                  b2.numel() == N, "b2: fp32 [N]");
+  TORCH_CHECK(w3.is_cuda() && w3.scalar_type() == at::kFloat &&
+                  w3.numel() == N, "w3: fp32 [N]");
+  auto y = at::zeros({M}, x.options().dtype(at::kFloat));
+  dim3 grid((unsigned)(N / G8_BN), (unsigned)(M / G8_BM));
+  auto stream = at::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL(
+      (gemm_nt_8phase_kernel<G8_EPI_RELU_DOT, true, true, false>), grid,
+      dim3(G8_THREADS), 0, stream, (const bf16_t*)x.data_ptr(),
+      (const bf16_t*)w.data_ptr(), b2.data_ptr<float>(),
+      (const unsigned char*)w3.data_ptr<float>(), nullptr, y.data_ptr(),
+      M, N, K);
+  return y;
 }
 
 // ---- launcher (called from gemm.hip's dispatch) ---------------------------

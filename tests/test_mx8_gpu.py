@@ -232,3 +232,30 @@ def test_mx8_supertile_clamp_path():
     y = ops.gemm_mx8_relu_dot(ops.quantize_e4m3(a, 0), 0,
                               ops.quantize_e4m3(b, 0), 0, b2, w3)
     assert torch.equal(y, torch.relu(a @ b.t() + b2) @ w3)
+
+
+def test_bf16_fused_head_matches_unfused():
+    """The bf16 twin of the fused GEMM+relu+rowdot head: must match the
+    unfused pipeline (GEMM -> +b2 -> relu -> @w3) at fp32-accumulation
+    tolerance, and the default MLP predict routes through it."""
+    g = torch.Generator(device=DEV).manual_seed(51)
+    x = (torch.randn(512, 512, generator=g, device=DEV) * 0.5).bfloat16()
+    w = (torch.randn(512, 512, generator=g, device=DEV) * 0.5).bfloat16()
+    b2 = torch.randn(512, generator=g, device=DEV)
+    w3 = torch.randn(512, generator=g, device=DEV)
+    y = ops.linear_relu_dot_bf16(x, w, b2, w3)
+    h2 = torch.relu(ops.linear_bf16(x, w, out_fp32=True) + b2)
+    want = h2 @ w3
+    torch.testing.assert_close(y, want, rtol=2e-3, atol=1e-2)
+
+    # predict-level: fused-head output == manual forward of same weights
+    from bodywork_mlops_demo_amd.models import GPUMLPRegressor
+
+    m = GPUMLPRegressor(hidden=512, device=DEV, seed=8)
+    X = torch.rand(2048, generator=g, device=DEV) * 100
+    p = m.predict(X)  # routes through the fused head (2048 % 256 == 0)
+    xn = (X.float() - m.X_MU) / m.X_SIGMA
+    h1 = ops.expand1d_bf16(xn, m.w1_bf, m.b1_bf, relu=True)
+    h2m = torch.relu(h1.float() @ m.W2w_bf.float().t() + m.b2)
+    want_p = h2m @ m.w3 + m.b3
+    torch.testing.assert_close(p, want_p, rtol=2e-2, atol=2e-2)
