@@ -141,7 +141,8 @@ __global__ void gemm_mx8_nt_kernel(
     const float* __restrict__ bias, const float* __restrict__ w3,
     void* __restrict__ C, long long M,
     long long N, long long K, int sa, int sb,
-    int super) {  // sa/sb: E8M0 bytes; super: supertile height in y-blocks
+    int super,      // supertile height in y-blocks (sa/sb: E8M0 bytes)
+    int xcd_aware) {
   __shared__ char lds[8 * MX_HTB];  // ONE __shared__ object (guide trap (a))
   // Supertile blockIdx remap (super > 1): consecutive dispatch ids walk
   // an (x-cols x super-rows) PATCH instead of a full grid-x row, so the
@@ -153,6 +154,15 @@ __global__ void gemm_mx8_nt_kernel(
   int bx = blockIdx.x, by = blockIdx.y;
   if (super > 1) {
     int linear = blockIdx.x + blockIdx.y * gridDim.x;
+    // XCD-aware transpose (env BODYWORK_MX_XCD=0 disables): dispatch
+    // round-robins workgroups across the 8 XCDs (block i -> XCD i%8),
+    // each with its OWN L2 — a naive patch spreads over 8 disjoint
+    // L2s.  Mapping i -> (i%8)*(G/8) + i/8 gives each XCD a CONTIGUOUS
+    // run of the supertile order, i.e. a compact sub-patch resident in
+    // ITS L2 (guide: +10-12% when HBM-bound).
+    const int G = (int)(gridDim.x * gridDim.y);
+    if (xcd_aware && G % 8 == 0)
+      linear = (linear % 8) * (G / 8) + (linear / 8);
     int band_sz = super * gridDim.x;
     int band = linear / band_sz;
     int in_band = linear % band_sz;
@@ -399,6 +409,14 @@ static int mx_supertile() {
   return v;
 }
 
+static int mx_xcd_aware() {
+  static int v = [] {
+    const char* e = getenv("BODYWORK_MX_XCD");
+    return e ? atoi(e) : 1;
+  }();
+  return v;
+}
+
 at::Tensor expand1d_e4m3_hip(const at::Tensor& x, const at::Tensor& w,
                              const c10::optional<at::Tensor>& b,
                              int64_t e) {
@@ -493,7 +511,8 @@ at::Tensor gemm_mx8_nt_hip(const at::Tensor& a8, int64_t ea,
 #define LMX(EPI_, HB_, OF_)                                                 \
   hipLaunchKernelGGL((gemm_mx8_nt_kernel<EPI_, HB_, OF_>), grid,            \
                      dim3(MX_THREADS), 0, stream, ap, bp, bias_p, nullptr, \
-                     C.data_ptr(), M, N, K, sa, sb, mx_supertile())
+                     C.data_ptr(), M, N, K, sa, sb, mx_supertile(),       \
+                     mx_xcd_aware())
   if (relu) {
     if (has_bias) { if (out_fp32) LMX(MX_EPI_BIAS_RELU, true, true);
                     else          LMX(MX_EPI_BIAS_RELU, true, false); }
@@ -539,6 +558,6 @@ at::Tensor gemm_mx8_relu_dot_hip(const at::Tensor& a8, int64_t ea,
                      a8.data_ptr<unsigned char>(),
                      b8.data_ptr<unsigned char>(), b2.data_ptr<float>(),
                      w3.data_ptr<float>(), y.data_ptr(), M, N, K, sa, sb,
-                     mx_supertile());
+                     mx_supertile(), mx_xcd_aware());
   return y;
 }
