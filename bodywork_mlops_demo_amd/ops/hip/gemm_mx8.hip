@@ -399,12 +399,15 @@ __global__ void expand1d_e4m3_kernel(const float* __restrict__ x,
 // ---- host wrappers --------------------------------------------------------
 
 // supertile height (y-blocks) for the blockIdx remap; latched once.
-// 16 -> a 16x16-block patch (~32 MB of A+B tiles) per 256 concurrent
-// workgroups.  BODYWORK_MX_SUPER=1 disables, other values A/B.
+// DEFAULT 1 (identity): the A/B matrix across supertile heights and
+// the XCD transpose measured within +-2-4% of the natural dispatch
+// order on every shape (profiles/r02_mx8_pmc.md) — L2 + dispatch
+// overlap already cover the tile re-streaming, so the least machinery
+// stays in play.  Envs kept for future A/B on other shapes.
 static int mx_supertile() {
   static int v = [] {
     const char* e = getenv("BODYWORK_MX_SUPER");
-    return e ? atoi(e) : 16;
+    return e ? atoi(e) : 1;
   }();
   return v;
 }
@@ -412,7 +415,7 @@ static int mx_supertile() {
 static int mx_xcd_aware() {
   static int v = [] {
     const char* e = getenv("BODYWORK_MX_XCD");
-    return e ? atoi(e) : 1;
+    return e ? atoi(e) : 0;
   }();
   return v;
 }
