@@ -62,11 +62,42 @@ def drift_report(store: ArtefactStore) -> dict:
     return report
 
 
+def plot_drift(report: dict, path: str) -> str:
+    """Render the offline-vs-online MAPE history and the drift gap to a
+    PNG (the reference analytics notebook's seaborn plots as a durable
+    CLI artefact — `model-performance-analytics.ipynb` capability)."""
+    import matplotlib
+
+    matplotlib.use("Agg")
+    import matplotlib.pyplot as plt
+
+    j = report["joined"]
+    fig, (ax1, ax2) = plt.subplots(2, 1, figsize=(10, 6), sharex=True)
+    ax1.plot(j["date"], j["MAPE_offline"], "o-",
+             label="offline MAPE (train-time)")
+    ax1.plot(j["date"], j["MAPE_online"], "s-",
+             label="online MAPE (live service, t+1 data)")
+    ax1.set_yscale("log")
+    ax1.legend()
+    ax1.grid(alpha=0.3)
+    ax1.set_title("model quality across pipeline days")
+    ax2.plot(j["date"], j["MAPE_online"] - j["MAPE_offline"], "k.-")
+    ax2.axhline(0, color="gray", lw=0.5)
+    ax2.grid(alpha=0.3)
+    ax2.set_title("drift gap: online - offline MAPE")
+    fig.autofmt_xdate()
+    fig.savefig(path, dpi=90, bbox_inches="tight")
+    plt.close(fig)
+    return path
+
+
 def main(argv=None) -> None:
     p = argparse.ArgumentParser(description=__doc__)
     p.add_argument("--store", default=None)
     p.add_argument("--csv-out", default=None,
                    help="write the joined history as CSV")
+    p.add_argument("--plot", default=None, metavar="PNG",
+                   help="render the drift history to a PNG")
     args = p.parse_args(argv)
     report = drift_report(open_store(args.store))
     if "joined" in report:
@@ -75,6 +106,8 @@ def main(argv=None) -> None:
         print("\nsummary:", report["summary"])
         if args.csv_out:
             report["joined"].to_csv(args.csv_out, index=False)
+        if args.plot:
+            print("wrote", plot_drift(report, args.plot))
     else:
         print("no joint metric history yet "
               f"(offline={len(report['offline'])}, online={len(report['online'])})")

@@ -225,3 +225,25 @@ def test_notebooks_carry_executed_outputs():
     md = "".join("".join(c["source"]) for c in serve["cells"]
                  if c["cell_type"] == "markdown")
     assert "curl" in md and "/score/v1" in md
+
+
+def test_analytics_plot_artifact(tmp_path):
+    """`analytics --plot out.png` renders the drift history (the
+    reference analytics notebook's plots as a durable CLI artefact)."""
+    from datetime import date
+
+    from bodywork_mlops_demo_amd.monitoring.analytics import (
+        drift_report, plot_drift)
+    from bodywork_mlops_demo_amd.pipeline.loop import run_loop
+    from bodywork_mlops_demo_amd.store import LocalStore
+
+    store = LocalStore(str(tmp_path / "store"))
+    run_loop(store, days=3, n_rows=500, model_type="linear", device="cpu",
+             start_date="2026-04-01")
+    report = drift_report(store)
+    out = str(tmp_path / "drift.png")
+    assert plot_drift(report, out) == out
+    import struct
+
+    with open(out, "rb") as f:
+        assert f.read(8) == b"\x89PNG\r\n\x1a\n"  # real PNG
