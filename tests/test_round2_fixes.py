@@ -337,3 +337,65 @@ def test_run_cycle_http_serving_and_skip_train(tmp_path):
     assert len(keys) == 3
     for k in keys:
         assert store.get_metrics_csv(k)["response_time_kind"] == "http-binary"
+
+
+def test_front_proxy_survives_backend_respawn():
+    """Watchdog integration: a backend dies and is respawned ON THE SAME
+    PORT; the proxy's kept-alive connections to the old process are
+    stale — the retry-once reconnect logic must recover without a
+    client-visible failure."""
+    import requests
+
+    b1 = _echo_backend("gen1")
+    port = b1.server_address[1]
+    proxy = FrontProxy([port]).start()
+    try:
+        url = f"http://127.0.0.1:{proxy.port}/score/v1"
+        assert requests.post(url, data=b"x", timeout=10).json()["tag"] == "gen1"
+        # kill and respawn on the same port (what the watchdog does)
+        b1.shutdown()
+        b1.server_close()
+        import time as _time
+
+        _time.sleep(0.1)
+        b2 = None
+        for _ in range(20):
+            try:
+                b2 = _respawn_echo("gen2", port)
+                break
+            except OSError:
+                _time.sleep(0.2)
+        assert b2 is not None, "could not rebind respawn port"
+        for _ in range(3):
+            r = requests.post(url, json={"X": 1}, timeout=10)
+            assert r.ok and r.json()["tag"] == "gen2"
+        b2.shutdown()
+        b2.server_close()
+    finally:
+        proxy.stop()
+
+
+def _respawn_echo(tag: str, port: int):
+    class H(BaseHTTPRequestHandler):
+        protocol_version = "HTTP/1.1"
+
+        def log_message(self, fmt, *args):
+            pass
+
+        def _respond(self):
+            n = int(self.headers.get("Content-Length") or 0)
+            self.rfile.read(n) if n else b""
+            out = json.dumps({"tag": tag}).encode()
+            self.send_response(200)
+            self.send_header("Content-Type", "application/json")
+            self.send_header("Content-Length", str(len(out)))
+            self.end_headers()
+            self.wfile.write(out)
+
+        do_GET = _respond
+        do_POST = _respond
+
+    srv = ThreadingHTTPServer(("127.0.0.1", port), H)
+    srv.daemon_threads = True
+    threading.Thread(target=srv.serve_forever, daemon=True).start()
+    return srv
