@@ -357,6 +357,155 @@ __global__ void quantize_e4m3_kernel(const T* __restrict__ x,
   }
 }
 
+// ---- single-phase-per-tile variant (env BODYWORK_MX_1P=1) -----------------
+//
+// ONE barrier per K-tile instead of two: A stays double-buffered, B
+// goes TRIPLE-buffered (LDS = 2x32 + 3x32 = 160 KB, the full CU), so
+// every overwrite stays barrier-separated with half the barrier count
+// per MFMA.  Schedule per tile T (slots: A T%2, B T%3; 6-phase unroll
+// so every slot index is compile-time):
+//
+//   wait vmcnt(2) [vmcnt(0) on the last tile]; barrier
+//   issue A(T+1) -> slot (T+1)%2   [overwrites A(T-1), read last phase]
+//   issue B(T+2) -> slot (T+2)%3   [overwrites B(T-1), read last phase]
+//   read B(T) x4 frags + A mfrag 0,1; 8 MFMAs; read A mfrag 2,3; 8 MFMAs
+//
+// Landing proof: per phase the issue order is [A(T+1), B(T+2)], so at
+// wait(T) the 2 newest outstanding glds are B(T+1)'s halves (issued at
+// phase T-1 after A(T)) — vmcnt(2) proves A(T) and everything older
+// (incl. B(T), issued at phase T-2) landed.  Near the end the guarded
+// issues only SHRINK the outstanding set, and the final tile waits
+// vmcnt(0).  Slot safety: both overwrites target buffers whose reads
+// were consumed by the PREVIOUS phase's MFMAs, before this phase's
+// barrier.
+#define P1_ASLOT(buf, half) (lds + ((buf) * 2 + (half)) * MX_HTB)
+#define P1_BSLOT(buf, half) (lds + (4 + (buf) * 2 + (half)) * MX_HTB)
+
+template <int EPI, bool HAS_BIAS, bool OUT_FP32>
+__launch_bounds__(MX_THREADS)
+__global__ void gemm_mx8_nt_1p_kernel(
+    const unsigned char* __restrict__ A, const unsigned char* __restrict__ B,
+    const float* __restrict__ bias, const float* __restrict__ w3,
+    void* __restrict__ C, long long M, long long N, long long K, int sa,
+    int sb) {
+  __shared__ char lds[10 * MX_HTB];  // A 2 slots + B 3 slots, 160 KB
+  const long long m0 = (long long)blockIdx.y * MX_BM;
+  const long long n0 = (long long)blockIdx.x * MX_BN;
+  const int wave = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int wm = wave >> 2;
+  const int wn = wave & 3;
+  const int fl = lane & 15;
+  const int kg = lane >> 4;
+  const int swz = fl & 3;
+  const long long nt = K / MX_BK;
+  const int a_inhalf = (wm & 1) * 64;
+  const int b_inhalf = (wn & 1) * 64;
+
+  mx_f32x4 acc[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+#define P1_READ8(dst, rowbase, rowoff)                                      \
+  do {                                                                      \
+    const char* _rb = (rowbase) + (rowoff) * 128 + ((kg ^ swz) * 32);       \
+    ((mx_i32x4*)&(dst))[0] = *(const mx_i32x4*)(_rb);                       \
+    ((mx_i32x4*)&(dst))[1] = *(const mx_i32x4*)(_rb + 16);                  \
+  } while (0)
+#define P1_AREAD(dst, buf, mfrag)                                           \
+  P1_READ8(dst, P1_ASLOT(buf, (wm >> 1)), a_inhalf + (mfrag) * 16 + fl)
+#define P1_BREAD(dst, buf, nfrag)                                           \
+  P1_READ8(dst, P1_BSLOT(buf, (wn >> 1)), b_inhalf + (nfrag) * 16 + fl)
+
+  mx_i32x8 a_q[2];
+  mx_i32x8 b_t[4];
+
+  int stg_off;
+  {
+    int ci = (int)threadIdx.x;
+    int row = ci >> 3;
+    int c32 = ((ci & 7) >> 1) ^ (row & 3);
+    stg_off = (int)(row * K + (c32 * 32 + (ci & 1) * 16));
+  }
+  const char* Ah0 = (const char*)(A + m0 * K);
+  const char* Ah1 = (const char*)(A + (m0 + 128) * K);
+  const char* Bh0 = (const char*)(B + n0 * K);
+  const char* Bh1 = (const char*)(B + (n0 + 128) * K);
+
+  // prologue: A(0), B(0), B(1)
+  mx_stage_half(P1_ASLOT(0, 0), Ah0, stg_off);
+  mx_stage_half(P1_ASLOT(0, 1), Ah1, stg_off);
+  mx_stage_half(P1_BSLOT(0, 0), Bh0, stg_off);
+  mx_stage_half(P1_BSLOT(0, 1), Bh1, stg_off);
+  if (nt > 1) {
+    mx_stage_half(P1_BSLOT(1, 0), Bh0 + MX_KOFF(1), stg_off);
+    mx_stage_half(P1_BSLOT(1, 1), Bh1 + MX_KOFF(1), stg_off);
+  }
+
+#define P1_MFMA(acc_i, afrag)                                               \
+  _Pragma("unroll") for (int nf = 0; nf < 4; ++nf)                          \
+      acc[acc_i][nf] = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(    \
+          afrag, b_t[nf], acc[acc_i][nf], 0, 0, 0, sa, 0, sb)
+
+#define P1_PHASE(T, AS, AN, BS, BN2)                                        \
+  do {                                                                      \
+    if ((T) + 1 < nt)                                                       \
+      asm volatile("s_waitcnt vmcnt(2)" ::: "memory");                      \
+    else                                                                    \
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");                      \
+    __builtin_amdgcn_s_barrier();                                           \
+    if ((T) + 1 < nt) {                                                     \
+      mx_stage_half(P1_ASLOT(AN, 0), Ah0 + MX_KOFF((T) + 1), stg_off);      \
+      mx_stage_half(P1_ASLOT(AN, 1), Ah1 + MX_KOFF((T) + 1), stg_off);      \
+    }                                                                       \
+    if ((T) + 2 < nt) {                                                     \
+      mx_stage_half(P1_BSLOT(BN2, 0), Bh0 + MX_KOFF((T) + 2), stg_off);     \
+      mx_stage_half(P1_BSLOT(BN2, 1), Bh1 + MX_KOFF((T) + 2), stg_off);     \
+    }                                                                       \
+    _Pragma("unroll") for (int nf = 0; nf < 4; ++nf)                        \
+        P1_BREAD(b_t[nf], BS, nf);                                          \
+    P1_AREAD(a_q[0], AS, 0);                                                \
+    P1_AREAD(a_q[1], AS, 1);                                                \
+    __builtin_amdgcn_s_setprio(1);                                          \
+    P1_MFMA(0, a_q[0]);                                                     \
+    P1_MFMA(1, a_q[1]);                                                     \
+    __builtin_amdgcn_s_setprio(0);                                          \
+    P1_AREAD(a_q[0], AS, 2);                                                \
+    P1_AREAD(a_q[1], AS, 3);                                                \
+    __builtin_amdgcn_s_setprio(1);                                          \
+    P1_MFMA(2, a_q[0]);                                                     \
+    P1_MFMA(3, a_q[1]);                                                     \
+    __builtin_amdgcn_s_setprio(0);                                          \
+  } while (0)
+
+  // 6-tile unroll (lcm of the A/B buffer periods); t stays ≡ 0 (mod 6)
+  // at the tail, so the tail phases' slot indices are compile-time too
+  long long t = 0;
+  for (; t + 6 <= nt; t += 6) {
+    P1_PHASE(t + 0, 0, 1, 0, 2);
+    P1_PHASE(t + 1, 1, 0, 1, 0);
+    P1_PHASE(t + 2, 0, 1, 2, 1);
+    P1_PHASE(t + 3, 1, 0, 0, 2);
+    P1_PHASE(t + 4, 0, 1, 1, 0);
+    P1_PHASE(t + 5, 1, 0, 2, 1);
+  }
+  if (t + 0 < nt) P1_PHASE(t + 0, 0, 1, 0, 2);
+  if (t + 1 < nt) P1_PHASE(t + 1, 1, 0, 1, 0);
+  if (t + 2 < nt) P1_PHASE(t + 2, 0, 1, 2, 1);
+  if (t + 3 < nt) P1_PHASE(t + 3, 1, 0, 0, 2);
+  if (t + 4 < nt) P1_PHASE(t + 4, 0, 1, 1, 0);
+#undef P1_PHASE
+#undef P1_MFMA
+#undef P1_AREAD
+#undef P1_BREAD
+#undef P1_READ8
+
+  mx_epilogue<EPI, HAS_BIAS, OUT_FP32>(acc, bias, w3, C, M, N, m0, n0, wm,
+                                       wn, fl, kg);
+}
+
 // ---- fused rank-1 expand -> e4m3 ------------------------------------------
 //
 // The MLP fp8 scoring forward's layer 1: h1 = relu(x*w + b) emitted
@@ -415,6 +564,16 @@ static int mx_supertile() {
 static int mx_xcd_aware() {
   static int v = [] {
     const char* e = getenv("BODYWORK_MX_XCD");
+    return e ? atoi(e) : 0;
+  }();
+  return v;
+}
+
+// single-phase-per-tile variant (1 barrier/K-tile, B triple-buffered,
+// 160 KB LDS).  Opt-in until its A/B wins on hardware.
+static int mx_one_phase() {
+  static int v = [] {
+    const char* e = getenv("BODYWORK_MX_1P");
     return e ? atoi(e) : 0;
   }();
   return v;
@@ -512,10 +671,17 @@ at::Tensor gemm_mx8_nt_hip(const at::Tensor& a8, int64_t ea,
   const unsigned char* ap = a8.data_ptr<unsigned char>();
   const unsigned char* bp = b8.data_ptr<unsigned char>();
 #define LMX(EPI_, HB_, OF_)                                                 \
-  hipLaunchKernelGGL((gemm_mx8_nt_kernel<EPI_, HB_, OF_>), grid,            \
-                     dim3(MX_THREADS), 0, stream, ap, bp, bias_p, nullptr, \
-                     C.data_ptr(), M, N, K, sa, sb, mx_supertile(),       \
-                     mx_xcd_aware())
+  do {                                                                      \
+    if (mx_one_phase())                                                     \
+      hipLaunchKernelGGL((gemm_mx8_nt_1p_kernel<EPI_, HB_, OF_>), grid,     \
+                         dim3(MX_THREADS), 0, stream, ap, bp, bias_p,       \
+                         nullptr, C.data_ptr(), M, N, K, sa, sb);           \
+    else                                                                    \
+      hipLaunchKernelGGL((gemm_mx8_nt_kernel<EPI_, HB_, OF_>), grid,        \
+                         dim3(MX_THREADS), 0, stream, ap, bp, bias_p,       \
+                         nullptr, C.data_ptr(), M, N, K, sa, sb,            \
+                         mx_supertile(), mx_xcd_aware());                   \
+  } while (0)
   if (relu) {
     if (has_bias) { if (out_fp32) LMX(MX_EPI_BIAS_RELU, true, true);
                     else          LMX(MX_EPI_BIAS_RELU, true, false); }
@@ -556,11 +722,18 @@ at::Tensor gemm_mx8_relu_dot_hip(const at::Tensor& a8, int64_t ea,
   auto y = torch::zeros({M}, a8.options().dtype(torch::kFloat));
   dim3 grid((unsigned)(N / MX_BN), (unsigned)(M / MX_BM));
   auto stream = at::cuda::getCurrentCUDAStream();
-  hipLaunchKernelGGL((gemm_mx8_nt_kernel<MX_EPI_RELU_DOT, true, true>),
-                     grid, dim3(MX_THREADS), 0, stream,
-                     a8.data_ptr<unsigned char>(),
-                     b8.data_ptr<unsigned char>(), b2.data_ptr<float>(),
-                     w3.data_ptr<float>(), y.data_ptr(), M, N, K, sa, sb,
-                     mx_supertile(), mx_xcd_aware());
+  if (mx_one_phase())
+    hipLaunchKernelGGL((gemm_mx8_nt_1p_kernel<MX_EPI_RELU_DOT, true, true>),
+                       grid, dim3(MX_THREADS), 0, stream,
+                       a8.data_ptr<unsigned char>(),
+                       b8.data_ptr<unsigned char>(), b2.data_ptr<float>(),
+                       w3.data_ptr<float>(), y.data_ptr(), M, N, K, sa, sb);
+  else
+    hipLaunchKernelGGL((gemm_mx8_nt_kernel<MX_EPI_RELU_DOT, true, true>),
+                       grid, dim3(MX_THREADS), 0, stream,
+                       a8.data_ptr<unsigned char>(),
+                       b8.data_ptr<unsigned char>(), b2.data_ptr<float>(),
+                       w3.data_ptr<float>(), y.data_ptr(), M, N, K, sa, sb,
+                       mx_supertile(), mx_xcd_aware());
   return y;
 }
