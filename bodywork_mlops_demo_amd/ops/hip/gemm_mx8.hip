@@ -398,7 +398,7 @@ __global__ void gemm_mx8_nt_1p_kernel(
   const int fl = lane & 15;
   const int kg = lane >> 4;
   const int swz = fl & 3;
-  const long long nt = K / MX_BK;
+  const int nt = (int)(K / MX_BK);
   const int a_inhalf = (wm & 1) * 64;
   const int b_inhalf = (wn & 1) * 64;
 
@@ -408,16 +408,26 @@ __global__ void gemm_mx8_nt_1p_kernel(
 #pragma unroll
     for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
+  // slot indices only affect LDS ADDRESSES (never register-array
+  // indices), so the A slots swap and the B slots rotate as plain
+  // pointer variables — the loop body is ONE phase, which keeps
+  // register pressure low (a 6-tile compile-time unroll of this
+  // schedule spilled ~400 VGPRs)
+  char* aw_cur = P1_ASLOT(0, 0);
+  char* aw_nxt = P1_ASLOT(1, 0);
+  char* bw0 = P1_BSLOT(0, 0);  // holds B(T)
+  char* bw1 = P1_BSLOT(1, 0);  // holds B(T+1) (in flight or staged)
+  char* bw2 = P1_BSLOT(2, 0);  // staging target for B(T+2)
+  // this wave reads its half directly: precompute read bases
+#define P1_AR_BASE(slotp) ((slotp) + (wm >> 1) * MX_HTB)
+#define P1_BR_BASE(slotp) ((slotp) + (wn >> 1) * MX_HTB)
+
 #define P1_READ8(dst, rowbase, rowoff)                                      \
   do {                                                                      \
     const char* _rb = (rowbase) + (rowoff) * 128 + ((kg ^ swz) * 32);       \
     ((mx_i32x4*)&(dst))[0] = *(const mx_i32x4*)(_rb);                       \
     ((mx_i32x4*)&(dst))[1] = *(const mx_i32x4*)(_rb + 16);                  \
   } while (0)
-#define P1_AREAD(dst, buf, mfrag)                                           \
-  P1_READ8(dst, P1_ASLOT(buf, (wm >> 1)), a_inhalf + (mfrag) * 16 + fl)
-#define P1_BREAD(dst, buf, nfrag)                                           \
-  P1_READ8(dst, P1_BSLOT(buf, (wn >> 1)), b_inhalf + (nfrag) * 16 + fl)
 
   mx_i32x8 a_q[2];
   mx_i32x8 b_t[4];
@@ -435,13 +445,13 @@ __global__ void gemm_mx8_nt_1p_kernel(
   const char* Bh1 = (const char*)(B + (n0 + 128) * K);
 
   // prologue: A(0), B(0), B(1)
-  mx_stage_half(P1_ASLOT(0, 0), Ah0, stg_off);
-  mx_stage_half(P1_ASLOT(0, 1), Ah1, stg_off);
-  mx_stage_half(P1_BSLOT(0, 0), Bh0, stg_off);
-  mx_stage_half(P1_BSLOT(0, 1), Bh1, stg_off);
+  mx_stage_half(aw_cur, Ah0, stg_off);
+  mx_stage_half(aw_cur + MX_HTB, Ah1, stg_off);
+  mx_stage_half(bw0, Bh0, stg_off);
+  mx_stage_half(bw0 + MX_HTB, Bh1, stg_off);
   if (nt > 1) {
-    mx_stage_half(P1_BSLOT(1, 0), Bh0 + MX_KOFF(1), stg_off);
-    mx_stage_half(P1_BSLOT(1, 1), Bh1 + MX_KOFF(1), stg_off);
+    mx_stage_half(bw1, Bh0 + MX_KOFF(1), stg_off);
+    mx_stage_half(bw1 + MX_HTB, Bh1 + MX_KOFF(1), stg_off);
   }
 
 #define P1_MFMA(acc_i, afrag)                                               \
@@ -449,58 +459,49 @@ __global__ void gemm_mx8_nt_1p_kernel(
       acc[acc_i][nf] = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(    \
           afrag, b_t[nf], acc[acc_i][nf], 0, 0, 0, sa, 0, sb)
 
-#define P1_PHASE(T, AS, AN, BS, BN2)                                        \
-  do {                                                                      \
-    if ((T) + 1 < nt)                                                       \
-      asm volatile("s_waitcnt vmcnt(2)" ::: "memory");                      \
-    else                                                                    \
-      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");                      \
-    __builtin_amdgcn_s_barrier();                                           \
-    if ((T) + 1 < nt) {                                                     \
-      mx_stage_half(P1_ASLOT(AN, 0), Ah0 + MX_KOFF((T) + 1), stg_off);      \
-      mx_stage_half(P1_ASLOT(AN, 1), Ah1 + MX_KOFF((T) + 1), stg_off);      \
-    }                                                                       \
-    if ((T) + 2 < nt) {                                                     \
-      mx_stage_half(P1_BSLOT(BN2, 0), Bh0 + MX_KOFF((T) + 2), stg_off);     \
-      mx_stage_half(P1_BSLOT(BN2, 1), Bh1 + MX_KOFF((T) + 2), stg_off);     \
-    }                                                                       \
-    _Pragma("unroll") for (int nf = 0; nf < 4; ++nf)                        \
-        P1_BREAD(b_t[nf], BS, nf);                                          \
-    P1_AREAD(a_q[0], AS, 0);                                                \
-    P1_AREAD(a_q[1], AS, 1);                                                \
-    __builtin_amdgcn_s_setprio(1);                                          \
-    P1_MFMA(0, a_q[0]);                                                     \
-    P1_MFMA(1, a_q[1]);                                                     \
-    __builtin_amdgcn_s_setprio(0);                                          \
-    P1_AREAD(a_q[0], AS, 2);                                                \
-    P1_AREAD(a_q[1], AS, 3);                                                \
-    __builtin_amdgcn_s_setprio(1);                                          \
-    P1_MFMA(2, a_q[0]);                                                     \
-    P1_MFMA(3, a_q[1]);                                                     \
-    __builtin_amdgcn_s_setprio(0);                                          \
-  } while (0)
-
-  // 6-tile unroll (lcm of the A/B buffer periods); t stays ≡ 0 (mod 6)
-  // at the tail, so the tail phases' slot indices are compile-time too
-  long long t = 0;
-  for (; t + 6 <= nt; t += 6) {
-    P1_PHASE(t + 0, 0, 1, 0, 2);
-    P1_PHASE(t + 1, 1, 0, 1, 0);
-    P1_PHASE(t + 2, 0, 1, 2, 1);
-    P1_PHASE(t + 3, 1, 0, 0, 2);
-    P1_PHASE(t + 4, 0, 1, 1, 0);
-    P1_PHASE(t + 5, 1, 0, 2, 1);
+  for (int t = 0; t < nt; ++t) {
+    if (t + 1 < nt)
+      asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
+    else
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    if (t + 1 < nt) {
+      mx_stage_half(aw_nxt, Ah0 + MX_KOFF(t + 1), stg_off);
+      mx_stage_half(aw_nxt + MX_HTB, Ah1 + MX_KOFF(t + 1), stg_off);
+    }
+    if (t + 2 < nt) {
+      mx_stage_half(bw2, Bh0 + MX_KOFF(t + 2), stg_off);
+      mx_stage_half(bw2 + MX_HTB, Bh1 + MX_KOFF(t + 2), stg_off);
+    }
+    {
+      const char* br = P1_BR_BASE(bw0);
+#pragma unroll
+      for (int nf = 0; nf < 4; ++nf)
+        P1_READ8(b_t[nf], br, b_inhalf + nf * 16 + fl);
+    }
+    {
+      const char* ar = P1_AR_BASE(aw_cur);
+      P1_READ8(a_q[0], ar, a_inhalf + 0 * 16 + fl);
+      P1_READ8(a_q[1], ar, a_inhalf + 1 * 16 + fl);
+      __builtin_amdgcn_s_setprio(1);
+      P1_MFMA(0, a_q[0]);
+      P1_MFMA(1, a_q[1]);
+      __builtin_amdgcn_s_setprio(0);
+      P1_READ8(a_q[0], ar, a_inhalf + 2 * 16 + fl);
+      P1_READ8(a_q[1], ar, a_inhalf + 3 * 16 + fl);
+      __builtin_amdgcn_s_setprio(1);
+      P1_MFMA(2, a_q[0]);
+      P1_MFMA(3, a_q[1]);
+      __builtin_amdgcn_s_setprio(0);
+    }
+    // rotate slots: A swaps, B rotates forward
+    char* tmp = aw_cur; aw_cur = aw_nxt; aw_nxt = tmp;
+    char* b0 = bw0; bw0 = bw1; bw1 = bw2; bw2 = b0;
   }
-  if (t + 0 < nt) P1_PHASE(t + 0, 0, 1, 0, 2);
-  if (t + 1 < nt) P1_PHASE(t + 1, 1, 0, 1, 0);
-  if (t + 2 < nt) P1_PHASE(t + 2, 0, 1, 2, 1);
-  if (t + 3 < nt) P1_PHASE(t + 3, 1, 0, 0, 2);
-  if (t + 4 < nt) P1_PHASE(t + 4, 0, 1, 1, 0);
-#undef P1_PHASE
 #undef P1_MFMA
-#undef P1_AREAD
-#undef P1_BREAD
 #undef P1_READ8
+#undef P1_AR_BASE
+#undef P1_BR_BASE
 
   mx_epilogue<EPI, HAS_BIAS, OUT_FP32>(acc, bias, w3, C, M, N, m0, n0, wm,
                                        wn, fl, kg);
