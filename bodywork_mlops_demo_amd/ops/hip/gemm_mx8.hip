@@ -571,11 +571,16 @@ static int mx_xcd_aware() {
 }
 
 // single-phase-per-tile variant (1 barrier/K-tile, B triple-buffered,
-// 160 KB LDS).  Opt-in until its A/B wins on hardware.
+// 160 KB LDS).  DEFAULT since its hardware A/B won on every shape
+// (+6-10%: 2042/2412/2186 TF vs 1920/2194/1998 same box) with the full
+// test file green normally AND under AMD_SERIALIZE_KERNEL/COPY,
+// bitwise-repeatable over 20 replays, exact at odd/clamp shapes
+// (profiles/r02_mx8_1p.md).  BODYWORK_MX_1P=0 reverts to the 2-phase
+// kernel.
 static int mx_one_phase() {
   static int v = [] {
     const char* e = getenv("BODYWORK_MX_1P");
-    return e ? atoi(e) : 0;
+    return e ? atoi(e) : 1;
   }();
   return v;
 }
