@@ -357,6 +357,139 @@ __global__ void gemm_nt_8phase_kernel(
       acc, bias, mask, mask_out, C, M, N, m0, n0, wm, wn, fl, kg);
 }
 
+// ---- single-phase-per-tile variant (env BODYWORK_GEMM_1P) -----------------
+//
+// The bf16 port of gemm_mx8.hip's single-phase schedule: ONE barrier
+// per K-tile (the 2-phase schedule above pays two), A double-buffered,
+// B TRIPLE-buffered into the full 160 KB LDS.  Slot indices only
+// affect LDS addresses, so the slots swap/rotate as plain pointer
+// variables and the loop body stays one phase (a compile-time slot
+// unroll of this schedule spilled ~400 VGPRs on the fp8 twin —
+// profiles/r02_mx8_1p.md).  Landing proof: per phase the issue order
+// is [A(T+1), B(T+2)], so at wait(T) the 2 newest outstanding glds are
+// B(T+1)'s halves — vmcnt(2) proves A(T) and everything older landed;
+// the last tile waits vmcnt(0).  Slot safety: both overwrites target
+// buffers read by the PREVIOUS phase, barrier-separated.
+template <int EPI, bool HAS_BIAS, bool OUT_FP32, bool EMIT_MASK = false>
+__launch_bounds__(G8_THREADS)
+__global__ void gemm_nt_1p_kernel(
+    const bf16_t* __restrict__ A, const bf16_t* __restrict__ B,
+    const float* __restrict__ bias, const unsigned char* __restrict__ mask,
+    unsigned char* __restrict__ mask_out, void* __restrict__ C, long long M,
+    long long N, long long K) {
+  __shared__ short lds[10 * G8_HT];  // A 2 slots + B 3 slots = 160 KB
+  const long long m0 = (long long)blockIdx.y * G8_BM;
+  const long long n0 = (long long)blockIdx.x * G8_BN;
+  const int wave = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int wm = wave >> 2;
+  const int wn = wave & 3;
+  const int fl = lane & 15;
+  const int kg = lane >> 4;
+  const int swz = fl & 7;
+  const int nt = (int)(K / G8_BK);
+  const int a_inhalf = (wm & 1) * 64;
+  const int b_inhalf = (wn & 1) * 64;
+
+  g8_f32x4 acc[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  short* aw_cur = lds;                 // A slot 0 (2 halves)
+  short* aw_nxt = lds + 2 * G8_HT;     // A slot 1
+  short* bw0 = lds + 4 * G8_HT;        // B(T)
+  short* bw1 = lds + 6 * G8_HT;        // B(T+1)
+  short* bw2 = lds + 8 * G8_HT;        // staging target for B(T+2)
+
+#define G1_READ(dst, rowbase, rowoff, ks)                                   \
+  dst = ((const lds_vec*)((rowbase) + (rowoff) * 64 +                       \
+                          (((ks) * 4 + kg) ^ swz) * 8))                     \
+            ->v
+
+  bf16x8_v a_q[2][2];  // current mfrag pair [mf&1][ks]
+  bf16x8_v b_t[4][2];
+
+  int stg_off;
+  {
+    int ci = (int)threadIdx.x;
+    int row = ci >> 3;
+    int sc = (ci & 7) ^ (row & 7);
+    stg_off = (int)((row * K + sc * 8) * 2);
+  }
+  const char* Ah0 = (const char*)(A + m0 * K);
+  const char* Ah1 = (const char*)(A + (m0 + 128) * K);
+  const char* Bh0 = (const char*)(B + n0 * K);
+  const char* Bh1 = (const char*)(B + (n0 + 128) * K);
+
+  // prologue: A(0), B(0), B(1)
+  g8_stage_half(aw_cur, Ah0, stg_off);
+  g8_stage_half(aw_cur + G8_HT, Ah1, stg_off);
+  g8_stage_half(bw0, Bh0, stg_off);
+  g8_stage_half(bw0 + G8_HT, Bh1, stg_off);
+  if (nt > 1) {
+    g8_stage_half(bw1, Bh0 + G8_KOFF(1), stg_off);
+    g8_stage_half(bw1 + G8_HT, Bh1 + G8_KOFF(1), stg_off);
+  }
+
+#define G1_MFMA(acc_i, apair)                                               \
+  _Pragma("unroll") for (int nf = 0; nf < 4; ++nf)                          \
+      _Pragma("unroll") for (int ks = 0; ks < 2; ++ks)                      \
+          acc[acc_i][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(         \
+              apair[ks], b_t[nf][ks], acc[acc_i][nf], 0, 0, 0)
+
+  for (int t = 0; t < nt; ++t) {
+    if (t + 1 < nt)
+      asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
+    else
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    if (t + 1 < nt) {
+      g8_stage_half(aw_nxt, Ah0 + G8_KOFF(t + 1), stg_off);
+      g8_stage_half(aw_nxt + G8_HT, Ah1 + G8_KOFF(t + 1), stg_off);
+    }
+    if (t + 2 < nt) {
+      g8_stage_half(bw2, Bh0 + G8_KOFF(t + 2), stg_off);
+      g8_stage_half(bw2 + G8_HT, Bh1 + G8_KOFF(t + 2), stg_off);
+    }
+    {
+      const short* br = bw0 + (wn >> 1) * G8_HT;
+#pragma unroll
+      for (int nf = 0; nf < 4; ++nf) {
+        G1_READ(b_t[nf][0], br, b_inhalf + nf * 16 + fl, 0);
+        G1_READ(b_t[nf][1], br, b_inhalf + nf * 16 + fl, 1);
+      }
+    }
+    {
+      const short* ar = aw_cur + (wm >> 1) * G8_HT;
+      G1_READ(a_q[0][0], ar, a_inhalf + 0 * 16 + fl, 0);
+      G1_READ(a_q[0][1], ar, a_inhalf + 0 * 16 + fl, 1);
+      G1_READ(a_q[1][0], ar, a_inhalf + 1 * 16 + fl, 0);
+      G1_READ(a_q[1][1], ar, a_inhalf + 1 * 16 + fl, 1);
+      __builtin_amdgcn_s_setprio(1);
+      G1_MFMA(0, a_q[0]);
+      G1_MFMA(1, a_q[1]);
+      __builtin_amdgcn_s_setprio(0);
+      G1_READ(a_q[0][0], ar, a_inhalf + 2 * 16 + fl, 0);
+      G1_READ(a_q[0][1], ar, a_inhalf + 2 * 16 + fl, 1);
+      G1_READ(a_q[1][0], ar, a_inhalf + 3 * 16 + fl, 0);
+      G1_READ(a_q[1][1], ar, a_inhalf + 3 * 16 + fl, 1);
+      __builtin_amdgcn_s_setprio(1);
+      G1_MFMA(2, a_q[0]);
+      G1_MFMA(3, a_q[1]);
+      __builtin_amdgcn_s_setprio(0);
+    }
+    short* tmp = aw_cur; aw_cur = aw_nxt; aw_nxt = tmp;
+    short* b0 = bw0; bw0 = bw1; bw1 = bw2; bw2 = b0;
+  }
+#undef G1_MFMA
+#undef G1_READ
+
+  g8_epilogue<EPI, HAS_BIAS, OUT_FP32, EMIT_MASK>(
+      acc, bias, mask, mask_out, C, M, N, m0, n0, wm, wn, fl, kg);
+}
+
 // ---- 8-wave / 512-thread variant (env BODYWORK_GEMM_WAVES=8) --------------
 //
 // Same 256^2 tile and LDS layout, but 8 waves in a 2(M)x4(N) grid with
@@ -669,9 +802,17 @@ void launch_gemm8(int epi, bool has_bias, bool out_fp32, bool emit_mask,
     const char* e = getenv("BODYWORK_GEMM_WAVES");
     return (e && e[0] == '8') ? 1 : 0;
   }();
+  static int onep = [] {  // single-phase variant (see gemm_nt_1p_kernel)
+    const char* e = getenv("BODYWORK_GEMM_1P");
+    return e ? atoi(e) : 0;
+  }();
 #define L8(EPI_, HB_, OF_, EM_)                                             \
   do {                                                                      \
-    if (w8)                                                                 \
+    if (onep)                                                               \
+      hipLaunchKernelGGL((gemm_nt_1p_kernel<EPI_, HB_, OF_, EM_>), grid,    \
+                         dim3(G8_THREADS), 0, stream, a, b, bias,           \
+                         mask, mask_out, cp, M, N, K);                      \
+    else if (w8)                                                            \
       hipLaunchKernelGGL((gemm_nt_8phase_w8_kernel<EPI_, HB_, OF_, EM_>),   \
                          grid, dim3(G8W_THREADS), 0, stream, a, b, bias,    \
                          mask, mask_out, cp, M, N, K);                      \
