@@ -8,11 +8,13 @@ shapes and the MLP fused shapes and prints TFLOP/s.
 """
 import argparse
 import os
+import sys
 import time
 
 import torch
 
-from bodywork_mlops_demo_amd import ops
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+from bodywork_mlops_demo_amd import ops  # noqa: E402
 
 
 def _oracle_nt(x, w):
