@@ -101,3 +101,24 @@ def test_bench_torchrun_world2_cpu(tmp_path):
     assert d["config"]["parallelism"] == "dp2"
     # whole-job rows/sec over 2 ranks of 20k rows each
     assert d["value"] > 0
+
+
+@pytest.mark.timeout(600)
+def test_bench_torchrun_world2_http_serving_cpu(tmp_path):
+    """world=2 with --serving http: each rank runs its OWN uvicorn
+    replica (port 5600+rank) against the shared store — the multi-rank
+    shape of the HTTP-inclusive cycle, rehearsed on gloo/CPU."""
+    env = dict(os.environ, MASTER_ADDR="127.0.0.1")
+    proc = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29637", "bench.py", "--gpus", "2",
+         "--rows", "4000", "--steps", "1", "--warmup", "1",
+         "--serving", "http",
+         "--store", str(tmp_path / "store")],
+        cwd=REPO, capture_output=True, text=True, timeout=540, env=env,
+    )
+    assert proc.returncode == 0, (proc.stdout + proc.stderr)[-3000:]
+    d = _parse_json_line(proc.stdout)
+    assert d["n_gpus"] == 2 and "HTTP" in d["metric"]
+    assert d["value"] > 0

@@ -247,3 +247,35 @@ def test_analytics_plot_artifact(tmp_path):
 
     with open(out, "rb") as f:
         assert f.read(8) == b"\x89PNG\r\n\x1a\n"  # real PNG
+
+
+def test_analytics_mixed_schema_history(tmp_path):
+    """A store with r1-era 5-column test-metrics CSVs AND r2 6-column
+    ones (response_time_kind added) must still join and summarise —
+    history written by an older build stays readable."""
+    from datetime import date
+
+    from bodywork_mlops_demo_amd.monitoring.analytics import drift_report
+    from bodywork_mlops_demo_amd.store import LocalStore, contract
+
+    store = LocalStore(str(tmp_path))
+    for i, d in enumerate((date(2026, 7, 1), date(2026, 7, 2))):
+        store.put_metrics_csv(
+            contract.model_metrics_key(d),
+            ["date", "MAPE", "r_squared", "max_residual"],
+            [d, 1.0 + i, 0.9, 5.0])
+    # day 1: old 5-column schema; day 2: new 6-column schema
+    store.put_metrics_csv(
+        contract.test_metrics_key(date(2026, 7, 1)),
+        ["date", "MAPE", "r_squared", "max_residual", "mean_response_time"],
+        [date(2026, 7, 1), 1.5, 0.85, 6.0, 1e-6])
+    store.put_metrics_csv(
+        contract.test_metrics_key(date(2026, 7, 2)),
+        ["date", "MAPE", "r_squared", "max_residual", "mean_response_time",
+         "response_time_kind"],
+        [date(2026, 7, 2), 1.7, 0.84, 6.5, 2e-6, "http-binary"])
+    report = drift_report(store)
+    assert report["summary"]["days"] == 2
+    assert abs(report["summary"]["mean_online_MAPE"] - 1.6) < 1e-9
+    kinds = report["online"]["response_time_kind"].tolist()
+    assert "http-binary" in kinds  # new rows keep the annotation
