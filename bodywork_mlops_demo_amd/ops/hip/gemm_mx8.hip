@@ -1,5 +1,12 @@
-// MX-FP8 (e4m3) 256x256-tile 8-phase NT GEMM for gfx950 — the 2x-rate
-// low-precision path.
+// MX-FP8 (e4m3) 256x256-tile NT GEMM for gfx950 — the 2x-rate
+// low-precision path.  TWO schedules live here:
+//   - gemm_mx8_nt_1p_kernel (DEFAULT): single-phase-per-tile, 1 barrier
+//     per K-tile, B triple-buffered into the full 160 KB LDS — PMC
+//     shows −20% wave cycles vs the 2-phase schedule at identical MFMA
+//     work, measured +6-10% on every shape (profiles/r02_mx8_1p.md);
+//   - gemm_mx8_nt_kernel: the original 2-phase port of gemm8.hip's
+//     8-phase structure (BODYWORK_MX_1P=0), kept as the reference
+//     schedule and for the supertile/XCD remap experiments.
 //
 // Why this shape: the non-scaled fp8 MFMAs (mfma_f32_16x16x32_fp8_fp8)
 // run at the BF16 rate on CDNA4 — the ONLY 2x-rate fp8 instruction is
